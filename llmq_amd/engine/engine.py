@@ -1,0 +1,259 @@
+"""LLMEngine — synchronous continuous-batching engine.
+
+One ``step()`` runs one scheduler iteration (a prefill batch or a decode
+batch), samples, advances sequences, and returns per-request progress.
+The async facade (async_engine.py) drives this loop on a dedicated thread.
+"""
+
+from __future__ import annotations
+
+import logging
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+from llmq_amd import ops
+from llmq_amd.engine.config import EngineConfig
+from llmq_amd.engine.kv_cache import BlockAllocator, KVCache
+from llmq_amd.engine.model_runner import ModelRunner
+from llmq_amd.engine.model_specs import ModelSpec, resolve_spec
+from llmq_amd.engine.models.llama import CausalLM
+from llmq_amd.engine.sampling_params import SamplingParams
+from llmq_amd.engine.scheduler import Scheduler, Sequence
+from llmq_amd.engine.tokenizer import load_tokenizer
+
+logger = logging.getLogger(__name__)
+
+# Memory the engine leaves free on top of weights+KV (workspace, graphs).
+_RESERVE_BYTES = 4 << 30
+
+
+@dataclass
+class RequestOutput:
+    request_id: str
+    new_token_ids: List[int] = field(default_factory=list)
+    finished: bool = False
+    finish_reason: Optional[str] = None
+    text: str = ""
+    prompt_tokens: int = 0
+    output_tokens: int = 0
+    queue_wait_ms: Optional[float] = None
+    prefill_ms: Optional[float] = None
+    decode_ms: Optional[float] = None
+
+
+class LLMEngine:
+    def __init__(self, config: EngineConfig, tp_rank: int = 0, tp_size: Optional[int] = None):
+        self.config = config
+        self.device = config.resolve_device()
+        self.dtype = config.resolve_dtype(self.device)
+        self.spec: ModelSpec = resolve_spec(config.model)
+        tp = tp_size if tp_size is not None else config.tensor_parallel_size
+        self.tp_size = tp
+        self.tp_rank = tp_rank
+        if self.device.type == "cuda" and not ops.has_hip_ext():
+            # Fail loudly: GPU execution must run the CDNA4 kernels.
+            ops._use_hip(torch.empty(1, device=self.device))
+        self.max_model_len = min(
+            config.max_model_len or self.spec.max_position_embeddings,
+            self.spec.max_position_embeddings,
+        )
+        logger.info(
+            "initialising engine: model=%s (%s, %.2fB params) device=%s dtype=%s tp=%d",
+            self.spec.name, self.spec.family, self.spec.param_count() / 1e9,
+            self.device, self.dtype, tp,
+        )
+        self.model = CausalLM(self.spec, self.device, self.dtype, tp_rank, tp)
+        self._load_or_init_weights()
+        self.tokenizer = load_tokenizer(
+            config.model, self.spec.vocab_size, self.spec.bos_token_id, self.spec.eos_token_id
+        )
+        num_blocks = self._size_kv_cache()
+        kv_heads_local = self.spec.num_kv_heads // tp
+        self.kv_cache = KVCache(
+            self.spec.num_layers, num_blocks, kv_heads_local,
+            config.kv_block_size, self.spec.head_dim, self.device, self.dtype,
+        )
+        self.allocator = BlockAllocator(num_blocks)
+        self.scheduler = Scheduler(
+            self.allocator,
+            config.kv_block_size,
+            config.max_num_seqs,
+            config.max_prefill_tokens,
+            self.max_model_len,
+        )
+        self.runner = ModelRunner(
+            self.model, self.kv_cache, config, self.device, self.max_model_len
+        )
+        self.runner.capture_graphs()
+        self._seqs: Dict[str, Sequence] = {}
+        self._prefill_done_at: Dict[str, float] = {}
+        self.steps = 0
+        logger.info(
+            "engine ready: %d KV blocks (%d tokens, %.1f GB), max_num_seqs=%d, graphs=%s",
+            num_blocks, num_blocks * config.kv_block_size,
+            num_blocks * KVCache.block_bytes(
+                self.spec.num_layers, kv_heads_local, config.kv_block_size,
+                self.spec.head_dim, self.dtype,
+            ) / 2 ** 30 * 1.0,
+            config.max_num_seqs,
+            self.runner.use_graphs,
+        )
+
+    # -- init helpers ----------------------------------------------------
+
+    def _load_or_init_weights(self) -> None:
+        from pathlib import Path
+
+        path = Path(self.config.model)
+        if self.config.load_weights and path.is_dir() and any(path.glob("*.safetensors")):
+            from llmq_amd.engine.weights import load_safetensors_weights
+
+            load_safetensors_weights(self.model, path)
+        else:
+            self.model.random_init(self.config.seed)
+
+    def _size_kv_cache(self) -> int:
+        cfg = self.config
+        kv_heads_local = self.spec.num_kv_heads // self.tp_size
+        if cfg.num_kv_blocks is not None:
+            return cfg.num_kv_blocks
+        if self.device.type != "cuda":
+            # CPU tests: enough for max_num_seqs × a modest context.
+            return max(
+                256,
+                (cfg.max_num_seqs * min(self.max_model_len, 1024)) // cfg.kv_block_size,
+            )
+        free, total = torch.cuda.mem_get_info(self.device)
+        used = total - free
+        budget = int(total * cfg.gpu_memory_utilization) - used - _RESERVE_BYTES
+        num = KVCache.num_blocks_for_budget(
+            budget, self.spec.num_layers, kv_heads_local, cfg.kv_block_size,
+            self.spec.head_dim, self.dtype,
+        )
+        # No point holding more KV than every admitted seq at full context.
+        cap = (cfg.max_num_seqs * self.max_model_len) // cfg.kv_block_size + cfg.max_num_seqs
+        num = min(num, cap)
+        if num < 16:
+            raise RuntimeError(
+                f"KV budget too small: {budget / 2**30:.1f} GB free for KV "
+                f"(gpu_memory_utilization={cfg.gpu_memory_utilization})"
+            )
+        return num
+
+    # -- request API -----------------------------------------------------
+
+    def add_request(
+        self,
+        request_id: str,
+        prompt: Optional[str] = None,
+        prompt_token_ids: Optional[List[int]] = None,
+        params: Optional[SamplingParams] = None,
+    ) -> None:
+        if request_id in self._seqs:
+            raise ValueError(f"duplicate request_id {request_id}")
+        params = params or SamplingParams()
+        if prompt_token_ids is None:
+            if prompt is None:
+                raise ValueError("need prompt or prompt_token_ids")
+            prompt_token_ids = self.tokenizer.encode(prompt)
+        if not prompt_token_ids:
+            prompt_token_ids = [self.spec.bos_token_id]
+        seq = Sequence(request_id, prompt_token_ids, params, arrival_time=time.time())
+        self._seqs[request_id] = seq
+        self.scheduler.add(seq)
+
+    def abort_request(self, request_id: str) -> None:
+        self.scheduler.abort(request_id)
+        self._seqs.pop(request_id, None)
+        self._prefill_done_at.pop(request_id, None)
+
+    def has_unfinished(self) -> bool:
+        return self.scheduler.has_work()
+
+    def num_unfinished(self) -> int:
+        return self.scheduler.num_waiting + self.scheduler.num_running
+
+    # -- stepping --------------------------------------------------------
+
+    def step(self) -> List[RequestOutput]:
+        batch = self.scheduler.schedule()
+        if batch.empty:
+            return []
+        t0 = time.perf_counter()
+        if batch.kind == "prefill":
+            tokens = self.runner.execute_prefill(batch.seqs)
+        else:
+            tokens = self.runner.execute_decode(batch.seqs)
+        token_list = tokens.tolist()
+        now = time.time()
+        self.steps += 1
+        outputs: List[RequestOutput] = []
+        for seq, tok in zip(batch.seqs, token_list):
+            if batch.kind == "prefill":
+                self._prefill_done_at[seq.request_id] = now
+                if seq.first_token_time is None:
+                    seq.first_token_time = now
+            seq.append_token(int(tok))
+            out = RequestOutput(
+                request_id=seq.request_id,
+                new_token_ids=[int(tok)],
+                prompt_tokens=seq.prompt_len,
+                output_tokens=seq.output_len,
+            )
+            reason = self._check_finish(seq)
+            if reason is not None:
+                self.scheduler.finish(seq, reason)
+                out.finished = True
+                out.finish_reason = reason
+                out.text = self._final_text(seq)
+                prefill_at = self._prefill_done_at.pop(seq.request_id, now)
+                out.queue_wait_ms = (prefill_at - seq.arrival_time) * 1000.0
+                out.decode_ms = (now - prefill_at) * 1000.0
+                self._seqs.pop(seq.request_id, None)
+            outputs.append(out)
+        return outputs
+
+    def _check_finish(self, seq: Sequence) -> Optional[str]:
+        params = seq.params
+        last = seq.token_ids[-1]
+        if not params.ignore_eos and last == self.tokenizer.eos_token_id:
+            return "eos"
+        if seq.output_len >= params.max_tokens:
+            return "length"
+        if seq.num_tokens >= self.max_model_len:
+            return "length"
+        if params.stop:
+            tail_ids = seq.token_ids[seq.prompt_len:]
+            text = self.tokenizer.decode(tail_ids)
+            for stop in params.stop:
+                idx = text.find(stop)
+                if idx >= 0:
+                    seq.output_text = text[:idx]
+                    return "stop"
+        return None
+
+    def _final_text(self, seq: Sequence) -> str:
+        if seq.output_text is not None:
+            return seq.output_text
+        out_ids = seq.token_ids[seq.prompt_len:]
+        if out_ids and out_ids[-1] == self.tokenizer.eos_token_id:
+            out_ids = out_ids[:-1]
+        return self.tokenizer.decode(out_ids)
+
+    # -- convenience (tests / bench) -------------------------------------
+
+    def generate_batch(
+        self, prompts: List[str], params: Optional[SamplingParams] = None
+    ) -> List[str]:
+        """Blocking helper: run all prompts to completion, return texts."""
+        results: Dict[str, str] = {}
+        for i, p in enumerate(prompts):
+            self.add_request(f"batch-{i}", prompt=p, params=params)
+        while self.has_unfinished():
+            for out in self.step():
+                if out.finished:
+                    results[out.request_id] = out.text
+        return [results[f"batch-{i}"] for i in range(len(prompts))]

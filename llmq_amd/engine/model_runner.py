@@ -1,0 +1,231 @@
+"""Model runner: batch tensor prep, forward execution, sampling, hipGraphs.
+
+Decode steps are captured as hipGraphs per batch-size bucket (the vLLM
+capability the reference relies on, SURVEY §2.9 "CUDA graph capture of
+decode" → hipGraph here): all decode inputs live in static device buffers;
+a step copies the small int tensors in, replays the graph, and samples from
+the static logits buffer. Python/launch overhead per decode step drops to
+one replay + one sampler call regardless of model depth.
+"""
+
+from __future__ import annotations
+
+import logging
+import time
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from llmq_amd import ops
+from llmq_amd.engine.config import EngineConfig
+from llmq_amd.engine.forward_meta import DecodeMeta, PrefillMeta
+from llmq_amd.engine.kv_cache import KVCache
+from llmq_amd.engine.models.llama import CausalLM
+from llmq_amd.engine.scheduler import ScheduledBatch, Sequence
+
+logger = logging.getLogger(__name__)
+
+
+def _buckets(max_bs: int) -> List[int]:
+    sizes = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256, 384, 512]
+    out = [b for b in sizes if b < max_bs]
+    out.append(max_bs)
+    return out
+
+
+class ModelRunner:
+    def __init__(
+        self,
+        model: CausalLM,
+        kv_cache: KVCache,
+        config: EngineConfig,
+        device: torch.device,
+        max_model_len: int,
+    ):
+        self.model = model
+        self.kv_cache = kv_cache
+        self.config = config
+        self.device = device
+        self.max_model_len = max_model_len
+        self.block_size = kv_cache.block_size
+        self.max_blocks_per_seq = (max_model_len + self.block_size - 1) // self.block_size
+        self.use_graphs = (
+            config.enable_hipgraph
+            and not config.enforce_eager
+            and device.type == "cuda"
+        )
+        self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
+        self._graph_pool = None
+        self.sampling_generator = torch.Generator(device=device if device.type == "cuda" else "cpu")
+        self.sampling_generator.manual_seed(config.seed)
+        if self.use_graphs:
+            self._alloc_static_buffers()
+
+    # -- static buffers / graph capture ---------------------------------
+
+    def _alloc_static_buffers(self) -> None:
+        B = min(self.config.max_num_seqs, self.config.hipgraph_max_batch)
+        dev = self.device
+        self.graph_max_bs = B
+        self.in_ids = torch.zeros(B, dtype=torch.long, device=dev)
+        self.in_pos = torch.zeros(B, dtype=torch.long, device=dev)
+        self.in_slots = torch.zeros(B, dtype=torch.long, device=dev)
+        self.in_block_tables = torch.zeros(
+            B, self.max_blocks_per_seq, dtype=torch.int32, device=dev
+        )
+        self.in_context_lens = torch.ones(B, dtype=torch.int32, device=dev)
+        self.out_hidden = torch.zeros(
+            B, self.model.spec.hidden_size, dtype=self.model.dtype, device=dev
+        )
+        # pinned host staging for fast H2D of the per-step metadata
+        self.h_ids = torch.zeros(B, dtype=torch.long, pin_memory=True)
+        self.h_pos = torch.zeros(B, dtype=torch.long, pin_memory=True)
+        self.h_slots = torch.zeros(B, dtype=torch.long, pin_memory=True)
+        self.h_block_tables = torch.zeros(
+            B, self.max_blocks_per_seq, dtype=torch.int32, pin_memory=True
+        )
+        self.h_context_lens = torch.ones(B, dtype=torch.int32, pin_memory=True)
+
+    def capture_graphs(self) -> None:
+        """Capture the decode forward for each batch-size bucket."""
+        if not self.use_graphs:
+            return
+        t0 = time.perf_counter()
+        torch.cuda.synchronize()
+        for bs in reversed(_buckets(self.graph_max_bs)):  # large→small shares pool
+            meta = DecodeMeta(
+                block_tables=self.in_block_tables[:bs],
+                context_lens=self.in_context_lens[:bs],
+                slot_mapping=self.in_slots[:bs],
+            )
+            # warmup (also materialises workspace allocations)
+            out = self.model.forward(self.in_ids[:bs], self.in_pos[:bs], self.kv_cache, meta)
+            torch.cuda.synchronize()
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph, pool=self._graph_pool):
+                out = self.model.forward(
+                    self.in_ids[:bs], self.in_pos[:bs], self.kv_cache, meta
+                )
+                self.out_hidden[:bs].copy_(out)
+            if self._graph_pool is None:
+                self._graph_pool = graph.pool()
+            self._graphs[bs] = graph
+        torch.cuda.synchronize()
+        logger.info(
+            "captured %d decode hipGraphs (max_bs=%d) in %.1fs",
+            len(self._graphs), self.graph_max_bs, time.perf_counter() - t0,
+        )
+
+    def _bucket_for(self, bs: int) -> Optional[int]:
+        for b in _buckets(self.graph_max_bs):
+            if b >= bs:
+                return b
+        return None
+
+    # -- prefill ---------------------------------------------------------
+
+    @torch.no_grad()
+    def execute_prefill(self, seqs: List[Sequence]) -> torch.Tensor:
+        """Run packed varlen prefill; returns sampled next tokens [B]."""
+        dev = self.device
+        all_ids: List[int] = []
+        all_pos: List[int] = []
+        all_slots: List[int] = []
+        cu = [0]
+        last_idx = []
+        for seq in seqs:
+            n = seq.num_tokens
+            all_ids.extend(seq.token_ids)
+            all_pos.extend(range(n))
+            bt = seq.block_table
+            bs = self.block_size
+            all_slots.extend(bt[p // bs] * bs + (p % bs) for p in range(n))
+            cu.append(cu[-1] + n)
+            last_idx.append(cu[-1] - 1)
+        input_ids = torch.tensor(all_ids, dtype=torch.long, device=dev)
+        positions = torch.tensor(all_pos, dtype=torch.long, device=dev)
+        meta = PrefillMeta(
+            cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+            max_seqlen=max(s.num_tokens for s in seqs),
+            slot_mapping=torch.tensor(all_slots, dtype=torch.long, device=dev),
+        )
+        hidden = self.model.forward(input_ids, positions, self.kv_cache, meta)
+        last_hidden = hidden[torch.tensor(last_idx, dtype=torch.long, device=dev)]
+        logits = self.model.compute_logits(last_hidden)
+        return self._sample(logits, seqs)
+
+    # -- decode ----------------------------------------------------------
+
+    @torch.no_grad()
+    def execute_decode(self, seqs: List[Sequence]) -> torch.Tensor:
+        dev = self.device
+        B = len(seqs)
+        bs = self.block_size
+        ids = [s.token_ids[-1] for s in seqs]
+        pos = [s.num_tokens - 1 for s in seqs]
+        slots = [s.block_table[p // bs] * bs + (p % bs) for s, p in zip(seqs, pos)]
+        ctx = [s.num_tokens for s in seqs]
+
+        bucket = self._bucket_for(B) if self.use_graphs else None
+        if bucket is not None and bucket in self._graphs:
+            return self._decode_with_graph(seqs, bucket, ids, pos, slots, ctx)
+
+        block_tables = torch.zeros(
+            B, max(len(s.block_table) for s in seqs), dtype=torch.int32, device=dev
+        )
+        bt_np = np.zeros(tuple(block_tables.shape), dtype=np.int32)
+        for i, s in enumerate(seqs):
+            bt_np[i, : len(s.block_table)] = s.block_table
+        block_tables.copy_(torch.from_numpy(bt_np))
+        meta = DecodeMeta(
+            block_tables=block_tables,
+            context_lens=torch.tensor(ctx, dtype=torch.int32, device=dev),
+            slot_mapping=torch.tensor(slots, dtype=torch.long, device=dev),
+        )
+        input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
+        positions = torch.tensor(pos, dtype=torch.long, device=dev)
+        hidden = self.model.forward(input_ids, positions, self.kv_cache, meta)
+        logits = self.model.compute_logits(hidden)
+        return self._sample(logits, seqs)
+
+    def _decode_with_graph(
+        self, seqs: List[Sequence], bucket: int, ids, pos, slots, ctx
+    ) -> torch.Tensor:
+        B = len(seqs)
+        # Stage metadata through pinned host buffers, one async copy each.
+        self.h_ids[:B] = torch.tensor(ids, dtype=torch.long)
+        self.h_pos[:B] = torch.tensor(pos, dtype=torch.long)
+        self.h_slots[:B] = torch.tensor(slots, dtype=torch.long)
+        self.h_context_lens[:B] = torch.tensor(ctx, dtype=torch.int32)
+        self.h_context_lens[B:bucket] = 1
+        bt = self.h_block_tables
+        bt_np = bt.numpy()
+        for i, s in enumerate(seqs):
+            n = len(s.block_table)
+            bt_np[i, :n] = s.block_table
+            bt_np[i, n:] = 0
+        if B < bucket:
+            bt_np[B:bucket] = 0
+            # pad rows: context_len 1 pointing at block 0 (defined garbage,
+            # their logits are never read)
+            self.h_slots[B:bucket] = 0
+            self.h_ids[B:bucket] = 0
+            self.h_pos[B:bucket] = 0
+        self.in_ids[:bucket].copy_(self.h_ids[:bucket], non_blocking=True)
+        self.in_pos[:bucket].copy_(self.h_pos[:bucket], non_blocking=True)
+        self.in_slots[:bucket].copy_(self.h_slots[:bucket], non_blocking=True)
+        self.in_context_lens[:bucket].copy_(self.h_context_lens[:bucket], non_blocking=True)
+        self.in_block_tables[:bucket].copy_(self.h_block_tables[:bucket], non_blocking=True)
+        self._graphs[bucket].replay()
+        logits = self.model.compute_logits(self.out_hidden[:B])
+        return self._sample(logits, seqs)
+
+    # -- sampling --------------------------------------------------------
+
+    def _sample(self, logits: torch.Tensor, seqs: List[Sequence]) -> torch.Tensor:
+        dev = logits.device
+        temps = torch.tensor([s.params.temperature for s in seqs], dtype=torch.float32, device=dev)
+        tps = torch.tensor([s.params.top_p for s in seqs], dtype=torch.float32, device=dev)
+        tks = torch.tensor([s.params.top_k for s in seqs], dtype=torch.int64, device=dev)
+        return ops.sample_tokens(logits, temps, tps, tks, self.sampling_generator)

@@ -1,0 +1,251 @@
+"""Decoder-only transformer covering the Llama / Qwen2 / Gemma-2 families.
+
+One functional implementation driven by ModelSpec switches (SURVEY §2.9:
+the model set the reference serves through vLLM). Design notes, MI355X-first:
+
+- Weights are plain tensors (no nn.Module graph): qkv packed into ONE GEMM
+  [q+2kv, hidden], gate+up packed into ONE GEMM [2*inter, hidden] — fewer,
+  larger hipBLASLt GEMMs (xGMI-budgeted TP shards stay big).
+- Hot elementwise/normalisation ops go through llmq_amd.ops → hand-written
+  CDNA4 kernels on GPU (fused residual+RMSNorm, fused RoPE q‖k, fused
+  SiLU·mul / GeGLU, paged attention, KV scatter).
+- Tensor parallelism: column-shard qkv & gate_up, row-shard o & down with a
+  single RCCL all-reduce after each of the two row GEMMs per layer (the
+  standard 2-allreduce/layer Megatron split; over 7×153 GB/s xGMI links the
+  payloads at decode batch sizes are latency-bound, so fewer+larger
+  collectives win — SURVEY §2.9 TP note).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional, Union
+
+import torch
+import torch.nn.functional as F
+
+from llmq_amd import ops
+from llmq_amd.engine.forward_meta import DecodeMeta, PrefillMeta
+from llmq_amd.engine.kv_cache import KVCache
+from llmq_amd.engine.model_specs import ModelSpec
+from llmq_amd.parallel import get_tp_group
+
+
+class LayerWeights:
+    __slots__ = (
+        "input_norm", "qkv", "qkv_bias", "o", "post_attn_norm",
+        "pre_mlp_norm", "gate_up", "down", "post_mlp_norm",
+    )
+
+    def __init__(self):
+        self.input_norm = None
+        self.qkv = None
+        self.qkv_bias = None
+        self.o = None
+        self.post_attn_norm = None
+        self.pre_mlp_norm = None
+        self.gate_up = None
+        self.down = None
+        self.post_mlp_norm = None
+
+
+class CausalLM:
+    def __init__(
+        self,
+        spec: ModelSpec,
+        device: torch.device,
+        dtype: torch.dtype,
+        tp_rank: int = 0,
+        tp_size: int = 1,
+    ):
+        self.spec = spec
+        self.device = device
+        self.dtype = dtype
+        self.tp_rank = tp_rank
+        self.tp_size = tp_size
+        if spec.num_heads % tp_size or spec.num_kv_heads % max(
+            1, min(tp_size, spec.num_kv_heads)
+        ):
+            raise ValueError(
+                f"num_heads={spec.num_heads} not divisible by tp_size={tp_size}"
+            )
+        self.heads = spec.num_heads // tp_size
+        # KV heads replicate when tp_size > num_kv_heads is not supported; shard.
+        if spec.num_kv_heads % tp_size:
+            raise ValueError(
+                f"num_kv_heads={spec.num_kv_heads} not divisible by tp_size={tp_size}"
+            )
+        self.kv_heads = spec.num_kv_heads // tp_size
+        self.q_size = self.heads * spec.head_dim
+        self.kv_size = self.kv_heads * spec.head_dim
+        self.inter = spec.intermediate_size // tp_size
+        if spec.intermediate_size % tp_size:
+            raise ValueError("intermediate_size not divisible by tp_size")
+        # Gemma norms use the (1 + w) convention.
+        self.norm_offset = 1.0 if spec.family == "gemma2" else 0.0
+
+        self.embedding: Optional[torch.Tensor] = None
+        self.layers: List[LayerWeights] = []
+        self.final_norm: Optional[torch.Tensor] = None
+        self.lm_head: Optional[torch.Tensor] = None
+        self.rope_cache = ops.build_rope_cache(
+            spec.max_position_embeddings, spec.head_dim, spec.rope_theta, device
+        )
+        self._alloc()
+
+    # -- weights ---------------------------------------------------------
+
+    def _empty(self, *shape) -> torch.Tensor:
+        return torch.empty(*shape, device=self.device, dtype=self.dtype)
+
+    def _alloc(self) -> None:
+        s = self.spec
+        self.embedding = self._empty(s.vocab_size, s.hidden_size)
+        for _ in range(s.num_layers):
+            lw = LayerWeights()
+            lw.input_norm = self._empty(s.hidden_size)
+            lw.qkv = self._empty(self.q_size + 2 * self.kv_size, s.hidden_size)
+            if s.qkv_bias:
+                lw.qkv_bias = self._empty(self.q_size + 2 * self.kv_size)
+            lw.o = self._empty(s.hidden_size, self.q_size)
+            lw.pre_mlp_norm = self._empty(s.hidden_size)
+            lw.gate_up = self._empty(2 * self.inter, s.hidden_size)
+            lw.down = self._empty(s.hidden_size, self.inter)
+            if s.post_norms:
+                lw.post_attn_norm = self._empty(s.hidden_size)
+                lw.post_mlp_norm = self._empty(s.hidden_size)
+            self.layers.append(lw)
+        self.final_norm = self._empty(s.hidden_size)
+        self.lm_head = self.embedding if s.tied_embeddings else self._empty(
+            s.vocab_size, s.hidden_size
+        )
+
+    @torch.no_grad()
+    def random_init(self, seed: int = 0) -> None:
+        """Random weights for synthetic benchmarking (no network for real
+        checkpoints here). Scaled so activations stay finite in bf16."""
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+
+        def fill(t: torch.Tensor, std: float) -> None:
+            cpu = torch.randn(t.shape, generator=gen, dtype=torch.float32) * std
+            t.copy_(cpu.to(t.dtype))
+
+        h = self.spec.hidden_size
+        std = 0.02
+        out_std = std / math.sqrt(2 * self.spec.num_layers)
+        fill(self.embedding, std)
+        for lw in self.layers:
+            lw.input_norm.fill_(1.0 - self.norm_offset)
+            fill(lw.qkv, std)
+            if lw.qkv_bias is not None:
+                lw.qkv_bias.zero_()
+            fill(lw.o, out_std)
+            lw.pre_mlp_norm.fill_(1.0 - self.norm_offset)
+            fill(lw.gate_up, std)
+            fill(lw.down, out_std)
+            if lw.post_attn_norm is not None:
+                lw.post_attn_norm.fill_(1.0 - self.norm_offset)
+                lw.post_mlp_norm.fill_(1.0 - self.norm_offset)
+        self.final_norm.fill_(1.0 - self.norm_offset)
+        if not self.spec.tied_embeddings:
+            fill(self.lm_head, std)
+
+    def named_tensors(self) -> Dict[str, torch.Tensor]:
+        out = {"embedding": self.embedding, "final_norm": self.final_norm}
+        if not self.spec.tied_embeddings:
+            out["lm_head"] = self.lm_head
+        for i, lw in enumerate(self.layers):
+            p = f"layers.{i}."
+            out[p + "input_norm"] = lw.input_norm
+            out[p + "qkv"] = lw.qkv
+            if lw.qkv_bias is not None:
+                out[p + "qkv_bias"] = lw.qkv_bias
+            out[p + "o"] = lw.o
+            out[p + "pre_mlp_norm"] = lw.pre_mlp_norm
+            out[p + "gate_up"] = lw.gate_up
+            out[p + "down"] = lw.down
+            if lw.post_attn_norm is not None:
+                out[p + "post_attn_norm"] = lw.post_attn_norm
+                out[p + "post_mlp_norm"] = lw.post_mlp_norm
+        return out
+
+    def weight_bytes(self) -> int:
+        return sum(t.numel() * t.element_size() for t in self.named_tensors().values())
+
+    # -- forward ---------------------------------------------------------
+
+    def _norm(self, x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+        return ops.rmsnorm(x, w, self.spec.rms_eps, self.norm_offset)
+
+    @torch.no_grad()
+    def forward(
+        self,
+        input_ids: torch.Tensor,  # [T]
+        positions: torch.Tensor,  # [T]
+        kv_cache: KVCache,
+        meta: Union[PrefillMeta, DecodeMeta],
+    ) -> torch.Tensor:
+        s = self.spec
+        tp = get_tp_group() if self.tp_size > 1 else None
+        x = F.embedding(input_ids, self.embedding)
+        if s.embedding_scale:
+            x = x * math.sqrt(s.hidden_size)
+            x = x.to(self.dtype)
+        residual: Optional[torch.Tensor] = None
+        for i, lw in enumerate(self.layers):
+            # ---- attention block
+            if residual is None:
+                residual = x
+                h = self._norm(x, lw.input_norm)
+            else:
+                h, residual = ops.fused_add_rmsnorm(
+                    x, residual, lw.input_norm, s.rms_eps, self.norm_offset
+                )
+            qkv = F.linear(h, lw.qkv, lw.qkv_bias)
+            q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
+            T = q.shape[0]
+            q = q.view(T, self.heads, s.head_dim)
+            k = k.view(T, self.kv_heads, s.head_dim)
+            v = v.view(T, self.kv_heads, s.head_dim)
+            ops.rope_inplace(q, k, positions, self.rope_cache)
+            ops.reshape_and_cache(k, v, kv_cache.k[i], kv_cache.v[i], meta.slot_mapping)
+            window = s.sliding_window if s.layer_uses_sliding_window(i) else 0
+            if meta.is_prefill:
+                attn = ops.varlen_prefill_attention(
+                    q, k, v, meta.cu_seqlens, meta.max_seqlen, s.scale,
+                    s.attn_softcap, window,
+                )
+            else:
+                attn = ops.paged_decode_attention(
+                    q, kv_cache.k[i], kv_cache.v[i], meta.block_tables,
+                    meta.context_lens, s.scale, s.attn_softcap, window,
+                )
+            attn_out = F.linear(attn.reshape(T, self.q_size), lw.o)
+            if tp is not None:
+                attn_out = tp.all_reduce(attn_out)
+            if s.post_norms:
+                attn_out = self._norm(attn_out, lw.post_attn_norm)
+            # ---- MLP block
+            h, residual = ops.fused_add_rmsnorm(
+                attn_out, residual, lw.pre_mlp_norm, s.rms_eps, self.norm_offset
+            )
+            gate_up = F.linear(h, lw.gate_up)
+            act = ops.gelu_tanh_and_mul(gate_up) if s.gelu else ops.silu_and_mul(gate_up)
+            mlp_out = F.linear(act, lw.down)
+            if tp is not None:
+                mlp_out = tp.all_reduce(mlp_out)
+            if s.post_norms:
+                mlp_out = self._norm(mlp_out, lw.post_mlp_norm)
+            x = mlp_out
+        # final residual add + norm
+        final = (residual.float() + x.float()).to(self.dtype)
+        return self._norm(final, self.final_norm)
+
+    @torch.no_grad()
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        """hidden [N, hidden] → logits [N, vocab] (fp32)."""
+        logits = F.linear(hidden, self.lm_head).float()
+        cap = self.spec.final_softcap
+        if cap and cap > 0:
+            logits = torch.tanh(logits / cap) * cap
+        return logits

@@ -1,0 +1,206 @@
+"""Continuous-batching scheduler.
+
+The in-tree equivalent of the capability the reference buys from vLLM
+(admit up to max_num_seqs sequences per step, prefill/decode
+interleave, paged-KV admission control). Policy (vLLM-v0-style):
+
+- FCFS waiting queue; a step is either one PREFILL batch (packed varlen,
+  bounded by max_prefill_tokens) or one DECODE batch (all running seqs).
+- Prefill has priority: new sequences join as soon as slots + KV blocks
+  allow, keeping the batch full (the broker-side prefetch ≫ batch trick,
+  SURVEY §3.1, keeps this queue fed).
+- Preemption by recompute: if a decode step cannot allocate the next KV
+  block, the youngest running sequence is evicted, its blocks freed, and it
+  re-enters the waiting queue with its generated tokens as prompt.
+"""
+
+from __future__ import annotations
+
+import logging
+from collections import deque
+from dataclasses import dataclass, field
+from enum import Enum
+from typing import Deque, List, Optional
+
+from llmq_amd.engine.kv_cache import BlockAllocator, blocks_needed
+from llmq_amd.engine.sampling_params import SamplingParams
+
+logger = logging.getLogger(__name__)
+
+
+class SeqStatus(Enum):
+    WAITING = "waiting"
+    RUNNING = "running"
+    FINISHED = "finished"
+
+
+class Sequence:
+    __slots__ = (
+        "request_id", "token_ids", "prompt_len", "params", "status",
+        "block_table", "arrival_time", "first_token_time", "finish_reason",
+        "output_text", "num_preemptions",
+    )
+
+    def __init__(self, request_id: str, prompt_token_ids: List[int], params: SamplingParams,
+                 arrival_time: float = 0.0):
+        self.request_id = request_id
+        self.token_ids: List[int] = list(prompt_token_ids)
+        self.prompt_len = len(prompt_token_ids)
+        self.params = params
+        self.status = SeqStatus.WAITING
+        self.block_table: List[int] = []
+        self.arrival_time = arrival_time
+        self.first_token_time: Optional[float] = None
+        self.finish_reason: Optional[str] = None
+        self.output_text = None  # set only when a stop string truncates
+        self.num_preemptions = 0
+
+    @property
+    def num_tokens(self) -> int:
+        return len(self.token_ids)
+
+    @property
+    def output_len(self) -> int:
+        return len(self.token_ids) - self.prompt_len
+
+    def append_token(self, token_id: int) -> None:
+        self.token_ids.append(token_id)
+
+
+@dataclass
+class ScheduledBatch:
+    kind: str  # "prefill" | "decode"
+    seqs: List[Sequence] = field(default_factory=list)
+
+    @property
+    def empty(self) -> bool:
+        return not self.seqs
+
+
+class Scheduler:
+    def __init__(
+        self,
+        allocator: BlockAllocator,
+        block_size: int,
+        max_num_seqs: int,
+        max_prefill_tokens: int,
+        max_model_len: int,
+    ):
+        self.allocator = allocator
+        self.block_size = block_size
+        self.max_num_seqs = max_num_seqs
+        self.max_prefill_tokens = max_prefill_tokens
+        self.max_model_len = max_model_len
+        self.waiting: Deque[Sequence] = deque()
+        self.running: List[Sequence] = []
+
+    # -- public ----------------------------------------------------------
+
+    @property
+    def num_waiting(self) -> int:
+        return len(self.waiting)
+
+    @property
+    def num_running(self) -> int:
+        return len(self.running)
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    def add(self, seq: Sequence) -> None:
+        if seq.num_tokens > self.max_model_len:
+            # Trim oversized prompts from the left (keep the recent window).
+            seq.token_ids = seq.token_ids[-self.max_model_len + 1 :]
+            seq.prompt_len = len(seq.token_ids)
+        self.waiting.append(seq)
+
+    def abort(self, request_id: str) -> bool:
+        for i, seq in enumerate(self.running):
+            if seq.request_id == request_id:
+                self._release(seq)
+                del self.running[i]
+                return True
+        for i, seq in enumerate(self.waiting):
+            if seq.request_id == request_id:
+                del self.waiting[i]
+                return True
+        return False
+
+    def finish(self, seq: Sequence, reason: str) -> None:
+        seq.status = SeqStatus.FINISHED
+        seq.finish_reason = reason
+        self._release(seq)
+        try:
+            self.running.remove(seq)
+        except ValueError:
+            pass
+
+    def schedule(self) -> ScheduledBatch:
+        prefill = self._schedule_prefill()
+        if not prefill.empty:
+            return prefill
+        return self._schedule_decode()
+
+    # -- internals -------------------------------------------------------
+
+    def _release(self, seq: Sequence) -> None:
+        if seq.block_table:
+            self.allocator.free(seq.block_table)
+            seq.block_table = []
+
+    def _schedule_prefill(self) -> ScheduledBatch:
+        batch = ScheduledBatch("prefill")
+        budget = self.max_prefill_tokens
+        while self.waiting and len(self.running) + len(batch.seqs) < self.max_num_seqs:
+            seq = self.waiting[0]
+            n = seq.num_tokens
+            if batch.seqs and n > budget:
+                break
+            need = blocks_needed(n, self.block_size)
+            blocks = self.allocator.allocate(need)
+            if blocks is None:
+                break
+            seq.block_table = blocks
+            seq.status = SeqStatus.RUNNING
+            self.waiting.popleft()
+            batch.seqs.append(seq)
+            budget -= n
+        if batch.seqs:
+            self.running.extend(batch.seqs)
+        return batch
+
+    def _schedule_decode(self) -> ScheduledBatch:
+        batch = ScheduledBatch("decode")
+        if not self.running:
+            return batch
+        # Ensure every running seq has a KV slot for its next position;
+        # preempt from the back (youngest) on allocation failure.
+        i = 0
+        while i < len(self.running):
+            seq = self.running[i]
+            need = blocks_needed(seq.num_tokens, self.block_size) - len(seq.block_table)
+            if need > 0:
+                blocks = self.allocator.allocate(need)
+                if blocks is None:
+                    victim = self.running[-1]
+                    if victim is seq and len(self.running) == 1:
+                        logger.error(
+                            "seq %s cannot fit in KV cache even alone; finishing as length",
+                            seq.request_id,
+                        )
+                        self.finish(seq, "length")
+                        continue
+                    self._preempt(victim)
+                    continue  # retry same index
+                seq.block_table.extend(blocks)
+            i += 1
+        batch.seqs = list(self.running)
+        return batch
+
+    def _preempt(self, seq: Sequence) -> None:
+        logger.warning("preempting seq %s (recompute)", seq.request_id)
+        self._release(seq)
+        seq.status = SeqStatus.WAITING
+        seq.num_preemptions += 1
+        self.running.remove(seq)
+        self.waiting.appendleft(seq)
