@@ -1,0 +1,62 @@
+"""Build the CDNA4 HIP extension in-tree.
+
+Produces llmq_amd/ops/_hip_ops.so (tracked location, gitignored) so the
+built artifact travels to GPU boxes with the repo snapshot. Cross-compiles
+for gfx950 without a GPU (PYTORCH_ROCM_ARCH pins the arch).
+
+Usage: python -m llmq_amd.ops.build [--verbose] [--force]
+"""
+
+from __future__ import annotations
+
+import os
+import shutil
+import sys
+from pathlib import Path
+
+HERE = Path(__file__).parent
+SRC = HERE / "hip" / "ops.hip"
+OUT = HERE / "_hip_ops.so"
+BUILD_DIR = HERE / "build"
+
+
+def _sources_mtime() -> float:
+    return max(p.stat().st_mtime for p in (HERE / "hip").iterdir() if p.is_file())
+
+
+def build(verbose: bool = False, force: bool = False) -> Path:
+    if OUT.is_file() and not force and OUT.stat().st_mtime >= _sources_mtime():
+        return OUT
+    # torch.utils.cpp_extension tracks only the listed sources (ops.hip), not
+    # the files it #includes — touch the umbrella TU so edits to any kernel
+    # file trigger a rebuild.
+    SRC.touch()
+    if force:
+        shutil.rmtree(BUILD_DIR, ignore_errors=True)
+    os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+    os.environ.setdefault("MAX_JOBS", "8")
+    from torch.utils.cpp_extension import load
+
+    BUILD_DIR.mkdir(exist_ok=True)
+    load(
+        name="llmq_amd_hip_ops",
+        sources=[str(SRC)],
+        extra_cflags=["-O3", "-std=c++20"],
+        extra_cuda_cflags=["-O3", "-std=c++20"],
+        build_directory=str(BUILD_DIR),
+        is_python_module=False,
+        verbose=verbose,
+    )
+    built = BUILD_DIR / "llmq_amd_hip_ops.so"
+    if not built.is_file():
+        candidates = list(BUILD_DIR.glob("*.so"))
+        if not candidates:
+            raise RuntimeError(f"build produced no .so under {BUILD_DIR}")
+        built = candidates[0]
+    shutil.copy2(built, OUT)
+    return OUT
+
+
+if __name__ == "__main__":
+    out = build(verbose="--verbose" in sys.argv, force="--force" in sys.argv)
+    print(f"built {out}")

@@ -1,0 +1,270 @@
+// Torch bindings for the llmq-amd CDNA4 kernels.
+// Registered under torch.ops.llmq_amd.* ; loaded via torch.ops.load_library.
+
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+#include <torch/library.h>
+
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+
+// kernel declarations (defined in the .hip TUs)
+template <typename T>
+__global__ void rmsnorm_kernel(T*, const T*, const T*, int, float, float);
+template <typename T>
+__global__ void fused_add_rmsnorm_kernel(T*, T*, const T*, int, float, float);
+template <typename T, bool GELU>
+__global__ void act_and_mul_kernel(T*, const T*, long, int);
+template <typename T>
+__global__ void rope_kernel(T*, T*, const long*, const float*, int, int, int,
+                            long, long);
+template <typename T>
+__global__ void reshape_and_cache_kernel(const T*, const T*, T*, T*,
+                                         const long*, int, int, int, long, long);
+template <typename T, int HEAD_DIM>
+__global__ void paged_decode_attention_kernel(T*, const T*, const T*, const T*,
+                                              const int*, const int*, int, int,
+                                              int, int, float, float, int, long,
+                                              long);
+template <typename T, int HEAD_DIM>
+__global__ void varlen_prefill_attention_kernel(T*, const T*, const T*, const T*,
+                                                const int*, int, int, float,
+                                                float, int, long, long, long,
+                                                long);
+
+namespace {
+
+#define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
+#define CHECK_LASTDIM(x) \
+  TORCH_CHECK(x.stride(-1) == 1, #x " innermost dim must be contiguous")
+
+hipStream_t stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+template <typename F>
+void dispatch_dtype(const at::Tensor& t, const char* name, F&& f) {
+  switch (t.scalar_type()) {
+    case at::kBFloat16:
+      f.template operator()<__hip_bfloat16>();
+      break;
+    case at::kHalf:
+      f.template operator()<_Float16>();
+      break;
+    case at::kFloat:
+      f.template operator()<float>();
+      break;
+    default:
+      TORCH_CHECK(false, name, ": unsupported dtype ", t.scalar_type());
+  }
+}
+
+// ---------------------------------------------------------------- norms --
+
+void rmsnorm(at::Tensor out, at::Tensor in, at::Tensor weight, double eps,
+             double offset) {
+  CHECK_GPU(in);
+  CHECK_LASTDIM(in);
+  const int hidden = in.size(-1);
+  const long rows = in.numel() / hidden;
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  dispatch_dtype(in, "rmsnorm", [&]<typename T>() {
+    hipLaunchKernelGGL(rmsnorm_kernel<T>, dim3(rows), dim3(256), 0, stream(),
+                       reinterpret_cast<T*>(out.data_ptr()),
+                       reinterpret_cast<const T*>(in.data_ptr()),
+                       reinterpret_cast<const T*>(weight.data_ptr()), hidden,
+                       (float)eps, (float)offset);
+  });
+}
+
+void fused_add_rmsnorm(at::Tensor x, at::Tensor residual, at::Tensor weight,
+                       double eps, double offset) {
+  CHECK_GPU(x);
+  CHECK_LASTDIM(x);
+  const int hidden = x.size(-1);
+  const long rows = x.numel() / hidden;
+  dispatch_dtype(x, "fused_add_rmsnorm", [&]<typename T>() {
+    hipLaunchKernelGGL(fused_add_rmsnorm_kernel<T>, dim3(rows), dim3(256), 0,
+                       stream(), reinterpret_cast<T*>(x.data_ptr()),
+                       reinterpret_cast<T*>(residual.data_ptr()),
+                       reinterpret_cast<const T*>(weight.data_ptr()), hidden,
+                       (float)eps, (float)offset);
+  });
+}
+
+// ----------------------------------------------------------- activations --
+
+template <bool GELU>
+void act_and_mul(at::Tensor out, at::Tensor in) {
+  CHECK_GPU(in);
+  CHECK_LASTDIM(in);
+  const int d = out.size(-1);
+  const long rows = out.numel() / d;
+  TORCH_CHECK(in.size(-1) == 2 * d, "in must be [..., 2*d]");
+  const long chunks = rows * (d / 8);
+  const int grid = std::min<long>((chunks + 255) / 256, 2048);
+  dispatch_dtype(in, "act_and_mul", [&]<typename T>() {
+    hipLaunchKernelGGL((act_and_mul_kernel<T, GELU>), dim3(grid), dim3(256), 0,
+                       stream(), reinterpret_cast<T*>(out.data_ptr()),
+                       reinterpret_cast<const T*>(in.data_ptr()), rows, d);
+  });
+}
+
+void silu_and_mul(at::Tensor out, at::Tensor in) { act_and_mul<false>(out, in); }
+void gelu_tanh_and_mul(at::Tensor out, at::Tensor in) { act_and_mul<true>(out, in); }
+
+// ----------------------------------------------------------------- rope --
+
+void rope_inplace(at::Tensor q, at::Tensor k, at::Tensor positions,
+                  at::Tensor cos_sin) {
+  CHECK_GPU(q);
+  CHECK_LASTDIM(q);
+  CHECK_LASTDIM(k);
+  TORCH_CHECK(q.dim() == 3 && k.dim() == 3, "q/k must be [T, H, D]");
+  TORCH_CHECK(cos_sin.scalar_type() == at::kFloat, "cos_sin must be f32");
+  const int T_ = q.size(0);
+  if (T_ == 0) return;
+  const int hq = q.size(1), hk = k.size(1), d = q.size(2);
+  TORCH_CHECK(q.stride(1) == d && k.stride(1) == d, "head dim must be packed");
+  dispatch_dtype(q, "rope", [&]<typename T>() {
+    hipLaunchKernelGGL(rope_kernel<T>, dim3(T_), dim3(256), 0, stream(),
+                       reinterpret_cast<T*>(q.data_ptr()),
+                       reinterpret_cast<T*>(k.data_ptr()),
+                       positions.data_ptr<long>(), cos_sin.data_ptr<float>(),
+                       hq, hk, d, q.stride(0), k.stride(0));
+  });
+}
+
+// ------------------------------------------------------------- KV cache --
+
+void reshape_and_cache(at::Tensor key, at::Tensor value, at::Tensor k_cache,
+                       at::Tensor v_cache, at::Tensor slot_mapping) {
+  CHECK_GPU(key);
+  CHECK_LASTDIM(key);
+  const int T_ = key.size(0);
+  if (T_ == 0) return;
+  const int kvh = key.size(1), d = key.size(2);
+  const int bs = k_cache.size(2);
+  TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
+  TORCH_CHECK(slot_mapping.scalar_type() == at::kLong);
+  dispatch_dtype(key, "reshape_and_cache", [&]<typename T>() {
+    hipLaunchKernelGGL(reshape_and_cache_kernel<T>, dim3(T_), dim3(128), 0,
+                       stream(), reinterpret_cast<const T*>(key.data_ptr()),
+                       reinterpret_cast<const T*>(value.data_ptr()),
+                       reinterpret_cast<T*>(k_cache.data_ptr()),
+                       reinterpret_cast<T*>(v_cache.data_ptr()),
+                       slot_mapping.data_ptr<long>(), kvh, d, bs,
+                       key.stride(0), value.stride(0));
+  });
+}
+
+// ------------------------------------------------------------ attention --
+
+template <typename T>
+void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
+                   const at::Tensor& vc, const at::Tensor& bt,
+                   const at::Tensor& cl, double scale, double softcap,
+                   long window) {
+  const int B = q.size(0), H = q.size(1), D = q.size(2);
+  const int KVH = kc.size(1);
+  const int bs = kc.size(2);
+  const int max_blocks = bt.size(1);
+  const int G = H / KVH;
+  TORCH_CHECK(H % KVH == 0 && G <= 8, "GQA group must be <= 8 (got ", G, ")");
+  TORCH_CHECK(bs == 16 || bs == 32, "block_size must be 16 or 32");
+  dim3 grid(B, KVH);
+  const int lds = (G * D + 64 * G) * sizeof(float);
+  auto l = [&]<int HD>() {
+    hipLaunchKernelGGL((paged_decode_attention_kernel<T, HD>), grid, dim3(256),
+                       lds, stream(), reinterpret_cast<T*>(out.data_ptr()),
+                       reinterpret_cast<const T*>(q.data_ptr()),
+                       reinterpret_cast<const T*>(kc.data_ptr()),
+                       reinterpret_cast<const T*>(vc.data_ptr()),
+                       bt.data_ptr<int>(), cl.data_ptr<int>(), H, KVH, bs,
+                       max_blocks, (float)scale, (float)softcap, (int)window,
+                       q.stride(0), out.stride(0));
+  };
+  switch (D) {
+    case 64: l.template operator()<64>(); break;
+    case 128: l.template operator()<128>(); break;
+    case 256: l.template operator()<256>(); break;
+    default: TORCH_CHECK(false, "unsupported head_dim ", D);
+  }
+}
+
+void paged_decode_attention(at::Tensor out, at::Tensor q, at::Tensor k_cache,
+                            at::Tensor v_cache, at::Tensor block_tables,
+                            at::Tensor context_lens, double scale,
+                            double softcap, long window) {
+  CHECK_GPU(q);
+  CHECK_LASTDIM(q);
+  TORCH_CHECK(block_tables.scalar_type() == at::kInt);
+  TORCH_CHECK(context_lens.scalar_type() == at::kInt);
+  dispatch_dtype(q, "paged_decode_attention", [&]<typename T>() {
+    launch_decode<T>(out, q, k_cache, v_cache, block_tables, context_lens,
+                     scale, softcap, window);
+  });
+}
+
+template <typename T>
+void launch_prefill(at::Tensor& out, const at::Tensor& q, const at::Tensor& k,
+                    const at::Tensor& v, const at::Tensor& cu, int B,
+                    double scale, double softcap, long window) {
+  const int H = q.size(1), D = q.size(2);
+  const int KVH = k.size(1);
+  dim3 grid(B, H);
+  auto l = [&]<int HD>() {
+    hipLaunchKernelGGL((varlen_prefill_attention_kernel<T, HD>), grid,
+                       dim3(256), 0, stream(),
+                       reinterpret_cast<T*>(out.data_ptr()),
+                       reinterpret_cast<const T*>(q.data_ptr()),
+                       reinterpret_cast<const T*>(k.data_ptr()),
+                       reinterpret_cast<const T*>(v.data_ptr()),
+                       cu.data_ptr<int>(), H, KVH, (float)scale, (float)softcap,
+                       (int)window, q.stride(0), k.stride(0), v.stride(0),
+                       out.stride(0));
+  };
+  switch (D) {
+    case 64: l.template operator()<64>(); break;
+    case 128: l.template operator()<128>(); break;
+    case 256: l.template operator()<256>(); break;
+    default: TORCH_CHECK(false, "unsupported head_dim ", D);
+  }
+}
+
+void varlen_prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,
+                              at::Tensor v, at::Tensor cu_seqlens,
+                              long max_seqlen, double scale, double softcap,
+                              long window) {
+  CHECK_GPU(q);
+  CHECK_LASTDIM(q);
+  TORCH_CHECK(cu_seqlens.scalar_type() == at::kInt);
+  const int B = cu_seqlens.size(0) - 1;
+  (void)max_seqlen;
+  dispatch_dtype(q, "varlen_prefill_attention", [&]<typename T>() {
+    launch_prefill<T>(out, q, k, v, cu_seqlens, B, scale, softcap, window);
+  });
+}
+
+}  // namespace
+
+TORCH_LIBRARY(llmq_amd, m) {
+  m.def("rmsnorm(Tensor(a!) out, Tensor input, Tensor weight, float eps, float offset) -> ()");
+  m.def("fused_add_rmsnorm(Tensor(a!) x, Tensor(b!) residual, Tensor weight, float eps, float offset) -> ()");
+  m.def("silu_and_mul(Tensor(a!) out, Tensor input) -> ()");
+  m.def("gelu_tanh_and_mul(Tensor(a!) out, Tensor input) -> ()");
+  m.def("rope_inplace(Tensor(a!) q, Tensor(b!) k, Tensor positions, Tensor cos_sin) -> ()");
+  m.def("reshape_and_cache(Tensor key, Tensor value, Tensor(a!) k_cache, Tensor(b!) v_cache, Tensor slot_mapping) -> ()");
+  m.def("paged_decode_attention(Tensor(a!) out, Tensor q, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor context_lens, float scale, float softcap, int window) -> ()");
+  m.def("varlen_prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, int max_seqlen, float scale, float softcap, int window) -> ()");
+}
+
+TORCH_LIBRARY_IMPL(llmq_amd, CUDA, m) {
+  m.impl("rmsnorm", &rmsnorm);
+  m.impl("fused_add_rmsnorm", &fused_add_rmsnorm);
+  m.impl("silu_and_mul", &silu_and_mul);
+  m.impl("gelu_tanh_and_mul", &gelu_tanh_and_mul);
+  m.impl("rope_inplace", &rope_inplace);
+  m.impl("reshape_and_cache", &reshape_and_cache);
+  m.impl("paged_decode_attention", &paged_decode_attention);
+  m.impl("varlen_prefill_attention", &varlen_prefill_attention);
+}

@@ -1,0 +1,6 @@
+// Umbrella translation unit: all kernels + torch bindings compiled together
+// (template instantiation happens at the launch sites in bindings.cpp).
+#include "elementwise.hip"
+#include "kvcache.hip"
+#include "attention.hip"
+#include "bindings.cpp"
