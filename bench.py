@@ -1,0 +1,188 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark (driver contract).
+
+Measures the BASELINE.json headline: output tokens/sec for Tower-Plus-9B
+(Gemma-2-9B architecture, random-init bf16, synthetic prompts) served by
+N data-parallel engine workers, one rank per MI355X over RCCL.
+
+A "step" is one continuous-batching decode iteration of the full resident
+batch (the engine's serving step: scheduler → hipGraph decode forward →
+sampling → bookkeeping). The engine is pre-filled with --batch sequences of
+--prompt-len synthetic tokens; after --warmup untimed steps, EXACTLY
+--steps steps are timed between barrier+synchronize fences; the slowest
+rank's wall time is used. value = N_ranks × batch × steps ÷ max_elapsed.
+
+Launch (multi-GPU, by the driver):
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=64)
+    parser.add_argument("--warmup", type=int, default=16)
+    parser.add_argument("--model", default="tower-plus-9b")
+    parser.add_argument("--batch", type=int, default=256,
+                        help="resident sequences per GPU (max_num_seqs)")
+    parser.add_argument("--prompt-len", type=int, default=1024)
+    parser.add_argument("--max-model-len", type=int, default=4096)
+    parser.add_argument("--temperature", type=float, default=0.7)
+    parser.add_argument("--eager", action="store_true", help="disable hipGraphs")
+    parser.add_argument("--profile-steps", type=int, default=0,
+                        help="extra untimed steps after the timed region (rocprof)")
+    args = parser.parse_args()
+
+    import torch
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world > 1
+
+    if distributed:
+        import torch.distributed as dist
+
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend="nccl")
+
+    use_gpu = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if use_gpu else "cpu"
+
+    from llmq_amd.engine.config import EngineConfig
+    from llmq_amd.engine.engine import LLMEngine
+    from llmq_amd.engine.sampling_params import SamplingParams
+
+    model = args.model
+    batch = args.batch
+    prompt_len = args.prompt_len
+    if not use_gpu:
+        # CPU fallback (contract: must run everywhere) — tiny shape.
+        model, batch, prompt_len = "tiny-llama", 4, 32
+
+    cfg = EngineConfig(
+        model=model,
+        max_num_seqs=batch,
+        max_model_len=args.max_model_len if use_gpu else 256,
+        max_prefill_tokens=16384,
+        load_weights=False,
+        fast_init=True,
+        device=device,
+        enforce_eager=args.eager,
+        hipgraph_max_batch=batch,
+        seed=1234 + rank,
+    )
+    t_init = time.perf_counter()
+    engine = LLMEngine(cfg)
+    if rank == 0:
+        print(f"[bench] engine init {time.perf_counter() - t_init:.1f}s", file=sys.stderr)
+
+    # --- load the resident batch (synthetic prompts, random token ids)
+    import numpy as np
+
+    rng = np.random.default_rng(42 + rank)
+    vocab = engine.spec.vocab_size
+    params = SamplingParams(
+        temperature=args.temperature, max_tokens=args.max_model_len, ignore_eos=True
+    )
+    for i in range(batch):
+        ids = rng.integers(0, vocab, size=prompt_len).tolist()
+        engine.add_request(f"bench-{rank}-{i}", prompt_token_ids=ids, params=params)
+
+    t_prefill = time.perf_counter()
+    while engine.scheduler.num_waiting > 0:
+        engine.step()  # prefill batches
+    prefill_s = time.perf_counter() - t_prefill
+    prefill_tokens = batch * prompt_len
+    if rank == 0:
+        print(
+            f"[bench] prefill {prefill_tokens} tokens in {prefill_s:.2f}s "
+            f"({prefill_tokens / prefill_s:.0f} tok/s/gpu)",
+            file=sys.stderr,
+        )
+
+    def barrier_sync():
+        if use_gpu:
+            torch.cuda.synchronize()
+        if distributed:
+            dist.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    # --- warmup
+    for _ in range(args.warmup):
+        engine.step()
+    barrier_sync()
+
+    # --- timed region
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        engine.step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    barrier_sync()
+
+    # slowest rank defines the job time
+    if distributed:
+        t = torch.tensor([elapsed], device=device if use_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    assert engine.scheduler.num_running == batch, (
+        f"batch decayed during timing: {engine.scheduler.num_running} != {batch}"
+    )
+
+    total_tokens = world * batch * args.steps
+    value = total_tokens / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    for _ in range(args.profile_steps):
+        engine.step()
+    if use_gpu:
+        torch.cuda.synchronize()
+
+    if rank == 0:
+        result = {
+            "metric": "output_tokens_per_sec (Tower-Plus-9B continuous-batching serving decode)",
+            "value": round(value, 1),
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"{model} ({engine.spec.name} arch, random init)",
+                "global_batch": world * batch,
+                "seq_len": prompt_len,
+                "parallelism": f"dp{world}",
+            },
+        }
+        # completed jobs/sec companion number: at steady-state decode one
+        # job finishes per output_len generated tokens.
+        out_len = args.max_model_len - prompt_len if use_gpu else 8
+        result["config"]["jobs_per_sec_at_output_len"] = {
+            str(out_len): round(value / max(out_len, 1), 3)
+        }
+        print(json.dumps(result))
+
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -23,6 +23,7 @@ class EngineConfig:
     seed: int = 0
     device: str = "auto"  # "auto" | "cuda" | "cpu"
     load_weights: bool = True  # False → random init (synthetic benchmarking)
+    fast_init: bool = False  # device-side random init (large models)
     # CPU-test override: number of KV blocks (None → sized from free HBM)
     num_kv_blocks: Optional[int] = None
     enforce_eager: bool = False

@@ -113,7 +113,7 @@ class LLMEngine:
 
             load_safetensors_weights(self.model, path)
         else:
-            self.model.random_init(self.config.seed)
+            self.model.random_init(self.config.seed, fast=self.config.fast_init)
 
     def _size_kv_cache(self) -> int:
         cfg = self.config

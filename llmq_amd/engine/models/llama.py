@@ -121,14 +121,25 @@ class CausalLM:
         )
 
     @torch.no_grad()
-    def random_init(self, seed: int = 0) -> None:
+    def random_init(self, seed: int = 0, fast: bool = False) -> None:
         """Random weights for synthetic benchmarking (no network for real
-        checkpoints here). Scaled so activations stay finite in bf16."""
-        gen = torch.Generator(device="cpu").manual_seed(seed)
+        checkpoints here). Scaled so activations stay finite in bf16.
 
-        def fill(t: torch.Tensor, std: float) -> None:
-            cpu = torch.randn(t.shape, generator=gen, dtype=torch.float32) * std
-            t.copy_(cpu.to(t.dtype))
+        fast=True generates on-device (needed for multi-B-param models:
+        host-side generation of a 9B model costs ~40 s); fast=False uses a
+        CPU generator so CPU and GPU engines with the same seed get
+        bit-identical weights (consistency tests)."""
+        if fast and self.device.type == "cuda":
+            gen = torch.Generator(device=self.device).manual_seed(seed)
+
+            def fill(t: torch.Tensor, std: float) -> None:
+                t.normal_(0.0, std, generator=gen)
+        else:
+            gen = torch.Generator(device="cpu").manual_seed(seed)
+
+            def fill(t: torch.Tensor, std: float) -> None:
+                cpu = torch.randn(t.shape, generator=gen, dtype=torch.float32) * std
+                t.copy_(cpu.to(t.dtype))
 
         h = self.spec.hidden_size
         std = 0.02

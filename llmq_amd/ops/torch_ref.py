@@ -54,8 +54,10 @@ def rope_inplace(
     cos = cs[:, :half].unsqueeze(1)  # [T, 1, half]
     sin = cs[:, half:].unsqueeze(1)
     for t in (q, k):
-        x1 = t[..., :half].float()
-        x2 = t[..., half:].float()
+        # clone: for f32 inputs .float() aliases, and the first write below
+        # would corrupt x1 before the second line reads it
+        x1 = t[..., :half].float().clone()
+        x2 = t[..., half:].float().clone()
         t[..., :half] = (x1 * cos - x2 * sin).to(t.dtype)
         t[..., half:] = (x2 * cos + x1 * sin).to(t.dtype)
 
