@@ -19,12 +19,17 @@ from llmq_amd.ops import torch_ref
 
 _EXT = None
 _EXT_ERROR: Optional[str] = None
+_EXT_TRIED = False
 
 
 def _try_load_extension() -> None:
-    global _EXT, _EXT_ERROR
-    if _EXT is not None:
+    """Lazy: called on first use, NOT at import (so `python -m
+    llmq_amd.ops.build` can rebuild without a namespace clash against a
+    stale loaded .so)."""
+    global _EXT, _EXT_ERROR, _EXT_TRIED
+    if _EXT_TRIED:
         return
+    _EXT_TRIED = True
     here = Path(__file__).parent
     candidates = sorted(here.glob("_hip_ops*.so"))
     if not candidates:
@@ -38,19 +43,25 @@ def _try_load_extension() -> None:
         _EXT = torch.ops.llmq_amd
         _EXT_ERROR = None
     except Exception as exc:  # noqa: BLE001
-        _EXT_ERROR = f"failed to load {candidates[0]}: {exc}"
-
-
-_try_load_extension()
+        # If an identical build was already loaded in this process (e.g.
+        # right after `build()`), the namespace exists — use it.
+        try:
+            torch.ops.llmq_amd.rmsnorm  # noqa: B018
+            _EXT = torch.ops.llmq_amd
+            _EXT_ERROR = None
+        except Exception:
+            _EXT_ERROR = f"failed to load {candidates[0]}: {exc}"
 
 
 def has_hip_ext() -> bool:
+    _try_load_extension()
     return _EXT is not None
 
 
 def _use_hip(t: torch.Tensor) -> bool:
     if not t.is_cuda:
         return False
+    _try_load_extension()
     if _EXT is None:
         raise RuntimeError(
             f"llmq_amd HIP extension required for GPU execution but unavailable: {_EXT_ERROR}"

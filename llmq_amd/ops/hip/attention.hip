@@ -43,44 +43,57 @@ __global__ __launch_bounds__(256) void paged_decode_attention_kernel(
 
   constexpr int VE = Vec8<T>::kElems;        // 8 for bf16, 4 for f32
   constexpr int D = HEAD_DIM;
+  constexpr int CHUNK = DECODE_CHUNK;
   const int lane16 = threadIdx.x & 15;       // lane within 16-lane score group
   const int g16 = threadIdx.x >> 4;          // score group id (0..15)
 
-  // LDS: q (G x D f32, pre-scaled), scores (CHUNK x G f32)
+  // Phase-B decomposition: SUBS token-parallel sub-groups per head, each of
+  // 32 threads covering all D dims. All 256 threads work for G*32*SUBS==256.
+  const int tph = 32 * G;                    // threads per head-set
+  const int SUBS = 256 / tph;                // 8,4,2,2,1,1,1,1 for G=1..8
+  const int sub = threadIdx.x / tph;         // token sub-group (>=SUBS: idle)
+  const int h_b = (threadIdx.x % tph) >> 5;  // head
+  constexpr int DPT = D / 32;                // dims per thread (2/4/8)
+  const int d0 = (threadIdx.x & 31) * DPT;
+
+  // LDS: q (G x D f32, pre-scaled) | scores (CHUNK x G) | chunk block ids |
+  //      merge buffers (SUBS x G x (D + 2))
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* q_s = reinterpret_cast<float*>(smem);              // [G][D]
   float* s_s = q_s + G * D;                                 // [CHUNK][G]
+  int* bt_s = reinterpret_cast<int*>(s_s + CHUNK * G);      // [CHUNK/bs up to 4]
+  float* merge_s = reinterpret_cast<float*>(bt_s + 8);      // [SUBS][G][D+2]
 
-  // Stage q, pre-scaled by `scale`.
   for (int i = threadIdx.x; i < G * D; i += blockDim.x) {
     const int h = i / D, d = i % D;
     q_s[i] = to_f32(q[(long)b * q_stride + (long)(kh * G + h) * D + d]) * scale;
   }
-  __syncthreads();
 
   const int start = (window > 0 && L > window) ? (L - window) : 0;
   const int* bt = block_tables + (long)b * max_blocks;
+  const int base0 = (start / CHUNK) * CHUNK;
 
-  // Phase-B per-thread state: head h, dims [d0, d0 + DPT)
-  constexpr int DPT = D / 32;               // dims per thread (2/4/8)
-  const int h_b = threadIdx.x >> 5;         // 0..7
-  const int d0 = (threadIdx.x & 31) * DPT;
   float acc[DPT];
 #pragma unroll
   for (int i = 0; i < DPT; ++i) acc[i] = 0.f;
   float m_run = -1e30f, l_run = 0.f;
 
-  for (int base = (start / DECODE_CHUNK) * DECODE_CHUNK; base < L;
-       base += DECODE_CHUNK) {
-    const int chunk_end = min(base + DECODE_CHUNK, L);
+  for (int base = base0; base < L; base += CHUNK) {
+    const int chunk_end = min(base + CHUNK, L);
+    const int n = chunk_end - base;
+    // stage this chunk's block ids (removes a dependent global load from the
+    // phase-B address chain)
+    if (threadIdx.x < (unsigned)((n + block_size - 1) / block_size))
+      bt_s[threadIdx.x] = bt[(base + threadIdx.x * block_size) / block_size];
+    __syncthreads();
     // ---- phase A: scores for tokens [base, chunk_end)
-    for (int t = base + g16; t < chunk_end; t += 16) {
-      const long blk = bt[t / block_size];
+    for (int t0 = g16; t0 < n; t0 += 16) {
+      const int t = base + t0;
+      const long blk = bt_s[t0 / block_size];
       const int off = t % block_size;
       const T* krow =
           k_cache + (((blk * num_kv_heads + kh) * (long)block_size + off)) * D;
-      // lane16 covers dims [lane16*D/16, ...): D/16 elems = 16B at D=128/bf16
-      constexpr int DL = D / 16;
+      constexpr int DL = D / 16;  // elems per lane16
       float kf[DL];
       if constexpr (DL % VE == 0) {
 #pragma unroll
@@ -102,16 +115,16 @@ __global__ __launch_bounds__(256) void paged_decode_attention_kernel(
         if (lane16 == 0) {
           if (softcap > 0.f) dot = tanhf(dot / softcap) * softcap;
           if (t < start) dot = -1e30f;
-          s_s[(t - base) * G + h] = dot;
+          s_s[t0 * G + h] = dot;
         }
       }
     }
     __syncthreads();
-    // ---- phase B: online softmax + V accumulation
-    if (h_b < G) {
-      const int n = chunk_end - base;
+    // ---- phase B: online softmax + V accumulation (sub-group token split)
+    if (sub < SUBS) {
       float m_chunk = -1e30f;
-      for (int t = 0; t < n; ++t) m_chunk = fmaxf(m_chunk, s_s[t * G + h_b]);
+      for (int t = sub; t < n; t += SUBS)
+        m_chunk = fmaxf(m_chunk, s_s[t * G + h_b]);
       const float m_new = fmaxf(m_run, m_chunk);
       if (m_new > -1e30f) {
         const float alpha = __expf(m_run - m_new);
@@ -119,11 +132,13 @@ __global__ __launch_bounds__(256) void paged_decode_attention_kernel(
         for (int i = 0; i < DPT; ++i) acc[i] *= alpha;
         l_run *= alpha;
         m_run = m_new;
-        for (int t = 0; t < n; ++t) {
+        // Iterations are independent (chunk-max softmax, no rescale inside)
+        // so V loads pipeline across tokens.
+        for (int t = sub; t < n; t += SUBS) {
           const float p = __expf(s_s[t * G + h_b] - m_new);
           l_run += p;
           const int tok = base + t;
-          const long blk = bt[tok / block_size];
+          const long blk = bt_s[t / block_size];
           const int off = tok % block_size;
           const T* vrow =
               v_cache + (((blk * num_kv_heads + kh) * (long)block_size + off)) * D + d0;
@@ -132,11 +147,42 @@ __global__ __launch_bounds__(256) void paged_decode_attention_kernel(
         }
       }
     }
-    __syncthreads();  // protect s_s for the next chunk
+    __syncthreads();  // protect s_s / bt_s for the next chunk
   }
 
-  if (h_b < G) {
-    const float inv = (l_run > 0.f) ? 1.0f / l_run : 0.f;
+  // ---- merge sub-group partials (flash-decoding style, within the block)
+  const int d2 = D + 2;
+  if (sub < SUBS && SUBS > 1) {
+    float* slot = merge_s + (sub * G + h_b) * d2;
+#pragma unroll
+    for (int i = 0; i < DPT; ++i) slot[d0 + i] = acc[i];
+    if (d0 == 0) {
+      slot[D] = m_run;
+      slot[D + 1] = l_run;
+    }
+  }
+  __syncthreads();
+  if (sub == 0) {
+    float m_tot = m_run, l_tot = 0.f;
+    if (SUBS > 1) {
+      for (int s2 = 0; s2 < SUBS; ++s2)
+        m_tot = fmaxf(m_tot, merge_s[(s2 * G + h_b) * d2 + D]);
+      float accm[DPT];
+#pragma unroll
+      for (int i = 0; i < DPT; ++i) accm[i] = 0.f;
+      for (int s2 = 0; s2 < SUBS; ++s2) {
+        const float* slot = merge_s + (s2 * G + h_b) * d2;
+        const float w = __expf(slot[D] - m_tot);
+        l_tot += w * slot[D + 1];
+#pragma unroll
+        for (int i = 0; i < DPT; ++i) accm[i] += w * slot[d0 + i];
+      }
+#pragma unroll
+      for (int i = 0; i < DPT; ++i) acc[i] = accm[i];
+    } else {
+      l_tot = l_run;
+    }
+    const float inv = (l_tot > 0.f) ? 1.0f / l_tot : 0.f;
     T* orow = out + (long)b * out_stride + (long)(kh * G + h_b) * D + d0;
 #pragma unroll
     for (int i = 0; i < DPT; ++i) orow[i] = from_f32<T>(acc[i] * inv);
@@ -222,5 +268,221 @@ __global__ __launch_bounds__(256) void varlen_prefill_attention_kernel(
 #pragma unroll
     for (int t = 0; t < DPT; ++t) orow[lane * DPT + t] = from_f32<T>(acc[t] * inv);
     __builtin_amdgcn_wave_barrier();
+  }
+}
+
+// ------------------------------------------------------ MFMA flash prefill --
+// Packed varlen causal GQA attention on matrix cores (bf16).
+// Grid: (seq, q_head, q_tile). Workgroup: 4 waves; each wave owns 16 query
+// rows of a 64-row Q tile. KV tiles of 64 keys staged in XOR-swizzled LDS
+// (shared by the 4 waves); per KV tile each wave computes
+//   S[16q,64k]  via mfma_f32_16x16x32_bf16 (Q frags in registers, K^T frags
+//               from LDS b128 reads)
+//   online softmax in the MFMA C-layout (row r = (lane>>4)*4+reg lives on
+//               the 16 col-lanes -> shfl_xor over the low 4 lane bits)
+//   OT[D,16q] += mfma(A=V^T, B=P^T)  -- transposed PV so the P fragment is a
+//               contiguous b128 read of the per-wave P_lds tile and the
+//               softmax rescale factor is lane-uniform (qrow = lane&15).
+// Epilogue transposes OT back with scalar stores.
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
+typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+
+#define PF_QT 64   // q rows per workgroup
+#define PF_KT 64   // keys per kv tile
+
+DEVINL int swz(int row, int d) {  // element-index XOR swizzle (16B granules)
+  return d ^ ((row & 7) << 3);
+}
+
+template <int HEAD_DIM>
+__global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
+    __hip_bfloat16* __restrict__ out,      // [T, H, D]
+    const __hip_bfloat16* __restrict__ q,  // [T, H, D]
+    const __hip_bfloat16* __restrict__ k,  // [T, KVH, D]
+    const __hip_bfloat16* __restrict__ v,
+    const int* __restrict__ cu_seqlens, int num_heads, int num_kv_heads,
+    float scale, float softcap, int window, long q_stride, long k_stride,
+    long v_stride, long o_stride) {
+  constexpr int D = HEAD_DIM;
+  const int seq = blockIdx.x;
+  const int h = blockIdx.y;
+  const int kvh = h / (num_heads / num_kv_heads);
+  const int s0 = cu_seqlens[seq];
+  const int L = cu_seqlens[seq + 1] - s0;
+  const int q_base = blockIdx.z * PF_QT;
+  if (q_base >= L) return;
+
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid & (WAVE - 1);
+  const int col = lane & 15;        // MFMA col lane (key for S, qrow for OT)
+  const int kgrp = lane >> 4;       // 0..3
+
+  // LDS: K tile | V tile | per-wave P tiles | per-wave row stats
+  __shared__ __attribute__((aligned(16))) short k_lds[PF_KT * D];
+  __shared__ __attribute__((aligned(16))) short v_lds[PF_KT * D];
+  __shared__ __attribute__((aligned(16))) short p_lds[4][16 * PF_KT];
+  __shared__ float stat_lds[4][2][16];  // [wave][alpha|inv_l][row]
+
+  // ---- load Q fragments (registers): frag[ks] covers dims ks*32+(kgrp*8..+7)
+  constexpr int KS = D / 32;
+  bf16x8_t qfrag[KS];
+  const int my_qrow = q_base + wid * 16 + col;  // A-frag row = lane&15
+  {
+    const int r = (my_qrow < L) ? my_qrow : (L - 1);
+    const __hip_bfloat16* qrow_p = q + (long)(s0 + r) * q_stride + (long)h * D;
+#pragma unroll
+    for (int ks = 0; ks < KS; ++ks)
+      qfrag[ks] = *reinterpret_cast<const bf16x8_t*>(qrow_p + ks * 32 + kgrp * 8);
+  }
+
+  // per-lane softmax state for rows rr = wid*16 + kgrp*4 + reg
+  float m_run[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float l_run[4] = {0.f, 0.f, 0.f, 0.f};
+  constexpr int DT = D / 16;
+  f32x4_t ot[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt) ot[dt] = {0.f, 0.f, 0.f, 0.f};
+
+  const int wave_max_row = min(q_base + wid * 16 + 15, L - 1);
+  const int block_max_row = min(q_base + PF_QT - 1, L - 1);
+  const int kv_end = block_max_row + 1;  // causal bound for the workgroup
+  int kv_begin = 0;
+  if (window > 0) {
+    const int wave_min_needed = q_base + 1 - window;  // earliest key any row sees
+    kv_begin = max(0, (wave_min_needed / PF_KT) * PF_KT);
+  }
+
+  for (int kt = kv_begin; kt < kv_end; kt += PF_KT) {
+    // ---- stage K/V tile (zero-fill beyond L so garbage never reaches MFMA)
+    {
+      constexpr int CHW = 8;  // elems per 16B chunk
+      const int chunks = PF_KT * D / CHW;
+      for (int c = tid; c < chunks; c += 256) {
+        const int key = c / (D / CHW);
+        const int d8 = (c % (D / CHW)) * CHW;
+        const int dst = key * D + swz(key, d8);
+        const int gkey = kt + key;
+        if (gkey < L) {
+          *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) =
+              *reinterpret_cast<const bf16x8_t*>(
+                  k + (long)(s0 + gkey) * k_stride + (long)kvh * D + d8);
+          *reinterpret_cast<bf16x8_t*>(&v_lds[dst]) =
+              *reinterpret_cast<const bf16x8_t*>(
+                  v + (long)(s0 + gkey) * v_stride + (long)kvh * D + d8);
+        } else {
+          *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+          *reinterpret_cast<bf16x8_t*>(&v_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+        }
+      }
+    }
+    __syncthreads();
+
+    if (kt <= wave_max_row) {  // this wave has rows that see this tile
+      // ---- S = Q K^T for 4 column tiles of 16 keys
+      f32x4_t s[4];
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        s[ct] = {0.f, 0.f, 0.f, 0.f};
+        const int key = ct * 16 + col;
+#pragma unroll
+        for (int ks = 0; ks < KS; ++ks) {
+          const int d8 = ks * 32 + kgrp * 8;
+          bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
+              &k_lds[key * D + swz(key, d8)]);
+          s[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks], bfrag, s[ct], 0, 0, 0);
+        }
+      }
+      // ---- scale, softcap, mask; rows rr = wid*16+kgrp*4+reg
+      float m_tile[4], l_tile[4], p[4][4];
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = q_base + wid * 16 + kgrp * 4 + reg;
+        float mx = -1e30f;
+#pragma unroll
+        for (int ct = 0; ct < 4; ++ct) {
+          const int key = kt + ct * 16 + col;
+          float x = s[ct][reg] * scale;
+          if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
+          const bool dead = key > row || key >= L || row >= L ||
+                            (window > 0 && key <= row - window);
+          x = dead ? -1e30f : x;
+          p[reg][ct] = x;
+          mx = fmaxf(mx, x);
+        }
+        // reduce max over the 16 col lanes
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
+        m_tile[reg] = mx;
+      }
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const float m_new = fmaxf(m_run[reg], m_tile[reg]);
+        const float alpha = (m_new > -1e30f) ? __expf(m_run[reg] - m_new) : 1.f;
+        float lsum = 0.f;
+#pragma unroll
+        for (int ct = 0; ct < 4; ++ct) {
+          const float pe = (m_new > -1e30f && p[reg][ct] > -1e29f)
+                               ? __expf(p[reg][ct] - m_new) : 0.f;
+          p[reg][ct] = pe;
+          lsum += pe;
+        }
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) lsum += __shfl_xor(lsum, off, WAVE);
+        l_run[reg] = l_run[reg] * alpha + lsum;
+        m_run[reg] = m_new;
+        // stash alpha for the OT lanes (indexed by qrow = lane&15)
+        if (col == 0) stat_lds[wid][0][kgrp * 4 + reg] = alpha;
+        // write P row to the wave's P tile (bf16)
+#pragma unroll
+        for (int ct = 0; ct < 4; ++ct) {
+          p_lds[wid][(kgrp * 4 + reg) * PF_KT + ct * 16 + col] =
+              __bfloat16_as_short(__float2bfloat16(p[reg][ct]));
+        }
+      }
+      // ---- OT += V^T P^T  (A = V^T frags via scalar LDS reads, B = P^T b128)
+      const float alpha_q = stat_lds[wid][0][col];
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
+        ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
+#pragma unroll
+        for (int ks = 0; ks < PF_KT / 32; ++ks) {
+          bf16x8_t a;
+          const int dim = dt * 16 + col;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int key = ks * 32 + kgrp * 8 + j;
+            a[j] = v_lds[key * D + swz(key, dim & ~7) + (dim & 7)];
+          }
+          bf16x8_t b = *reinterpret_cast<const bf16x8_t*>(
+              &p_lds[wid][col * PF_KT + ks * 32 + kgrp * 8]);
+          ot[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, ot[dt], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: normalise and store (transpose OT back per lane)
+  if (col == 0) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const float l = l_run[reg];
+      stat_lds[wid][1][kgrp * 4 + reg] = (l > 0.f) ? 1.0f / l : 0.f;
+    }
+  }
+  __builtin_amdgcn_s_waitcnt(0);  // lgkm drain before same-wave read
+  const float inv = stat_lds[wid][1][col];
+  const int orow = q_base + wid * 16 + col;
+  if (orow < L) {
+    __hip_bfloat16* op = out + (long)(s0 + orow) * o_stride + (long)h * D;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg)
+        op[dt * 16 + kgrp * 4 + reg] = __float2bfloat16(ot[dt][reg] * inv);
+    }
   }
 }

@@ -32,6 +32,11 @@ __global__ void varlen_prefill_attention_kernel(T*, const T*, const T*, const T*
                                                 const int*, int, int, float,
                                                 float, int, long, long, long,
                                                 long);
+template <int HEAD_DIM>
+__global__ void flash_prefill_bf16_kernel(__hip_bfloat16*, const __hip_bfloat16*,
+                                          const __hip_bfloat16*, const __hip_bfloat16*,
+                                          const int*, int, int, float, float, int,
+                                          long, long, long, long);
 
 namespace {
 
@@ -172,7 +177,8 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
   TORCH_CHECK(H % KVH == 0 && G <= 8, "GQA group must be <= 8 (got ", G, ")");
   TORCH_CHECK(bs == 16 || bs == 32, "block_size must be 16 or 32");
   dim3 grid(B, KVH);
-  const int lds = (G * D + 64 * G) * sizeof(float);
+  const int subs = 256 / (32 * G);
+  const int lds = (G * D + 64 * G + subs * G * (D + 2)) * sizeof(float) + 8 * sizeof(int);
   auto l = [&]<int HD>() {
     hipLaunchKernelGGL((paged_decode_attention_kernel<T, HD>), grid, dim3(256),
                        lds, stream(), reinterpret_cast<T*>(out.data_ptr()),
@@ -208,9 +214,31 @@ void paged_decode_attention(at::Tensor out, at::Tensor q, at::Tensor k_cache,
 template <typename T>
 void launch_prefill(at::Tensor& out, const at::Tensor& q, const at::Tensor& k,
                     const at::Tensor& v, const at::Tensor& cu, int B,
-                    double scale, double softcap, long window) {
+                    double scale, double softcap, long window, long max_seqlen) {
   const int H = q.size(1), D = q.size(2);
   const int KVH = k.size(1);
+  if constexpr (std::is_same_v<T, __hip_bfloat16>) {
+    // MFMA flash path (bf16): grid (seq, head, q-tile)
+    const int qtiles = (int)((max_seqlen + 63) / 64);
+    dim3 fgrid(B, H, std::max(qtiles, 1));
+    auto lf = [&]<int HD>() {
+      hipLaunchKernelGGL((flash_prefill_bf16_kernel<HD>), fgrid, dim3(256), 0,
+                         stream(),
+                         reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+                         cu.data_ptr<int>(), H, KVH, (float)scale,
+                         (float)softcap, (int)window, q.stride(0), k.stride(0),
+                         v.stride(0), out.stride(0));
+    };
+    switch (D) {
+      case 64: lf.template operator()<64>(); return;
+      case 128: lf.template operator()<128>(); return;
+      case 256: lf.template operator()<256>(); return;
+      default: TORCH_CHECK(false, "unsupported head_dim ", D);
+    }
+  }
   dim3 grid(B, H);
   auto l = [&]<int HD>() {
     hipLaunchKernelGGL((varlen_prefill_attention_kernel<T, HD>), grid,
@@ -239,9 +267,9 @@ void varlen_prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,
   CHECK_LASTDIM(q);
   TORCH_CHECK(cu_seqlens.scalar_type() == at::kInt);
   const int B = cu_seqlens.size(0) - 1;
-  (void)max_seqlen;
   dispatch_dtype(q, "varlen_prefill_attention", [&]<typename T>() {
-    launch_prefill<T>(out, q, k, v, cu_seqlens, B, scale, softcap, window);
+    launch_prefill<T>(out, q, k, v, cu_seqlens, B, scale, softcap, window,
+                      max_seqlen);
   });
 }
 
