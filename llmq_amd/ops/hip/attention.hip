@@ -22,7 +22,7 @@
 
 #include "common.h"
 
-#define DECODE_CHUNK 64  // tokens per two-phase iteration (4 KV blocks @ bs 16)
+#define DECODE_CHUNK 128  // tokens per two-phase iteration (8 KV blocks @ bs 16)
 
 template <typename T, int HEAD_DIM>
 __global__ __launch_bounds__(256) void paged_decode_attention_kernel(
@@ -61,8 +61,8 @@ __global__ __launch_bounds__(256) void paged_decode_attention_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* q_s = reinterpret_cast<float*>(smem);              // [G][D]
   float* s_s = q_s + G * D;                                 // [CHUNK][G]
-  int* bt_s = reinterpret_cast<int*>(s_s + CHUNK * G);      // [CHUNK/bs up to 4]
-  float* merge_s = reinterpret_cast<float*>(bt_s + 8);      // [SUBS][G][D+2]
+  int* bt_s = reinterpret_cast<int*>(s_s + CHUNK * G);      // [CHUNK/bs up to 16]
+  float* merge_s = reinterpret_cast<float*>(bt_s + 16);     // [SUBS][G][D+2]
 
   for (int i = threadIdx.x; i < G * D; i += blockDim.x) {
     const int h = i / D, d = i % D;
@@ -142,8 +142,29 @@ __global__ __launch_bounds__(256) void paged_decode_attention_kernel(
           const int off = tok % block_size;
           const T* vrow =
               v_cache + (((blk * num_kv_heads + kh) * (long)block_size + off)) * D + d0;
+          // Vectorized V row slice: DPT contiguous elems per thread
+          // (16B for bf16 D=256, 8B for D=128) — scalar 2B loads here
+          // were the v2 bandwidth ceiling.
+          if constexpr (DPT % Vec8<T>::kElems == 0) {
+            constexpr int VE8 = Vec8<T>::kElems;
 #pragma unroll
-          for (int i = 0; i < DPT; ++i) acc[i] += p * to_f32(vrow[i]);
+            for (int c = 0; c < DPT; c += VE8) {
+              Vec8<T> vv = load16(vrow + c);
+#pragma unroll
+              for (int j = 0; j < VE8; ++j) acc[c + j] += p * to_f32(vv.data[j]);
+            }
+          } else if constexpr (DPT % Vec4<T>::kElems == 0) {
+            constexpr int VE4 = Vec4<T>::kElems;
+#pragma unroll
+            for (int c = 0; c < DPT; c += VE4) {
+              Vec4<T> vv = load8(vrow + c);
+#pragma unroll
+              for (int j = 0; j < VE4; ++j) acc[c + j] += p * to_f32(vv.data[j]);
+            }
+          } else {
+#pragma unroll
+            for (int i = 0; i < DPT; ++i) acc[i] += p * to_f32(vrow[i]);
+          }
         }
       }
     }

@@ -178,7 +178,8 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
   TORCH_CHECK(bs == 16 || bs == 32, "block_size must be 16 or 32");
   dim3 grid(B, KVH);
   const int subs = 256 / (32 * G);
-  const int lds = (G * D + 64 * G + subs * G * (D + 2)) * sizeof(float) + 8 * sizeof(int);
+  const int lds = (G * D + DECODE_CHUNK * G + subs * G * (D + 2)) * sizeof(float) +
+                  16 * sizeof(int);
   auto l = [&]<int HD>() {
     hipLaunchKernelGGL((paged_decode_attention_kernel<T, HD>), grid, dim3(256),
                        lds, stream(), reinterpret_cast<T*>(out.data_ptr()),

@@ -48,6 +48,19 @@ DEVINL void store16(T* p, const Vec8<T>& v) {
   *reinterpret_cast<int4*>(p) = *reinterpret_cast<const int4*>(v.data);
 }
 
+template <typename T> struct Vec4 {
+  // 8-byte vector: 4 elements of a 2-byte T, 2 of a 4-byte T.
+  static constexpr int kElems = (sizeof(T) == 2) ? 4 : 2;
+  T data[kElems];
+};
+
+template <typename T>
+DEVINL Vec4<T> load8(const T* p) {
+  Vec4<T> v;
+  *reinterpret_cast<int2*>(v.data) = *reinterpret_cast<const int2*>(p);
+  return v;
+}
+
 // ---- wave / group reductions ----------------------------------------------
 
 DEVINL float wave_reduce_sum(float x) {
