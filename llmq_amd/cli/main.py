@@ -190,8 +190,13 @@ def worker() -> None:
 @click.option("--max-num-seqs", type=int, default=None)
 @click.option("--max-model-len", type=int, default=None)
 @click.option("--prefetch", type=int, default=None)
-def worker_run(model, queue_name, tensor_parallel_size, max_num_seqs, max_model_len, prefetch):
+@click.option("--engine-overrides", default=None,
+              help='JSON dict of EngineConfig overrides, e.g. \'{"enforce_eager": true}\'')
+def worker_run(model, queue_name, tensor_parallel_size, max_num_seqs, max_model_len,
+               prefetch, engine_overrides):
     """GPU inference worker (in-tree MI355X engine)."""
+    import json as _json
+
     from llmq_amd.cli.worker import run_engine_worker
 
     run_engine_worker(
@@ -200,6 +205,7 @@ def worker_run(model, queue_name, tensor_parallel_size, max_num_seqs, max_model_
         max_num_seqs=max_num_seqs,
         max_model_len=max_model_len,
         prefetch=prefetch,
+        engine_overrides=_json.loads(engine_overrides) if engine_overrides else None,
     )
 
 

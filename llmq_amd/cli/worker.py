@@ -28,6 +28,7 @@ def run_engine_worker(
     pipeline: Optional[PipelineConfig] = None,
     stage_name: Optional[str] = None,
     stage_config: Optional[dict] = None,
+    engine_overrides: Optional[dict] = None,
 ) -> None:
     setup_logging(json_output=True)
     from llmq_amd.workers.engine_worker import EngineWorker
@@ -42,6 +43,7 @@ def run_engine_worker(
         pipeline=pipeline,
         stage_name=stage_name,
         stage_config=stage_config or {},
+        engine_overrides=engine_overrides,
     )
     asyncio.run(worker.run())
 
