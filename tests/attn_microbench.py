@@ -1,0 +1,90 @@
+#!/usr/bin/env python3
+"""Standalone decode-attention microbenchmark (GPU box tool, not a pytest).
+
+Times paged_decode_attention at the serving shape (Tower-Plus-9B/gemma2:
+B=256 seqs, ctx ~1064, 16 q heads / 8 kv heads, D=256, block 16) and
+reports ms + achieved KV-read bandwidth. Run under rocprofv3 --pmc for
+counter analysis.
+
+  python tests/attn_microbench.py [--ctx 1064] [--batch 256] [--iters 50]
+"""
+
+from __future__ import annotations
+
+import argparse
+import time
+
+import torch
+
+from llmq_amd import ops
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--batch", type=int, default=256)
+    ap.add_argument("--ctx", type=int, default=1064)
+    ap.add_argument("--heads", type=int, default=16)
+    ap.add_argument("--kv-heads", type=int, default=8)
+    ap.add_argument("--head-dim", type=int, default=256)
+    ap.add_argument("--block-size", type=int, default=16)
+    ap.add_argument("--iters", type=int, default=50)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--softcap", type=float, default=0.0)
+    ap.add_argument("--window", type=int, default=0)
+    ap.add_argument("--check", action="store_true", help="verify vs torch ref")
+    args = ap.parse_args()
+
+    assert torch.cuda.is_available()
+    assert ops.has_hip_ext()
+    dev = torch.device("cuda:0")
+    dtype = torch.bfloat16
+    B, L, H, KVH, D, bs = (args.batch, args.ctx, args.heads, args.kv_heads,
+                           args.head_dim, args.block_size)
+    blocks_per_seq = (L + bs - 1) // bs
+    num_blocks = B * blocks_per_seq + 1
+
+    torch.manual_seed(0)
+    q = torch.randn(B, H, D, device=dev, dtype=dtype)
+    k_cache = torch.randn(num_blocks, KVH, bs, D, device=dev, dtype=dtype)
+    v_cache = torch.randn(num_blocks, KVH, bs, D, device=dev, dtype=dtype)
+    block_tables = torch.arange(
+        1, 1 + B * blocks_per_seq, device=dev, dtype=torch.int32
+    ).reshape(B, blocks_per_seq)
+    context_lens = torch.full((B,), L, device=dev, dtype=torch.int32)
+    out = torch.empty_like(q)
+    scale = D ** -0.5
+
+    def run() -> None:
+        ops.paged_decode_attention(
+            out, q, k_cache, v_cache, block_tables, context_lens,
+            scale, args.softcap, args.window,
+        )
+
+    for _ in range(args.warmup):
+        run()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.iters):
+        run()
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / args.iters
+
+    kv_bytes = 2 * B * L * KVH * D * 2  # K+V bf16 read once per (b, kh)
+    print(f"shape B={B} L={L} H={H} KVH={KVH} D={D} bs={bs}")
+    print(f"{dt * 1e3:.3f} ms/iter   KV {kv_bytes / 2**30:.2f} GiB   "
+          f"{kv_bytes / dt / 1e12:.2f} TB/s effective")
+
+    if args.check:
+        from llmq_amd.ops import torch_ref
+
+        ref = torch_ref.paged_decode_attention(
+            q.float(), k_cache.float(), v_cache.float(), block_tables,
+            context_lens, scale, args.softcap, args.window,
+        ).float()
+        err = (out.float() - ref).abs().max().item()
+        print(f"max|err| vs f32 ref: {err:.4e}")
+        assert err < 0.05
+
+
+if __name__ == "__main__":
+    main()
