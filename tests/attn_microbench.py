@@ -12,11 +12,15 @@ counter analysis.
 from __future__ import annotations
 
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
-from llmq_amd import ops
+from llmq_amd import ops  # noqa: E402
 
 
 def main() -> None:
