@@ -38,10 +38,11 @@ def main() -> None:
     ap.add_argument("--check", action="store_true", help="verify vs torch ref")
     args = ap.parse_args()
 
-    assert torch.cuda.is_available()
-    assert ops.has_hip_ext()
-    dev = torch.device("cuda:0")
-    dtype = torch.bfloat16
+    use_gpu = torch.cuda.is_available()
+    if use_gpu:
+        assert ops.has_hip_ext()
+    dev = torch.device("cuda:0" if use_gpu else "cpu")
+    dtype = torch.bfloat16 if use_gpu else torch.float32
     B, L, H, KVH, D, bs = (args.batch, args.ctx, args.heads, args.kv_heads,
                            args.head_dim, args.block_size)
     blocks_per_seq = (L + bs - 1) // bs
@@ -55,23 +56,28 @@ def main() -> None:
         1, 1 + B * blocks_per_seq, device=dev, dtype=torch.int32
     ).reshape(B, blocks_per_seq)
     context_lens = torch.full((B,), L, device=dev, dtype=torch.int32)
-    out = torch.empty_like(q)
     scale = D ** -0.5
+    out = q
 
     def run() -> None:
-        ops.paged_decode_attention(
-            out, q, k_cache, v_cache, block_tables, context_lens,
+        nonlocal_out[0] = ops.paged_decode_attention(
+            q, k_cache, v_cache, block_tables, context_lens,
             scale, args.softcap, args.window,
         )
 
+    nonlocal_out = [None]
+
     for _ in range(args.warmup):
         run()
-    torch.cuda.synchronize()
+    if use_gpu:
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.iters):
         run()
-    torch.cuda.synchronize()
+    if use_gpu:
+        torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / args.iters
+    out = nonlocal_out[0]
 
     kv_bytes = 2 * B * L * KVH * D * 2  # K+V bf16 read once per (b, kh)
     print(f"shape B={B} L={L} H={H} KVH={KVH} D={D} bs={bs}")
