@@ -30,7 +30,7 @@ logger = logging.getLogger(__name__)
 _RESERVE_BYTES = 4 << 30
 
 
-@dataclass
+@dataclass(slots=True)
 class RequestOutput:
     request_id: str
     new_token_ids: List[int] = field(default_factory=list)
@@ -191,19 +191,33 @@ class LLMEngine:
         now = time.time()
         self.steps += 1
         outputs: List[RequestOutput] = []
+        is_prefill = batch.kind == "prefill"
+        eos_id = self.tokenizer.eos_token_id
+        mml = self.max_model_len
         for seq, tok in zip(batch.seqs, token_list):
-            if batch.kind == "prefill":
+            if is_prefill:
                 self._prefill_done_at[seq.request_id] = now
                 if seq.first_token_time is None:
                     seq.first_token_time = now
-            seq.append_token(int(tok))
+            seq.token_ids.append(tok)
+            params = seq.params
+            nt = len(seq.token_ids)
             out = RequestOutput(
                 request_id=seq.request_id,
-                new_token_ids=[int(tok)],
+                new_token_ids=[tok],
                 prompt_tokens=seq.prompt_len,
-                output_tokens=seq.output_len,
+                output_tokens=nt - seq.prompt_len,
             )
-            reason = self._check_finish(seq)
+            # Cheap pre-check: most decode steps finish nothing; only take
+            # the full _check_finish path (stop-string decode etc.) when a
+            # finish condition can actually hold.
+            maybe_done = (
+                (tok == eos_id and not params.ignore_eos)
+                or nt - seq.prompt_len >= params.max_tokens
+                or nt >= mml
+                or params.stop
+            )
+            reason = self._check_finish(seq) if maybe_done else None
             if reason is not None:
                 self.scheduler.finish(seq, reason)
                 out.finished = True

@@ -175,9 +175,16 @@ class Scheduler:
             return batch
         # Ensure every running seq has a KV slot for its next position;
         # preempt from the back (youngest) on allocation failure.
+        # Cheap gate first: a decode step grows a seq by one token, so a new
+        # block is needed only when the tokens exceed the table's capacity —
+        # ~1/block_size of seqs per step take the slow path.
+        bs = self.block_size
         i = 0
         while i < len(self.running):
             seq = self.running[i]
+            if len(seq.token_ids) <= len(seq.block_table) * bs:
+                i += 1
+                continue
             need = blocks_needed(seq.num_tokens, self.block_size) - len(seq.block_table)
             if need > 0:
                 blocks = self.allocator.allocate(need)
