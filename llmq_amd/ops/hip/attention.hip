@@ -507,3 +507,200 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
     }
   }
 }
+
+// ------------------------------------------------------- MFMA paged decode --
+// Single-token GQA decode attention on matrix cores (bf16, D in {128,256},
+// G <= 16). One workgroup per (sequence, kv_head); the GQA query group is
+// padded to a 16-row MFMA tile (pad rows repeat q-head 0 and are never
+// stored). Per 64-key chunk, the 4 waves cooperate:
+//   stage: all 256 threads gather the chunk's K and V rows from the paged
+//          cache into XOR-swizzled LDS (zero-filled beyond L)
+//   S:     wave w computes S[16q, 16keys] for key slab w via
+//          mfma_f32_16x16x32_bf16; row-max partials exchanged through LDS
+//   P:     every wave combines the 4 partial maxes identically (shared
+//          m_run), builds its P slab in LDS (bf16), keeps a PER-WAVE l_run
+//          (summed once at the end — avoids a third stats barrier)
+//   PV:    wave w owns dim slab [w*D/4, (w+1)*D/4): OT[dims,16q] +=
+//          mfma(A = V^T frags, B = P^T b128 reads); the online-softmax
+//          rescale factor is lane-uniform (qrow = lane&15), read from LDS.
+// The decode-v2 kernel above is VALU-issue-bound (~120 cyc/token/wave:
+// scalar dot + exp + per-token addressing); this one moves QK^T and PV onto
+// MFMA (~6 cyc/token/wave) so the kernel runs at the KV HBM stream rate.
+
+#define PD_KT 64  // keys per chunk
+
+template <int HEAD_DIM>
+__global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
+    __hip_bfloat16* __restrict__ out,      // [B, H, D]
+    const __hip_bfloat16* __restrict__ q,  // [B, H, D]
+    const __hip_bfloat16* __restrict__ k_cache,  // [nb, KVH, bs, D]
+    const __hip_bfloat16* __restrict__ v_cache,
+    const int* __restrict__ block_tables,  // [B, max_blocks]
+    const int* __restrict__ context_lens,  // [B]
+    int num_heads, int num_kv_heads, int block_size, int max_blocks,
+    float scale, float softcap, int window, long q_stride, long out_stride) {
+  constexpr int D = HEAD_DIM;
+  constexpr int KS = D / 32;   // MFMA K-steps over the head dim
+  constexpr int D4 = D / 4;    // dim slab per wave
+  constexpr int DT = D4 / 16;  // 16-dim MFMA tiles per wave
+  const int b = blockIdx.x;
+  const int kh = blockIdx.y;
+  const int G = num_heads / num_kv_heads;  // <= 16
+  const int L = context_lens[b];
+  if (L <= 0) return;
+
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid & (WAVE - 1);
+  const int col = lane & 15;   // MFMA col lane (key for S, qrow for OT)
+  const int kgrp = lane >> 4;  // 0..3
+
+  __shared__ __attribute__((aligned(16))) short k_lds[PD_KT * D];
+  __shared__ __attribute__((aligned(16))) short v_lds[PD_KT * D];
+  __shared__ __attribute__((aligned(16))) short p_lds[16 * PD_KT];  // [qrow][key]
+  __shared__ float mpart_lds[4][16];  // per-wave row-max partials
+  __shared__ float alpha_lds[16];     // per-row rescale for the OT lanes
+  __shared__ float l_lds[4][16];      // per-wave l_run (end merge)
+
+  // ---- Q fragments: A-operand rows = padded q rows (row = lane&15)
+  bf16x8_t qfrag[KS];
+  {
+    const int qrow = (col < G) ? col : 0;
+    const __hip_bfloat16* qp = q + (long)b * q_stride + (long)(kh * G + qrow) * D;
+#pragma unroll
+    for (int ks = 0; ks < KS; ++ks)
+      qfrag[ks] = *reinterpret_cast<const bf16x8_t*>(qp + ks * 32 + kgrp * 8);
+  }
+
+  // softmax state: every wave redundantly tracks rows r = kgrp*4 + reg
+  float m_regs[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float l_regs[4] = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t ot[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt) ot[dt] = {0.f, 0.f, 0.f, 0.f};
+
+  const int start = (window > 0 && L > window) ? (L - window) : 0;
+  const int* bt = block_tables + (long)b * max_blocks;
+  const int base0 = (start / PD_KT) * PD_KT;
+
+  for (int base = base0; base < L; base += PD_KT) {
+    // ---- stage K/V chunk (gather via block table; zeros beyond L)
+    {
+      constexpr int CPK = D / 8;        // 16B chunks per key row
+      const int nck = PD_KT * CPK;      // chunks per tile
+      for (int c = tid; c < nck; c += 256) {
+        const int key = c / CPK;
+        const int d8 = (c % CPK) * 8;
+        const int dst = key * D + swz(key, d8);
+        const int gkey = base + key;
+        if (gkey < L) {
+          const int blk = bt[gkey / block_size];
+          const long rowoff =
+              (((long)blk * num_kv_heads + kh) * block_size + gkey % block_size) * D + d8;
+          *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) =
+              *reinterpret_cast<const bf16x8_t*>(k_cache + rowoff);
+          *reinterpret_cast<bf16x8_t*>(&v_lds[dst]) =
+              *reinterpret_cast<const bf16x8_t*>(v_cache + rowoff);
+        } else {
+          *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+          *reinterpret_cast<bf16x8_t*>(&v_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- S[16, 16] for this wave's key slab
+    f32x4_t s = {0.f, 0.f, 0.f, 0.f};
+    {
+      const int key = wid * 16 + col;
+#pragma unroll
+      for (int ks = 0; ks < KS; ++ks) {
+        const int d8 = ks * 32 + kgrp * 8;
+        bf16x8_t bfrag =
+            *reinterpret_cast<const bf16x8_t*>(&k_lds[key * D + swz(key, d8)]);
+        s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks], bfrag, s, 0, 0, 0);
+      }
+    }
+    // scale, softcap, bounds mask; rows r = kgrp*4 + reg
+    float sv[4], mx[4];
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int key = base + wid * 16 + col;
+      float x = s[reg] * scale;
+      if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
+      const bool dead = key >= L || key < start;
+      sv[reg] = dead ? -1e30f : x;
+      float m = sv[reg];
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) m = fmaxf(m, __shfl_xor(m, off, WAVE));
+      mx[reg] = m;  // this wave's slab max for row reg
+    }
+    if (col == 0) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) mpart_lds[wid][kgrp * 4 + reg] = mx[reg];
+    }
+    __syncthreads();
+
+    // ---- combine maxes (identical on every wave), build P, update l
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int row = kgrp * 4 + reg;
+      float m_tile = fmaxf(fmaxf(mpart_lds[0][row], mpart_lds[1][row]),
+                           fmaxf(mpart_lds[2][row], mpart_lds[3][row]));
+      const float m_new = fmaxf(m_regs[reg], m_tile);
+      const float alpha = (m_new > -1e30f) ? __expf(m_regs[reg] - m_new) : 1.f;
+      const float pe = (m_new > -1e30f && sv[reg] > -1e29f)
+                           ? __expf(sv[reg] - m_new) : 0.f;
+      float lsum = pe;
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) lsum += __shfl_xor(lsum, off, WAVE);
+      l_regs[reg] = l_regs[reg] * alpha + lsum;
+      m_regs[reg] = m_new;
+      if (wid == 0 && col == 0) alpha_lds[row] = alpha;
+      p_lds[row * PD_KT + wid * 16 + col] =
+          __bfloat16_as_short(__float2bfloat16(pe));
+    }
+    __syncthreads();
+
+    // ---- OT[dims, 16q] += V^T P^T over this wave's dim slab
+    const float alpha_q = alpha_lds[col];
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
+      ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
+      const int dim = wid * D4 + dt * 16 + col;
+#pragma unroll
+      for (int ks = 0; ks < PD_KT / 32; ++ks) {
+        bf16x8_t a;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int key = ks * 32 + kgrp * 8 + j;
+          a[j] = v_lds[key * D + swz(key, dim & ~7) + (dim & 7)];
+        }
+        bf16x8_t bb = *reinterpret_cast<const bf16x8_t*>(
+            &p_lds[col * PD_KT + ks * 32 + kgrp * 8]);
+        ot[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bb, ot[dt], 0, 0, 0);
+      }
+    }
+    __syncthreads();  // v_lds/p_lds consumed; next chunk may overwrite
+  }
+
+  // ---- merge per-wave l, normalise, store this wave's dim slab
+  if (col == 0) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) l_lds[wid][kgrp * 4 + reg] = l_regs[reg];
+  }
+  __syncthreads();
+  const float l_tot = l_lds[0][col] + l_lds[1][col] + l_lds[2][col] + l_lds[3][col];
+  const float inv = (l_tot > 0.f) ? 1.0f / l_tot : 0.f;
+  if (col < G) {
+    __hip_bfloat16* op = out + (long)b * out_stride + (long)(kh * G + col) * D;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      const int dim0 = wid * D4 + dt * 16 + kgrp * 4;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg)
+        op[dim0 + reg] = __float2bfloat16(ot[dt][reg] * inv);
+    }
+  }
+}

@@ -174,9 +174,29 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
   const int bs = kc.size(2);
   const int max_blocks = bt.size(1);
   const int G = H / KVH;
-  TORCH_CHECK(H % KVH == 0 && G <= 8, "GQA group must be <= 8 (got ", G, ")");
+  TORCH_CHECK(H % KVH == 0, "num_heads must divide num_kv_heads");
   TORCH_CHECK(bs == 16 || bs == 32, "block_size must be 16 or 32");
   dim3 grid(B, KVH);
+  // bf16 + MFMA-supported head dims → matrix-core kernel (G padded to 16)
+  if constexpr (std::is_same_v<T, __hip_bfloat16>) {
+    if ((D == 128 || D == 256) && G <= 16) {
+      auto lm = [&]<int HD>() {
+        hipLaunchKernelGGL((paged_decode_mfma_kernel<HD>), grid, dim3(256), 0,
+                           stream(),
+                           reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                           reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                           reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()),
+                           reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()),
+                           bt.data_ptr<int>(), cl.data_ptr<int>(), H, KVH, bs,
+                           max_blocks, (float)scale, (float)softcap,
+                           (int)window, q.stride(0), out.stride(0));
+      };
+      if (D == 128) lm.template operator()<128>();
+      else lm.template operator()<256>();
+      return;
+    }
+  }
+  TORCH_CHECK(G <= 8, "GQA group must be <= 8 for the VALU decode kernel");
   const int subs = 256 / (32 * G);
   const int lds = (G * D + DECODE_CHUNK * G + subs * G * (D + 2)) * sizeof(float) +
                   16 * sizeof(int);

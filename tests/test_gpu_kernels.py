@@ -147,6 +147,25 @@ class TestDecodeAttention:
         atol, rtol = TOL[dtype]
         assert_close_to_f32_ref(out.cpu(), ref, atol, 5 * rtol)
 
+    @pytest.mark.parametrize("G,D", [(12, 128), (16, 256)])
+    def test_wide_gqa_group_bf16(self, G, D):
+        """G > 8 is only supported by the MFMA decode kernel (padded 16-row
+        Q tile); the VALU kernel caps at 8."""
+        dtype = torch.bfloat16
+        B, KVH, BS = 5, 2, 16
+        H = G * KVH
+        torch.manual_seed(3)
+        ctx = torch.tensor([1, 31, 64, 129, 333], dtype=torch.int32, device=DEV)
+        kc, vc, bt = _build_cache(B, KVH, D, BS, 333, dtype)
+        q = torch.randn(B, H, D, device=DEV, dtype=dtype)
+        scale = D ** -0.5
+        out = ops.paged_decode_attention(q, kc, vc, bt, ctx, scale)
+        ref = torch_ref.paged_decode_attention(
+            q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(), ctx.cpu(), scale
+        )
+        atol, rtol = TOL[dtype]
+        assert_close_to_f32_ref(out.cpu(), ref, atol, 5 * rtol)
+
     @pytest.mark.parametrize("softcap,window", [(50.0, 0), (0.0, 64), (50.0, 64)])
     def test_softcap_window(self, softcap, window):
         dtype = torch.bfloat16
