@@ -206,15 +206,18 @@ def spec_from_hf_config(path: Path, name: str) -> ModelSpec:
         attn_scale=(
             cfg.get("query_pre_attn_scalar") and 1.0 / math.sqrt(cfg["query_pre_attn_scalar"])
         ),
-        eos_token_id=_first(cfg.get("eos_token_id", 2)),
-        bos_token_id=_first(cfg.get("bos_token_id", 1)),
+        eos_token_id=_first(cfg.get("eos_token_id"), 2),
+        bos_token_id=_first(cfg.get("bos_token_id"), 1),
     )
     return spec
 
 
-def _first(x) -> int:
+def _first(x, default: int = 2) -> int:
+    # HF configs may carry an int, a list, or null (e.g. Qwen2 bos_token_id).
+    if x is None:
+        return default
     if isinstance(x, list):
-        return int(x[0]) if x else 2
+        return int(x[0]) if x else default
     return int(x)
 
 
