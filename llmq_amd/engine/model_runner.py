@@ -59,6 +59,11 @@ class ModelRunner:
         self._graph_pool = None
         self.sampling_generator = torch.Generator(device=device if device.type == "cuda" else "cpu")
         self.sampling_generator.manual_seed(config.seed)
+        self._sample_step = 0
+        self._sample_out: Optional[torch.Tensor] = None
+        self._sample_keys: Optional[torch.Tensor] = None
+        self._sample_temps: Optional[torch.Tensor] = None
+        self._sample_temps_h: Optional[torch.Tensor] = None
         if self.use_graphs:
             self._alloc_static_buffers()
 
@@ -225,6 +230,28 @@ class ModelRunner:
 
     def _sample(self, logits: torch.Tensor, seqs: List[Sequence]) -> torch.Tensor:
         dev = logits.device
+        self._sample_step += 1
+        simple = all(
+            s.params.top_k == 0 and s.params.top_p >= 1.0 for s in seqs
+        )
+        if simple and dev.type == "cuda":
+            # Fused one-pass HIP sampler (gumbel-max / greedy argmax).
+            B = len(seqs)
+            if self._sample_out is None or self._sample_out.numel() < B:
+                cap = max(B, self.config.max_num_seqs)
+                self._sample_out = torch.empty(cap, dtype=torch.int64, device=dev)
+                self._sample_keys = torch.empty(cap, dtype=torch.int64, device=dev)
+                self._sample_temps = torch.empty(cap, dtype=torch.float32, device=dev)
+                self._sample_temps_h = torch.empty(cap, dtype=torch.float32, pin_memory=True)
+            th = self._sample_temps_h
+            for i, s in enumerate(seqs):
+                th[i] = s.params.temperature
+            self._sample_temps[:B].copy_(th[:B], non_blocking=True)
+            ops.sample_gumbel_argmax(
+                self._sample_out[:B], self._sample_keys[:B], logits.float(),
+                self._sample_temps[:B], self.config.seed, self._sample_step,
+            )
+            return self._sample_out[:B]
         temps = torch.tensor([s.params.temperature for s in seqs], dtype=torch.float32, device=dev)
         tps = torch.tensor([s.params.top_p for s in seqs], dtype=torch.float32, device=dev)
         tks = torch.tensor([s.params.top_k for s in seqs], dtype=torch.int64, device=dev)

@@ -188,6 +188,22 @@ def varlen_prefill_attention(
 # ------------------------------------------------------------- sampling --
 
 
+def sample_gumbel_argmax(
+    out: torch.Tensor,      # [B] int64
+    keys: torch.Tensor,     # [B] int64 scratch (packed value|~index)
+    logits: torch.Tensor,   # [B, V] fp32
+    temps: torch.Tensor,    # [B] fp32 (<= 0 → greedy row)
+    seed: int,
+    step: int,
+) -> None:
+    """Fused one-pass sampler: per-row Gumbel-max (== softmax sampling) or
+    argmax for greedy rows. Counter-based RNG keyed on (seed, step) so
+    results are launch-geometry- and graph-independent."""
+    assert _use_hip(logits)
+    keys.zero_()
+    _EXT.sample_gumbel_argmax(out, keys, logits, temps, seed, step)
+
+
 def sample_tokens(
     logits: torch.Tensor,
     temperatures: torch.Tensor,
@@ -195,8 +211,8 @@ def sample_tokens(
     top_ks: torch.Tensor,
     generator: Optional[torch.Generator] = None,
 ) -> torch.Tensor:
-    # Vectorized torch path is used on GPU too for now (not a per-step
-    # bottleneck at moderate batch; a fused HIP sampler is planned).
+    # Generic path (top-k/top-p capable). The serving hot loop uses the
+    # fused sample_gumbel_argmax above when no top-k/top-p is requested.
     if logits.is_cuda:
         return _sample_tokens_gpu(logits, temperatures, top_ps, top_ks, generator)
     return torch_ref.sample_tokens(logits, temperatures, top_ps, top_ks, generator)

@@ -294,6 +294,29 @@ void varlen_prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,
   });
 }
 
+// ------------------------------------------------------------- sampling --
+
+void sample_gumbel_argmax(at::Tensor out, at::Tensor keys, at::Tensor logits,
+                          at::Tensor temps, long seed, long step) {
+  CHECK_GPU(logits);
+  TORCH_CHECK(logits.scalar_type() == at::kFloat && logits.dim() == 2);
+  TORCH_CHECK(out.scalar_type() == at::kLong);
+  TORCH_CHECK(keys.scalar_type() == at::kLong && keys.numel() == logits.size(0));
+  const int B = logits.size(0), V = logits.size(1);
+  // enough splits to fill the chip at small B
+  int nsplit = 1;
+  while (B * nsplit < 2048 && nsplit < 64 && (V / nsplit) > 4096) nsplit *= 2;
+  hipLaunchKernelGGL(sample_argmax_kernel, dim3(B, nsplit), dim3(256), 0,
+                     stream(),
+                     reinterpret_cast<unsigned long long*>(keys.data_ptr()),
+                     logits.data_ptr<float>(), temps.data_ptr<float>(), V,
+                     (unsigned int)seed, (unsigned int)step);
+  hipLaunchKernelGGL(unpack_keys_kernel, dim3((B + 255) / 256), dim3(256), 0,
+                     stream(), out.data_ptr<long>(),
+                     reinterpret_cast<const unsigned long long*>(keys.data_ptr()),
+                     B);
+}
+
 }  // namespace
 
 TORCH_LIBRARY(llmq_amd, m) {
@@ -305,6 +328,7 @@ TORCH_LIBRARY(llmq_amd, m) {
   m.def("reshape_and_cache(Tensor key, Tensor value, Tensor(a!) k_cache, Tensor(b!) v_cache, Tensor slot_mapping) -> ()");
   m.def("paged_decode_attention(Tensor(a!) out, Tensor q, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor context_lens, float scale, float softcap, int window) -> ()");
   m.def("varlen_prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, int max_seqlen, float scale, float softcap, int window) -> ()");
+  m.def("sample_gumbel_argmax(Tensor(a!) out, Tensor(b!) keys, Tensor logits, Tensor temps, int seed, int step) -> ()");
 }
 
 TORCH_LIBRARY_IMPL(llmq_amd, CUDA, m) {
@@ -316,4 +340,5 @@ TORCH_LIBRARY_IMPL(llmq_amd, CUDA, m) {
   m.impl("reshape_and_cache", &reshape_and_cache);
   m.impl("paged_decode_attention", &paged_decode_attention);
   m.impl("varlen_prefill_attention", &varlen_prefill_attention);
+  m.impl("sample_gumbel_argmax", &sample_gumbel_argmax);
 }

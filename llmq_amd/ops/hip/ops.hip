@@ -3,4 +3,5 @@
 #include "elementwise.hip"
 #include "kvcache.hip"
 #include "attention.hip"
+#include "sampling.hip"
 #include "bindings.cpp"

@@ -1,0 +1,92 @@
+// Fused token sampling for gfx950.
+//
+// One pass over the fp32 logits [B, V]: each workgroup scans a vocab chunk
+// of one row and computes a local argmax of
+//   temp <= 0:  logit                      (greedy)
+//   temp  > 0:  logit / temp + Gumbel(g)   (Gumbel-max == softmax sampling)
+// then publishes via one atomicMax on a packed (orderable-f32 | index) key.
+// Replaces the torch sampler's 4+ full passes (softmax, multinomial, ...)
+// with a single logits read — at B=256, V=256k fp32 that is 262 MB ≈ 35 µs
+// at stream rate. Top-k/top-p rows fall back to the torch path (wrapper).
+//
+// RNG: counter-based (seed, step, element) hash — deterministic per engine
+// step, independent of launch geometry, so eager and hipGraph runs sample
+// identical tokens.
+
+#include "common.h"
+
+DEVINL unsigned int hash_u32(unsigned int x) {
+  // finalizer-strength integer hash (xxhash/murmur-style avalanche)
+  x ^= x >> 16; x *= 0x7feb352dU;
+  x ^= x >> 15; x *= 0x846ca68bU;
+  x ^= x >> 16;
+  return x;
+}
+
+DEVINL float uniform01(unsigned int seed, unsigned int step, unsigned int b,
+                       unsigned int v) {
+  unsigned int h = hash_u32(seed ^ hash_u32(step ^ hash_u32((b << 20) ^ v)));
+  // (0, 1]: avoid 0 so log() is finite
+  return (h + 1u) * (1.0f / 4294967296.0f);
+}
+
+DEVINL unsigned long long pack_key(float val, int idx) {
+  unsigned int u = __float_as_uint(val);
+  u = (u & 0x80000000u) ? ~u : (u | 0x80000000u);  // order-preserving map
+  // index bits inverted so equal values tie-break toward the LOWER index
+  // (matches torch argmax) under atomicMax
+  return ((unsigned long long)u << 32) | (unsigned int)(~idx);
+}
+
+__global__ __launch_bounds__(256) void sample_argmax_kernel(
+    unsigned long long* __restrict__ out_keys,  // [B], pre-zeroed
+    const float* __restrict__ logits,           // [B, V]
+    const float* __restrict__ temps,            // [B]
+    int V, unsigned int seed, unsigned int step) {
+  const int b = blockIdx.x;
+  const int nsplit = gridDim.y;
+  const int chunk = (V + nsplit - 1) / nsplit;
+  const int v0 = blockIdx.y * chunk;
+  const int v1 = min(v0 + chunk, V);
+  const float temp = temps[b];
+  const bool greedy = temp <= 0.f;
+  const float inv_t = greedy ? 1.0f : 1.0f / temp;
+  const float* row = logits + (long)b * V;
+
+  float best = -3e38f;
+  int best_i = v0;
+  for (int v = v0 + threadIdx.x; v < v1; v += 256) {
+    float x = row[v] * inv_t;
+    if (!greedy) {
+      const float u = uniform01(seed, step, b, v);
+      x += -__logf(-__logf(u));  // Gumbel(0,1)
+    }
+    if (x > best) { best = x; best_i = v; }
+  }
+  // wave reduce (value, index)
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    const float ov = __shfl_xor(best, off, WAVE);
+    const int oi = __shfl_xor(best_i, off, WAVE);
+    if (ov > best || (ov == best && oi < best_i)) { best = ov; best_i = oi; }
+  }
+  __shared__ float vals[4];
+  __shared__ int idxs[4];
+  const int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) { vals[wid] = best; idxs[wid] = best_i; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < 4; ++w) {
+      if (vals[w] > best || (vals[w] == best && idxs[w] < best_i)) {
+        best = vals[w]; best_i = idxs[w];
+      }
+    }
+    atomicMax(out_keys + b, pack_key(best, best_i));
+  }
+}
+
+__global__ void unpack_keys_kernel(long* __restrict__ out,
+                                   const unsigned long long* __restrict__ keys,
+                                   int B) {
+  const int b = blockIdx.x * blockDim.x + threadIdx.x;
+  if (b < B) out[b] = (long)(unsigned int)(~(keys[b] & 0xffffffffu));
+}

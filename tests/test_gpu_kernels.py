@@ -262,3 +262,47 @@ class TestEngineGPU:
         # First sampled token comes from the SAME prefill path in both modes.
         for e, g in zip(eager_out, graph_out):
             assert e[0] == g[0]
+
+
+class TestFusedSampler:
+    def _run(self, logits, temps, seed=7, step=3):
+        B = logits.shape[0]
+        out = torch.empty(B, dtype=torch.int64, device=DEV)
+        keys = torch.empty(B, dtype=torch.int64, device=DEV)
+        ops.sample_gumbel_argmax(out, keys, logits, temps, seed, step)
+        return out
+
+    def test_greedy_matches_argmax(self):
+        torch.manual_seed(0)
+        logits = torch.randn(16, 50000, device=DEV, dtype=torch.float32)
+        temps = torch.zeros(16, device=DEV)
+        out = self._run(logits, temps)
+        assert torch.equal(out, logits.argmax(dim=-1))
+
+    def test_deterministic_per_step(self):
+        torch.manual_seed(0)
+        logits = torch.randn(8, 4096, device=DEV)
+        temps = torch.full((8,), 0.7, device=DEV)
+        a = self._run(logits, temps, seed=1, step=5)
+        b = self._run(logits, temps, seed=1, step=5)
+        c = self._run(logits, temps, seed=1, step=6)
+        assert torch.equal(a, b)
+        assert not torch.equal(a, c)  # different step → different draws
+
+    def test_distribution_matches_softmax(self):
+        """Gumbel-max sampling must match the softmax distribution."""
+        torch.manual_seed(0)
+        V = 8
+        logits_row = torch.tensor([2.0, 1.0, 0.0, -1.0, 3.0, 0.5, -2.0, 1.5],
+                                  device=DEV)
+        temp = 0.8
+        N = 20000
+        logits = logits_row.expand(N, V).contiguous()
+        temps = torch.full((N,), temp, device=DEV)
+        out = self._run(logits, temps, seed=123, step=1)
+        counts = torch.bincount(out.cpu(), minlength=V).float()
+        expected = torch.softmax(logits_row.cpu() / temp, dim=-1) * N
+        # chi-square-ish check: every bucket within 5 sigma
+        sigma = (expected.clamp_min(1.0)).sqrt()
+        assert ((counts - expected).abs() < 5 * sigma + 10).all(), (
+            counts, expected)
