@@ -512,22 +512,21 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
 // Single-token GQA decode attention on matrix cores (bf16, D in {128,256},
 // G <= 16). One workgroup per (sequence, kv_head); the GQA query group is
 // padded to a 16-row MFMA tile (pad rows repeat q-head 0 and are never
-// stored). Per 64-key chunk, the 4 waves cooperate:
-//   stage: all 256 threads gather the chunk's K and V rows from the paged
-//          cache into XOR-swizzled LDS (zero-filled beyond L)
-//   S:     wave w computes S[16q, 16keys] for key slab w via
-//          mfma_f32_16x16x32_bf16; row-max partials exchanged through LDS
-//   P:     every wave combines the 4 partial maxes identically (shared
-//          m_run), builds its P slab in LDS (bf16), keeps a PER-WAVE l_run
-//          (summed once at the end — avoids a third stats barrier)
-//   PV:    wave w owns dim slab [w*D/4, (w+1)*D/4): OT[dims,16q] +=
-//          mfma(A = V^T frags, B = P^T b128 reads); the online-softmax
-//          rescale factor is lane-uniform (qrow = lane&15), read from LDS.
-// The decode-v2 kernel above is VALU-issue-bound (~120 cyc/token/wave:
-// scalar dot + exp + per-token addressing); this one moves QK^T and PV onto
-// MFMA (~6 cyc/token/wave) so the kernel runs at the KV HBM stream rate.
+// stored). K and V SHARE one LDS tile buffer: per 64-key chunk
+//   stage K -> sync -> S = QK^T (wave w: 16-key slab, MFMA 16x16x32) ->
+//   sync -> stage V into the same buffer (its HBM latency hides under the
+//   softmax VALU work) + combine row maxes / build P / update per-wave l ->
+//   sync -> PV: wave w owns dim slab [w*D/4,(w+1)*D/4): OT[dims,16q] +=
+//   mfma(A = V^T frags, B = P^T b128 reads) -> sync.
+// The block table is staged to LDS once up front (the per-load global
+// bt[] read was a dependent-latency chain on every staging address).
+// Sharing the tile keeps LDS ~36 KB -> 4 workgroups/CU (the v1 split-K/V
+// layout was 68 KB -> 2 WGs/CU and ran SLOWER than the VALU kernel).
+// The VALU kernel above is issue-bound (~120 cyc/token/wave); this one
+// targets the KV HBM stream rate.
 
-#define PD_KT 64  // keys per chunk
+#define PD_KT 64       // keys per chunk
+#define PD_MAX_BT 1024  // max staged block-table entries (host falls back past this)
 
 template <int HEAD_DIM>
 __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
@@ -555,12 +554,20 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
   const int col = lane & 15;   // MFMA col lane (key for S, qrow for OT)
   const int kgrp = lane >> 4;  // 0..3
 
-  __shared__ __attribute__((aligned(16))) short k_lds[PD_KT * D];
-  __shared__ __attribute__((aligned(16))) short v_lds[PD_KT * D];
+  __shared__ __attribute__((aligned(16))) short kv_lds[PD_KT * D];  // K then V
   __shared__ __attribute__((aligned(16))) short p_lds[16 * PD_KT];  // [qrow][key]
   __shared__ float mpart_lds[4][16];  // per-wave row-max partials
   __shared__ float alpha_lds[16];     // per-row rescale for the OT lanes
   __shared__ float l_lds[4][16];      // per-wave l_run (end merge)
+  __shared__ int bt_lds[PD_MAX_BT];
+
+  // ---- stage the whole block table once (removes a dependent global load
+  // from every staging address)
+  {
+    const int nb = (L + block_size - 1) / block_size;
+    const int* bt = block_tables + (long)b * max_blocks;
+    for (int i = tid; i < nb; i += 256) bt_lds[i] = bt[i];
+  }
 
   // ---- Q fragments: A-operand rows = padded q rows (row = lane&15)
   bf16x8_t qfrag[KS];
@@ -580,31 +587,26 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
   for (int dt = 0; dt < DT; ++dt) ot[dt] = {0.f, 0.f, 0.f, 0.f};
 
   const int start = (window > 0 && L > window) ? (L - window) : 0;
-  const int* bt = block_tables + (long)b * max_blocks;
   const int base0 = (start / PD_KT) * PD_KT;
+  constexpr int CPK = D / 8;    // 16B chunks per key row
+  constexpr int NCK = PD_KT * CPK;
+  __syncthreads();  // bt_lds ready
 
   for (int base = base0; base < L; base += PD_KT) {
-    // ---- stage K/V chunk (gather via block table; zeros beyond L)
-    {
-      constexpr int CPK = D / 8;        // 16B chunks per key row
-      const int nck = PD_KT * CPK;      // chunks per tile
-      for (int c = tid; c < nck; c += 256) {
-        const int key = c / CPK;
-        const int d8 = (c % CPK) * 8;
-        const int dst = key * D + swz(key, d8);
-        const int gkey = base + key;
-        if (gkey < L) {
-          const int blk = bt[gkey / block_size];
-          const long rowoff =
-              (((long)blk * num_kv_heads + kh) * block_size + gkey % block_size) * D + d8;
-          *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) =
-              *reinterpret_cast<const bf16x8_t*>(k_cache + rowoff);
-          *reinterpret_cast<bf16x8_t*>(&v_lds[dst]) =
-              *reinterpret_cast<const bf16x8_t*>(v_cache + rowoff);
-        } else {
-          *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
-          *reinterpret_cast<bf16x8_t*>(&v_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
-        }
+    // ---- stage K chunk (gather via LDS block table; zeros beyond L)
+    for (int c = tid; c < NCK; c += 256) {
+      const int key = c / CPK;
+      const int d8 = (c % CPK) * 8;
+      const int dst = key * D + swz(key, d8);
+      const int gkey = base + key;
+      if (gkey < L) {
+        const long rowoff =
+            (((long)bt_lds[gkey / block_size] * num_kv_heads + kh) * block_size +
+             gkey % block_size) * D + d8;
+        *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) =
+            *reinterpret_cast<const bf16x8_t*>(k_cache + rowoff);
+      } else {
+        *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
       }
     }
     __syncthreads();
@@ -617,29 +619,48 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
       for (int ks = 0; ks < KS; ++ks) {
         const int d8 = ks * 32 + kgrp * 8;
         bf16x8_t bfrag =
-            *reinterpret_cast<const bf16x8_t*>(&k_lds[key * D + swz(key, d8)]);
+            *reinterpret_cast<const bf16x8_t*>(&kv_lds[key * D + swz(key, d8)]);
         s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks], bfrag, s, 0, 0, 0);
       }
     }
     // scale, softcap, bounds mask; rows r = kgrp*4 + reg
     float sv[4], mx[4];
-#pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
+    {
       const int key = base + wid * 16 + col;
-      float x = s[reg] * scale;
-      if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
       const bool dead = key >= L || key < start;
-      sv[reg] = dead ? -1e30f : x;
-      float m = sv[reg];
 #pragma unroll
-      for (int off = 1; off < 16; off <<= 1) m = fmaxf(m, __shfl_xor(m, off, WAVE));
-      mx[reg] = m;  // this wave's slab max for row reg
+      for (int reg = 0; reg < 4; ++reg) {
+        float x = s[reg] * scale;
+        if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
+        sv[reg] = dead ? -1e30f : x;
+        float m = sv[reg];
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) m = fmaxf(m, __shfl_xor(m, off, WAVE));
+        mx[reg] = m;
+      }
     }
     if (col == 0) {
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) mpart_lds[wid][kgrp * 4 + reg] = mx[reg];
     }
-    __syncthreads();
+    __syncthreads();  // mparts ready; every wave is past its K reads
+
+    // ---- stage V into the SAME buffer; its latency hides under softmax VALU
+    for (int c = tid; c < NCK; c += 256) {
+      const int key = c / CPK;
+      const int d8 = (c % CPK) * 8;
+      const int dst = key * D + swz(key, d8);
+      const int gkey = base + key;
+      if (gkey < L) {
+        const long rowoff =
+            (((long)bt_lds[gkey / block_size] * num_kv_heads + kh) * block_size +
+             gkey % block_size) * D + d8;
+        *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) =
+            *reinterpret_cast<const bf16x8_t*>(v_cache + rowoff);
+      } else {
+        *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+    }
 
     // ---- combine maxes (identical on every wave), build P, update l
 #pragma unroll
@@ -660,7 +681,7 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
       p_lds[row * PD_KT + wid * 16 + col] =
           __bfloat16_as_short(__float2bfloat16(pe));
     }
-    __syncthreads();
+    __syncthreads();  // V + P + alpha ready
 
     // ---- OT[dims, 16q] += V^T P^T over this wave's dim slab
     const float alpha_q = alpha_lds[col];
@@ -675,14 +696,14 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const int key = ks * 32 + kgrp * 8 + j;
-          a[j] = v_lds[key * D + swz(key, dim & ~7) + (dim & 7)];
+          a[j] = kv_lds[key * D + swz(key, dim & ~7) + (dim & 7)];
         }
         bf16x8_t bb = *reinterpret_cast<const bf16x8_t*>(
             &p_lds[col * PD_KT + ks * 32 + kgrp * 8]);
         ot[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bb, ot[dt], 0, 0, 0);
       }
     }
-    __syncthreads();  // v_lds/p_lds consumed; next chunk may overwrite
+    __syncthreads();  // V/P consumed; next chunk may overwrite
   }
 
   // ---- merge per-wave l, normalise, store this wave's dim slab
