@@ -49,13 +49,15 @@ def main() -> None:
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     distributed = world > 1
 
+    use_gpu = torch.cuda.is_available()
     if distributed:
         import torch.distributed as dist
 
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group(backend="nccl")
+        if use_gpu:
+            torch.cuda.set_device(local_rank)
+        # "nccl" IS RCCL on ROCm; gloo covers the CPU test path.
+        dist.init_process_group(backend="nccl" if use_gpu else "gloo")
 
-    use_gpu = torch.cuda.is_available()
     device = f"cuda:{local_rank}" if use_gpu else "cpu"
 
     from llmq_amd.engine.config import EngineConfig
