@@ -29,6 +29,32 @@ logger = logging.getLogger(__name__)
 # Memory the engine leaves free on top of weights+KV (workspace, graphs).
 _RESERVE_BYTES = 4 << 30
 
+_TUNED_GEMMS_DONE = False
+
+
+def _enable_tuned_gemms() -> None:
+    """Load the shipped TunableOp results (hipBLASLt/rocBLAS algorithm
+    selections tuned on MI355X for the serving GEMM shapes). No-op when the
+    user drives TunableOp via PYTORCH_TUNABLEOP_* env vars themselves."""
+    global _TUNED_GEMMS_DONE
+    if _TUNED_GEMMS_DONE or os.environ.get("PYTORCH_TUNABLEOP_ENABLED"):
+        return
+    _TUNED_GEMMS_DONE = True
+    from pathlib import Path
+
+    csv = Path(__file__).parent.parent / "ops" / "tunableop_gfx950.csv"
+    if not csv.is_file():
+        return
+    try:
+        import torch.cuda.tunable as tunable
+
+        tunable.enable(True)
+        tunable.tuning_enable(False)
+        tunable.read_file(str(csv))
+        logger.info("TunableOp GEMM selections loaded from %s", csv.name)
+    except Exception as exc:  # noqa: BLE001 — tuning is an optimisation only
+        logger.warning("TunableOp setup failed (%s); using default GEMMs", exc)
+
 
 @dataclass(slots=True)
 class RequestOutput:
@@ -56,6 +82,8 @@ class LLMEngine:
         if self.device.type == "cuda" and not ops.has_hip_ext():
             # Fail loudly: GPU execution must run the CDNA4 kernels.
             ops._use_hip(torch.empty(1, device=self.device))
+        if self.device.type == "cuda":
+            _enable_tuned_gemms()
         self.max_model_len = min(
             config.max_model_len or self.spec.max_position_embeddings,
             self.spec.max_position_embeddings,
