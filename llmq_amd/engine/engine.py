@@ -8,6 +8,7 @@ The async facade (async_engine.py) drives this loop on a dedicated thread.
 from __future__ import annotations
 
 import logging
+import os
 import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
