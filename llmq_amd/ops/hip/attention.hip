@@ -529,7 +529,7 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
 #define PD_MAX_BT 1024  // max staged block-table entries (host falls back past this)
 
 template <int HEAD_DIM>
-__global__ __launch_bounds__(256, 3) void paged_decode_mfma_kernel(
+__global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
     __hip_bfloat16* __restrict__ out,      // [B, H, D]
     const __hip_bfloat16* __restrict__ q,  // [B, H, D]
     const __hip_bfloat16* __restrict__ k_cache,  // [nb, KVH, bs, D]
@@ -590,43 +590,25 @@ __global__ __launch_bounds__(256, 3) void paged_decode_mfma_kernel(
   const int base0 = (start / PD_KT) * PD_KT;
   constexpr int CPK = D / 8;    // 16B chunks per key row
   constexpr int NCK = PD_KT * CPK;
-  constexpr int RPT = NCK / 256;  // 16B pieces per thread per tile
   __syncthreads();  // bt_lds ready
 
-  // ---- K prefetch registers: piece r covers element chunk c = r*256 + tid
-  bf16x8_t kreg[RPT];
-  auto issue_k = [&](int base) {
-#pragma unroll
-    for (int r = 0; r < RPT; ++r) {
-      const int c = r * 256 + tid;
+  for (int base = base0; base < L; base += PD_KT) {
+    // ---- stage K chunk (gather via LDS block table; zeros beyond L)
+    for (int c = tid; c < NCK; c += 256) {
       const int key = c / CPK;
+      const int d8 = (c % CPK) * 8;
+      const int dst = key * D + swz(key, d8);
       const int gkey = base + key;
       if (gkey < L) {
-        const int d8 = (c % CPK) * 8;
         const long rowoff =
             (((long)bt_lds[gkey / block_size] * num_kv_heads + kh) * block_size +
              gkey % block_size) * D + d8;
-        kreg[r] = *reinterpret_cast<const bf16x8_t*>(k_cache + rowoff);
+        *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) =
+            *reinterpret_cast<const bf16x8_t*>(k_cache + rowoff);
       } else {
-        kreg[r] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+        *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
       }
     }
-  };
-  auto write_k = [&]() {
-#pragma unroll
-    for (int r = 0; r < RPT; ++r) {
-      const int c = r * 256 + tid;
-      const int key = c / CPK;
-      const int d8 = (c % CPK) * 8;
-      *reinterpret_cast<bf16x8_t*>(&kv_lds[key * D + swz(key, d8)]) = kreg[r];
-    }
-  };
-
-  issue_k(base0);
-
-  for (int base = base0; base < L; base += PD_KT) {
-    // ---- K tile: registers (prefetched during the previous PV) -> LDS
-    write_k();
     __syncthreads();
 
     // ---- S[16, 16] for this wave's key slab
@@ -721,9 +703,6 @@ __global__ __launch_bounds__(256, 3) void paged_decode_mfma_kernel(
         ot[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bb, ot[dt], 0, 0, 0);
       }
     }
-    // prefetch the next chunk's K while PV's MFMAs drain (the loads only
-    // need to land before the post-barrier write_k)
-    if (base + PD_KT < L) issue_k(base + PD_KT);
     __syncthreads();  // V/P consumed; next chunk may overwrite
   }
 
