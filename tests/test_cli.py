@@ -1,0 +1,143 @@
+"""CLI surface tests via click's CliRunner against a live in-process broker.
+
+The reference has NO tests for submit/receive/monitor CLI —
+these exceed it: submit from a JSONL file with --map templating, receive to
+stdout JSONL, status/health/errors/clear against real queues.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import threading
+
+import pytest
+from click.testing import CliRunner
+
+from llmq_amd.cli.main import cli
+from llmq_amd.core.client import BrokerClient
+from llmq_amd.core.config import Config
+from llmq_amd.core.models import Job, Result
+
+pytestmark = pytest.mark.integration
+
+
+@pytest.fixture()
+def broker_env(monkeypatch, tmp_path):
+    """A live broker in a background thread + LLMQ_BROKER_URL env."""
+    from llmq_amd.broker.server import BrokerServer
+
+    loop = asyncio.new_event_loop()
+    server = BrokerServer("127.0.0.1", 0, data_dir=None, max_retries=2)
+    started = threading.Event()
+
+    def run():
+        asyncio.set_event_loop(loop)
+        loop.run_until_complete(server.serve())
+        started.set()
+        loop.run_forever()
+
+    t = threading.Thread(target=run, daemon=True)
+    t.start()
+    started.wait(10)
+    url = f"llmq://127.0.0.1:{server.port}"
+    monkeypatch.setenv("LLMQ_BROKER_URL", url)
+    yield url
+    loop.call_soon_threadsafe(loop.stop)
+    t.join(timeout=5)
+
+
+def _client(url: str) -> BrokerClient:
+    return BrokerClient(Config(broker_url=url))
+
+
+def test_submit_and_status_and_clear(broker_env, tmp_path):
+    jobs = tmp_path / "jobs.jsonl"
+    jobs.write_text("\n".join(
+        json.dumps({"id": f"c{i}", "text": f"hello {i}"}) for i in range(7)
+    ))
+    runner = CliRunner()
+    res = runner.invoke(cli, ["submit", "cliq", str(jobs),
+                              "--template", "Say: {text}"])
+    assert res.exit_code == 0, res.output
+    assert "7" in res.output
+
+    res = runner.invoke(cli, ["status", "cliq"])
+    assert res.exit_code == 0, res.output
+    assert "cliq" in res.output
+
+    res = runner.invoke(cli, ["health", "cliq"])
+    # no workers consuming → unhealthy exit code 1 by design
+    assert res.exit_code == 1 and "UNHEALTHY" in res.output, res.output
+
+    res = runner.invoke(cli, ["clear", "cliq", "-y"])
+    assert res.exit_code == 0, res.output
+
+    async def check():
+        c = _client(broker_env)
+        await c.connect()
+        stats = await c.get_queue_stats("cliq")
+        await c.disconnect()
+        return stats
+
+    stats = asyncio.new_event_loop().run_until_complete(check())
+    assert stats.message_count == 0
+
+
+def test_receive_drains_results(broker_env):
+    async def seed():
+        c = _client(broker_env)
+        await c.connect()
+        await c.setup_queue_infrastructure("clir")
+        for i in range(3):
+            await c.publish_result("clir", Result(
+                id=f"r{i}", prompt="p", result=f"out {i}",
+                worker_id="w", duration_ms=1.0,
+            ))
+        await c.disconnect()
+
+    asyncio.new_event_loop().run_until_complete(seed())
+    runner = CliRunner()
+    res = runner.invoke(cli, ["receive", "clir", "--timeout", "2"])
+    assert res.exit_code == 0, res.output
+    lines = [json.loads(l) for l in res.output.splitlines() if l.startswith("{")]
+    assert {r["id"] for r in lines} == {"r0", "r1", "r2"}
+
+
+def test_errors_shows_dead_letters(broker_env):
+    async def seed():
+        c = _client(broker_env)
+        await c.connect()
+        await c.setup_queue_infrastructure("clie")
+        await c.publish_jobs("clie", [Job(id="bad", prompt="x")])
+        # consume and dead-letter it
+        done = asyncio.Event()
+
+        async def cb(delivery):
+            await delivery.nack(requeue=False, error="boom", worker="w-test")
+            done.set()
+
+        await c.consume_jobs("clie", cb, prefetch=1)
+        await asyncio.wait_for(done.wait(), 5)
+        await c.disconnect()
+
+    asyncio.new_event_loop().run_until_complete(seed())
+    runner = CliRunner()
+    res = runner.invoke(cli, ["errors", "clie"])
+    assert res.exit_code == 0, res.output
+    assert "boom" in res.output
+
+
+def test_status_all_queues(broker_env):
+    async def seed():
+        c = _client(broker_env)
+        await c.connect()
+        await c.setup_queue_infrastructure("q-one")
+        await c.setup_queue_infrastructure("q-two")
+        await c.disconnect()
+
+    asyncio.new_event_loop().run_until_complete(seed())
+    runner = CliRunner()
+    res = runner.invoke(cli, ["status"])
+    assert res.exit_code == 0, res.output
+    assert "q-one" in res.output and "q-two" in res.output
