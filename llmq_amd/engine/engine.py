@@ -2,7 +2,8 @@
 
 One ``step()`` runs one scheduler iteration (a prefill batch or a decode
 batch), samples, advances sequences, and returns per-request progress.
-The async facade (async_engine.py) drives this loop on a dedicated thread.
+The async facade (workers/engine_worker.py AsyncEngineBridge) drives this
+loop on a dedicated engine thread.
 """
 
 from __future__ import annotations
