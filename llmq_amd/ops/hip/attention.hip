@@ -561,12 +561,14 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
   __shared__ float l_lds[4][16];      // per-wave l_run (end merge)
   __shared__ int bt_lds[PD_MAX_BT];
 
-  // ---- stage the whole block table once (removes a dependent global load
-  // from every staging address)
+  // ---- stage the block table once (removes a dependent global load from
+  // every staging address); contexts past PD_MAX_BT blocks (16k tokens at
+  // bs 16) read the tail entries from global instead
+  const int* bt_glob = block_tables + (long)b * max_blocks;
   {
     const int nb = (L + block_size - 1) / block_size;
-    const int* bt = block_tables + (long)b * max_blocks;
-    for (int i = tid; i < nb; i += 256) bt_lds[i] = bt[i];
+    const int nstage = nb < PD_MAX_BT ? nb : PD_MAX_BT;
+    for (int i = tid; i < nstage; i += 256) bt_lds[i] = bt_glob[i];
   }
 
   // ---- Q fragments: A-operand rows = padded q rows (row = lane&15)
@@ -600,9 +602,10 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
       const int dst = key * D + swz(key, d8);
       const int gkey = base + key;
       if (gkey < L) {
+        const int bidx = gkey / block_size;
+        const long blk = (bidx < PD_MAX_BT) ? bt_lds[bidx] : bt_glob[bidx];
         const long rowoff =
-            (((long)bt_lds[gkey / block_size] * num_kv_heads + kh) * block_size +
-             gkey % block_size) * D + d8;
+            ((blk * num_kv_heads + kh) * block_size + gkey % block_size) * D + d8;
         *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) =
             *reinterpret_cast<const bf16x8_t*>(k_cache + rowoff);
       } else {
@@ -652,9 +655,10 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
       const int dst = key * D + swz(key, d8);
       const int gkey = base + key;
       if (gkey < L) {
+        const int bidx = gkey / block_size;
+        const long blk = (bidx < PD_MAX_BT) ? bt_lds[bidx] : bt_glob[bidx];
         const long rowoff =
-            (((long)bt_lds[gkey / block_size] * num_kv_heads + kh) * block_size +
-             gkey % block_size) * D + d8;
+            ((blk * num_kv_heads + kh) * block_size + gkey % block_size) * D + d8;
         *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) =
             *reinterpret_cast<const bf16x8_t*>(v_cache + rowoff);
       } else {

@@ -179,7 +179,7 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
   dim3 grid(B, KVH);
   // bf16 + MFMA-supported head dims → matrix-core kernel (G padded to 16)
   if constexpr (std::is_same_v<T, __hip_bfloat16>) {
-    if ((D == 128 || D == 256) && G <= 16 && max_blocks <= 1024) {
+    if ((D == 128 || D == 256) && G <= 16) {
       auto lm = [&]<int HD>() {
         hipLaunchKernelGGL((paged_decode_mfma_kernel<HD>), grid, dim3(256), 0,
                            stream(),
