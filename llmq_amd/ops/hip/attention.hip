@@ -618,7 +618,6 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
     f32x4_t s = {0.f, 0.f, 0.f, 0.f};
     {
       const int key = wid * 16 + col;
-      __builtin_amdgcn_s_setprio(1);  // keep MFMA waves ahead of stagers
 #pragma unroll
       for (int ks = 0; ks < KS; ++ks) {
         const int d8 = ks * 32 + kgrp * 8;
@@ -626,7 +625,6 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
             *reinterpret_cast<const bf16x8_t*>(&kv_lds[key * D + swz(key, d8)]);
         s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks], bfrag, s, 0, 0, 0);
       }
-      __builtin_amdgcn_s_setprio(0);
     }
     // scale, softcap, bounds mask; rows r = kgrp*4 + reg
     float sv[4], mx[4];
@@ -691,7 +689,6 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
 
     // ---- OT[dims, 16q] += V^T P^T over this wave's dim slab
     const float alpha_q = alpha_lds[col];
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt) {
       ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
@@ -710,7 +707,6 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
         ot[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bb, ot[dt], 0, 0, 0);
       }
     }
-    __builtin_amdgcn_s_setprio(0);
     __syncthreads();  // V/P consumed; next chunk may overwrite
   }
 
