@@ -525,11 +525,12 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
 // The VALU kernel above is issue-bound (~120 cyc/token/wave); this one
 // targets the KV HBM stream rate.
 
-#define PD_KT 64       // keys per chunk
-#define PD_MAX_BT 1024  // max staged block-table entries (host falls back past this)
+#define PD_MAX_BT 1024  // block-table entries staged in LDS (tail from global)
 
-template <int HEAD_DIM>
-__global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
+// NW = waves per workgroup (4 -> 256 threads/64-key chunks; 8 -> 512
+// threads/128-key chunks: same waves/SIMD at half the barriers per token).
+template <int HEAD_DIM, int NW>
+__global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
     __hip_bfloat16* __restrict__ out,      // [B, H, D]
     const __hip_bfloat16* __restrict__ q,  // [B, H, D]
     const __hip_bfloat16* __restrict__ k_cache,  // [nb, KVH, bs, D]
@@ -539,9 +540,11 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
     int num_heads, int num_kv_heads, int block_size, int max_blocks,
     float scale, float softcap, int window, long q_stride, long out_stride) {
   constexpr int D = HEAD_DIM;
-  constexpr int KS = D / 32;   // MFMA K-steps over the head dim
-  constexpr int D4 = D / 4;    // dim slab per wave
-  constexpr int DT = D4 / 16;  // 16-dim MFMA tiles per wave
+  constexpr int KS = D / 32;      // MFMA K-steps over the head dim
+  constexpr int PD_KT = NW * 16;  // keys per chunk
+  constexpr int NT = NW * WAVE;   // threads per workgroup
+  constexpr int D4 = D / NW;      // dim slab per wave
+  constexpr int DT = D4 / 16;     // 16-dim MFMA tiles per wave
   const int b = blockIdx.x;
   const int kh = blockIdx.y;
   const int G = num_heads / num_kv_heads;  // <= 16
@@ -556,9 +559,9 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
 
   __shared__ __attribute__((aligned(16))) short kv_lds[PD_KT * D];  // K then V
   __shared__ __attribute__((aligned(16))) short p_lds[16 * PD_KT];  // [qrow][key]
-  __shared__ float mpart_lds[4][16];  // per-wave row-max partials
-  __shared__ float alpha_lds[16];     // per-row rescale for the OT lanes
-  __shared__ float l_lds[4][16];      // per-wave l_run (end merge)
+  __shared__ float mpart_lds[NW][16];  // per-wave row-max partials
+  __shared__ float alpha_lds[16];      // per-row rescale for the OT lanes
+  __shared__ float l_lds[NW][16];      // per-wave l_run (end merge)
   __shared__ int bt_lds[PD_MAX_BT];
 
   // ---- stage the block table once (removes a dependent global load from
@@ -568,7 +571,7 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
   {
     const int nb = (L + block_size - 1) / block_size;
     const int nstage = nb < PD_MAX_BT ? nb : PD_MAX_BT;
-    for (int i = tid; i < nstage; i += 256) bt_lds[i] = bt_glob[i];
+    for (int i = tid; i < nstage; i += NT) bt_lds[i] = bt_glob[i];
   }
 
   // ---- Q fragments: A-operand rows = padded q rows (row = lane&15)
@@ -596,7 +599,7 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
 
   for (int base = base0; base < L; base += PD_KT) {
     // ---- stage K chunk (gather via LDS block table; zeros beyond L)
-    for (int c = tid; c < NCK; c += 256) {
+    for (int c = tid; c < NCK; c += NT) {
       const int key = c / CPK;
       const int d8 = (c % CPK) * 8;
       const int dst = key * D + swz(key, d8);
@@ -649,7 +652,7 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
     __syncthreads();  // mparts ready; every wave is past its K reads
 
     // ---- stage V into the SAME buffer; its latency hides under softmax VALU
-    for (int c = tid; c < NCK; c += 256) {
+    for (int c = tid; c < NCK; c += NT) {
       const int key = c / CPK;
       const int d8 = (c % CPK) * 8;
       const int dst = key * D + swz(key, d8);
@@ -670,8 +673,9 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       const int row = kgrp * 4 + reg;
-      float m_tile = fmaxf(fmaxf(mpart_lds[0][row], mpart_lds[1][row]),
-                           fmaxf(mpart_lds[2][row], mpart_lds[3][row]));
+      float m_tile = mpart_lds[0][row];
+#pragma unroll
+      for (int w = 1; w < NW; ++w) m_tile = fmaxf(m_tile, mpart_lds[w][row]);
       const float m_new = fmaxf(m_regs[reg], m_tile);
       const float alpha = (m_new > -1e30f) ? __expf(m_regs[reg] - m_new) : 1.f;
       const float pe = (m_new > -1e30f && sv[reg] > -1e29f)
@@ -716,7 +720,9 @@ __global__ __launch_bounds__(256) void paged_decode_mfma_kernel(
     for (int reg = 0; reg < 4; ++reg) l_lds[wid][kgrp * 4 + reg] = l_regs[reg];
   }
   __syncthreads();
-  const float l_tot = l_lds[0][col] + l_lds[1][col] + l_lds[2][col] + l_lds[3][col];
+  float l_tot = l_lds[0][col];
+#pragma unroll
+  for (int w = 1; w < NW; ++w) l_tot += l_lds[w][col];
   const float inv = (l_tot > 0.f) ? 1.0f / l_tot : 0.f;
   if (col < G) {
     __hip_bfloat16* op = out + (long)b * out_stride + (long)(kh * G + col) * D;

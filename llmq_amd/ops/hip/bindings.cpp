@@ -177,12 +177,19 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
   TORCH_CHECK(H % KVH == 0, "num_heads must divide num_kv_heads");
   TORCH_CHECK(bs == 16 || bs == 32, "block_size must be 16 or 32");
   dim3 grid(B, KVH);
-  // bf16 + MFMA-supported head dims → matrix-core kernel (G padded to 16)
+  // bf16 + MFMA-supported head dims → matrix-core kernel (G padded to 16).
+  // NW = waves per workgroup (keys-per-chunk = 16·NW); 8 halves the
+  // barriers per token vs 4 at the same waves/SIMD (A/B-able via
+  // LLMQ_DECODE_NW).
   if constexpr (std::is_same_v<T, __hip_bfloat16>) {
     if ((D == 128 || D == 256) && G <= 16) {
-      auto lm = [&]<int HD>() {
-        hipLaunchKernelGGL((paged_decode_mfma_kernel<HD>), grid, dim3(256), 0,
-                           stream(),
+      static const int nw_env = [] {
+        const char* e = getenv("LLMQ_DECODE_NW");
+        return e ? atoi(e) : 8;
+      }();
+      auto lm = [&]<int HD, int NW>() {
+        hipLaunchKernelGGL((paged_decode_mfma_kernel<HD, NW>), grid,
+                           dim3(NW * 64), 0, stream(),
                            reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
                            reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                            reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()),
@@ -191,8 +198,13 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
                            max_blocks, (float)scale, (float)softcap,
                            (int)window, q.stride(0), out.stride(0));
       };
-      if (D == 128) lm.template operator()<128>();
-      else lm.template operator()<256>();
+      if (D == 128) {
+        if (nw_env == 4) lm.template operator()<128, 4>();
+        else lm.template operator()<128, 8>();
+      } else {
+        if (nw_env == 4) lm.template operator()<256, 4>();
+        else lm.template operator()<256, 8>();
+      }
       return;
     }
   }
