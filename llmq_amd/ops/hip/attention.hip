@@ -529,6 +529,11 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
 
 // NW = waves per workgroup (4 -> 256 threads/64-key chunks; 8 -> 512
 // threads/128-key chunks: same waves/SIMD at half the barriers per token).
+// With gridDim.z == NSPLIT > 1 (flash-decoding split-KV: small B×KVH, e.g.
+// tensor-parallel ranks holding one kv head), each z-workgroup covers a
+// slice of the context and writes UNNORMALISED fp32 partials
+// (acc[G][D], m, l) to scratch[B][KVH][NSPLIT][G][D+2]; the
+// decode_splitkv_merge_kernel combines them. NSPLIT==1 writes out directly.
 template <int HEAD_DIM, int NW>
 __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
     __hip_bfloat16* __restrict__ out,      // [B, H, D]
@@ -537,6 +542,7 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
     const __hip_bfloat16* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [B, max_blocks]
     const int* __restrict__ context_lens,  // [B]
+    float* __restrict__ scratch,           // [B,KVH,NSPLIT,G,D+2] (NSPLIT>1)
     int num_heads, int num_kv_heads, int block_size, int max_blocks,
     float scale, float softcap, int window, long q_stride, long out_stride) {
   constexpr int D = HEAD_DIM;
@@ -595,9 +601,18 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
   const int base0 = (start / PD_KT) * PD_KT;
   constexpr int CPK = D / 8;    // 16B chunks per key row
   constexpr int NCK = PD_KT * CPK;
+  // split-KV: this z-block's slice of the chunk range
+  const int nsplit = gridDim.z;
+  int range_lo = base0, range_hi = L;
+  if (nsplit > 1) {
+    const int nchunks = (L - base0 + PD_KT - 1) / PD_KT;
+    const int per = (nchunks + nsplit - 1) / nsplit;
+    range_lo = base0 + (int)blockIdx.z * per * PD_KT;
+    range_hi = min(L, range_lo + per * PD_KT);
+  }
   __syncthreads();  // bt_lds ready
 
-  for (int base = base0; base < L; base += PD_KT) {
+  for (int base = range_lo; base < range_hi; base += PD_KT) {
     // ---- stage K chunk (gather via LDS block table; zeros beyond L)
     for (int c = tid; c < NCK; c += NT) {
       const int key = c / CPK;
@@ -714,7 +729,7 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
     __syncthreads();  // V/P consumed; next chunk may overwrite
   }
 
-  // ---- merge per-wave l, normalise, store this wave's dim slab
+  // ---- merge per-wave l, then store (normalised out, or raw partials)
   if (col == 0) {
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) l_lds[wid][kgrp * 4 + reg] = l_regs[reg];
@@ -723,6 +738,34 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
   float l_tot = l_lds[0][col];
 #pragma unroll
   for (int w = 1; w < NW; ++w) l_tot += l_lds[w][col];
+  if (nsplit > 1) {
+    float* slot = scratch +
+        ((((long)b * num_kv_heads + kh) * nsplit + blockIdx.z) * G) * (D + 2);
+    if (col < G) {
+      float* row = slot + (long)col * (D + 2);
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        const int dim0 = wid * D4 + dt * 16 + kgrp * 4;
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) row[dim0 + reg] = ot[dt][reg];
+      }
+    }
+    // Row stats: wave-0 lanes with col==0 cover rows r = kgrp*4+reg
+    // (m_regs is identical across waves after the shared-max combine).
+    if (wid == 0 && col == 0) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int r = kgrp * 4 + reg;
+        if (r < G) {
+          float* row = slot + (long)r * (D + 2);
+          row[D] = m_regs[reg];
+          row[D + 1] = l_lds[0][r];
+          for (int w = 1; w < NW; ++w) row[D + 1] += l_lds[w][r];
+        }
+      }
+    }
+    return;
+  }
   const float inv = (l_tot > 0.f) ? 1.0f / l_tot : 0.f;
   if (col < G) {
     __hip_bfloat16* op = out + (long)b * out_stride + (long)(kh * G + col) * D;
@@ -734,4 +777,41 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
         op[dim0 + reg] = __float2bfloat16(ot[dt][reg] * inv);
     }
   }
+}
+
+// Combine split-KV partials: out[b, kh*G+g] = Σ_s w_s·acc_s / Σ_s w_s·l_s,
+// w_s = exp(m_s − max_s m). One wave per (b, kh, g).
+template <int HEAD_DIM>
+__global__ __launch_bounds__(64) void decode_splitkv_merge_kernel(
+    __hip_bfloat16* __restrict__ out, const float* __restrict__ scratch,
+    int num_kv_heads, int G, int nsplit, long out_stride) {
+  constexpr int D = HEAD_DIM;
+  constexpr int DPL = D / WAVE;  // dims per lane (2 or 4)
+  const int b = blockIdx.x;
+  const int kh = blockIdx.y;
+  const int g = blockIdx.z;
+  const int lane = threadIdx.x;
+  const float* base = scratch +
+      ((((long)b * num_kv_heads + kh) * nsplit) * G + g) * (D + 2);
+  const long sstride = (long)G * (D + 2);
+  float m_max = -1e30f;
+  for (int s2 = 0; s2 < nsplit; ++s2) m_max = fmaxf(m_max, base[s2 * sstride + D]);
+  float acc[DPL];
+#pragma unroll
+  for (int i = 0; i < DPL; ++i) acc[i] = 0.f;
+  float l = 0.f;
+  for (int s2 = 0; s2 < nsplit; ++s2) {
+    const float* row = base + s2 * sstride;
+    const float w = __expf(row[D] - m_max);
+    if (w > 0.f) {
+      l += w * row[D + 1];
+#pragma unroll
+      for (int i = 0; i < DPL; ++i) acc[i] += w * row[lane * DPL + i];
+    }
+  }
+  const float inv = (l > 0.f) ? 1.0f / l : 0.f;
+  __hip_bfloat16* op =
+      out + (long)b * out_stride + (long)(kh * G + g) * D + lane * DPL;
+#pragma unroll
+  for (int i = 0; i < DPL; ++i) op[i] = __float2bfloat16(acc[i] * inv);
 }

@@ -187,16 +187,36 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
         const char* e = getenv("LLMQ_DECODE_NW");
         return e ? atoi(e) : 8;
       }();
+      // Split-KV: when B×KVH under-fills the chip (TP ranks hold few kv
+      // heads; small batches), split the context across z-workgroups and
+      // merge fp32 partials. Target ≥512 workgroups (2 resident per CU).
+      int nsplit = 1;
+      while (B * KVH * nsplit < 512 && nsplit < 16) nsplit *= 2;
+      at::Tensor scratch;
+      float* scratch_ptr = nullptr;
+      if (nsplit > 1) {
+        scratch = at::empty({(long)B, KVH, nsplit, G, D + 2},
+                            q.options().dtype(at::kFloat));
+        scratch_ptr = scratch.data_ptr<float>();
+      }
+      dim3 sgrid(B, KVH, nsplit);
       auto lm = [&]<int HD, int NW>() {
-        hipLaunchKernelGGL((paged_decode_mfma_kernel<HD, NW>), grid,
+        hipLaunchKernelGGL((paged_decode_mfma_kernel<HD, NW>), sgrid,
                            dim3(NW * 64), 0, stream(),
                            reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
                            reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                            reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()),
                            reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()),
-                           bt.data_ptr<int>(), cl.data_ptr<int>(), H, KVH, bs,
-                           max_blocks, (float)scale, (float)softcap,
-                           (int)window, q.stride(0), out.stride(0));
+                           bt.data_ptr<int>(), cl.data_ptr<int>(), scratch_ptr,
+                           H, KVH, bs, max_blocks, (float)scale,
+                           (float)softcap, (int)window, q.stride(0),
+                           out.stride(0));
+        if (nsplit > 1) {
+          hipLaunchKernelGGL((decode_splitkv_merge_kernel<HD>),
+                             dim3(B, KVH, G), dim3(64), 0, stream(),
+                             reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                             scratch_ptr, KVH, G, nsplit, out.stride(0));
+        }
       };
       if (D == 128) {
         // NW=16 needs D/NW >= 16 (a full MFMA dim tile per wave)
