@@ -38,7 +38,8 @@ def _looks_like_dataset(source: str) -> bool:
         return False
     p = Path(source)
     if p.exists():
-        return False
+        # an existing directory is a local dataset; an existing file is JSONL
+        return p.is_dir()
     return "/" in source or p.suffix not in (".jsonl", ".json", ".txt")
 
 

@@ -141,3 +141,41 @@ def test_status_all_queues(broker_env):
     res = runner.invoke(cli, ["status"])
     assert res.exit_code == 0, res.output
     assert "q-one" in res.output and "q-two" in res.output
+
+
+def test_submit_local_dataset_with_map(broker_env, tmp_path):
+    """HF-dataset ingestion path (reference submit.py:96-160) using a local
+    dataset directory — works offline via `datasets` local file loading."""
+    ds_dir = tmp_path / "myds"
+    ds_dir.mkdir()
+    with open(ds_dir / "train.jsonl", "w") as f:
+        for i in range(5):
+            f.write(json.dumps({"content": f"zin {i}", "lang": "nl"}) + "\n")
+    runner = CliRunner()
+    res = runner.invoke(cli, [
+        "submit", "dsq", str(ds_dir),
+        "--template", "Translate from {lang}: {text}",
+        "--map", "text=content",
+    ])
+    assert res.exit_code == 0, res.output
+
+    async def drain():
+        c = _client(broker_env)
+        await c.connect()
+        out = []
+        done = asyncio.Event()
+
+        async def cb(delivery):
+            out.append(Job.model_validate_json(delivery.body))
+            await delivery.ack()
+            if len(out) >= 5:
+                done.set()
+
+        await c.consume_jobs("dsq", cb, prefetch=10)
+        await asyncio.wait_for(done.wait(), 10)
+        await c.disconnect()
+        return out
+
+    jobs = asyncio.new_event_loop().run_until_complete(drain())
+    prompts = sorted(j.get_formatted_prompt() for j in jobs)
+    assert prompts[0] == "Translate from nl: zin 0"
