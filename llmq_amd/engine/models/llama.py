@@ -202,24 +202,22 @@ class CausalLM:
         if s.embedding_scale:
             x = x * math.sqrt(s.hidden_size)
             x = x.to(self.dtype)
-        residual: Optional[torch.Tensor] = None
+        residual = x
+        h = self._norm(x, self.layers[0].input_norm)
+        n_layers = len(self.layers)
         for i, lw in enumerate(self.layers):
-            # ---- attention block
-            if residual is None:
-                residual = x
-                h = self._norm(x, lw.input_norm)
-            else:
-                h, residual = ops.fused_add_rmsnorm(
-                    x, residual, lw.input_norm, s.rms_eps, self.norm_offset
-                )
+            # ---- attention block (h = normed input, set by the previous
+            # block's fused norm — launch count is the decode-step bound)
             qkv = F.linear(h, lw.qkv, lw.qkv_bias)
             q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
             T = q.shape[0]
             q = q.view(T, self.heads, s.head_dim)
             k = k.view(T, self.kv_heads, s.head_dim)
             v = v.view(T, self.kv_heads, s.head_dim)
-            ops.rope_inplace(q, k, positions, self.rope_cache)
-            ops.reshape_and_cache(k, v, kv_cache.k[i], kv_cache.v[i], meta.slot_mapping)
+            ops.rope_and_cache(
+                q, k, v, kv_cache.k[i], kv_cache.v[i], positions,
+                self.rope_cache, meta.slot_mapping,
+            )
             window = s.sliding_window if s.layer_uses_sliding_window(i) else 0
             if meta.is_prefill:
                 attn = ops.varlen_prefill_attention(
@@ -235,22 +233,37 @@ class CausalLM:
             if tp is not None:
                 attn_out = tp.all_reduce(attn_out)
             if s.post_norms:
-                attn_out = self._norm(attn_out, lw.post_attn_norm)
+                # Gemma-2 sandwich fused: residual += norm(attn_out, post);
+                # h = norm(residual, pre_mlp)
+                h, residual = ops.norm_add_norm(
+                    attn_out, residual, lw.post_attn_norm, lw.pre_mlp_norm,
+                    s.rms_eps, self.norm_offset,
+                )
+            else:
+                h, residual = ops.fused_add_rmsnorm(
+                    attn_out, residual, lw.pre_mlp_norm, s.rms_eps, self.norm_offset
+                )
             # ---- MLP block
-            h, residual = ops.fused_add_rmsnorm(
-                attn_out, residual, lw.pre_mlp_norm, s.rms_eps, self.norm_offset
-            )
             gate_up = F.linear(h, lw.gate_up)
             act = ops.gelu_tanh_and_mul(gate_up) if s.gelu else ops.silu_and_mul(gate_up)
             mlp_out = F.linear(act, lw.down)
             if tp is not None:
                 mlp_out = tp.all_reduce(mlp_out)
+            # The next block's input norm (or the final norm) fuses with this
+            # block's residual add (+ post norm for Gemma-2).
+            w_next = (
+                self.layers[i + 1].input_norm if i + 1 < n_layers else self.final_norm
+            )
             if s.post_norms:
-                mlp_out = self._norm(mlp_out, lw.post_mlp_norm)
-            x = mlp_out
-        # final residual add + norm
-        final = (residual.float() + x.float()).to(self.dtype)
-        return self._norm(final, self.final_norm)
+                h, residual = ops.norm_add_norm(
+                    mlp_out, residual, lw.post_mlp_norm, w_next,
+                    s.rms_eps, self.norm_offset,
+                )
+            else:
+                h, residual = ops.fused_add_rmsnorm(
+                    mlp_out, residual, w_next, s.rms_eps, self.norm_offset
+                )
+        return h
 
     @torch.no_grad()
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:

@@ -90,6 +90,18 @@ def fused_add_rmsnorm(
     return torch_ref.fused_add_rmsnorm(x, residual, weight, eps, offset)
 
 
+def norm_add_norm(
+    x: torch.Tensor, residual: torch.Tensor, w_post: torch.Tensor,
+    w_pre: torch.Tensor, eps: float, offset: float = 0.0,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Gemma-2 sandwich: residual += rmsnorm(x, w_post);
+    x = rmsnorm(residual, w_pre). In place on GPU; returns (x, residual)."""
+    if _use_hip(x):
+        _EXT.norm_add_norm(x, residual, w_post, w_pre, eps, offset)
+        return x, residual
+    return torch_ref.norm_add_norm(x, residual, w_post, w_pre, eps, offset)
+
+
 # ----------------------------------------------------------------- rope --
 
 
@@ -124,6 +136,20 @@ def gelu_tanh_and_mul(x: torch.Tensor) -> torch.Tensor:
         _EXT.gelu_tanh_and_mul(out, x)
         return out
     return torch_ref.gelu_tanh_and_mul(x)
+
+
+def rope_and_cache(
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+    k_cache: torch.Tensor, v_cache: torch.Tensor,
+    positions: torch.Tensor, cos_sin: torch.Tensor, slot_mapping: torch.Tensor,
+) -> None:
+    """Fused: RoPE(q, k) in place + scatter rotated k and v into the paged
+    cache (one launch instead of two per layer)."""
+    if _use_hip(q):
+        _EXT.rope_and_cache(q, k, v, k_cache, v_cache, positions, cos_sin, slot_mapping)
+        return
+    torch_ref.rope_inplace(q, k, positions, cos_sin)
+    torch_ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
 
 
 # ------------------------------------------------------------- KV cache --

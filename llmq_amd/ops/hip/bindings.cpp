@@ -139,6 +139,22 @@ void rope_inplace(at::Tensor q, at::Tensor k, at::Tensor positions,
   });
 }
 
+void norm_add_norm(at::Tensor x, at::Tensor residual, at::Tensor w_post,
+                   at::Tensor w_pre, double eps, double offset) {
+  CHECK_GPU(x);
+  CHECK_LASTDIM(x);
+  const int hidden = x.size(-1);
+  const long rows = x.numel() / hidden;
+  dispatch_dtype(x, "norm_add_norm", [&]<typename T>() {
+    hipLaunchKernelGGL(norm_add_norm_kernel<T>, dim3(rows), dim3(256), 0,
+                       stream(), reinterpret_cast<T*>(x.data_ptr()),
+                       reinterpret_cast<T*>(residual.data_ptr()),
+                       reinterpret_cast<const T*>(w_post.data_ptr()),
+                       reinterpret_cast<const T*>(w_pre.data_ptr()), hidden,
+                       (float)eps, (float)offset);
+  });
+}
+
 // ------------------------------------------------------------- KV cache --
 
 void reshape_and_cache(at::Tensor key, at::Tensor value, at::Tensor k_cache,
@@ -159,6 +175,34 @@ void reshape_and_cache(at::Tensor key, at::Tensor value, at::Tensor k_cache,
                        reinterpret_cast<T*>(v_cache.data_ptr()),
                        slot_mapping.data_ptr<long>(), kvh, d, bs,
                        key.stride(0), value.stride(0));
+  });
+}
+
+void rope_and_cache(at::Tensor q, at::Tensor k, at::Tensor value,
+                    at::Tensor k_cache, at::Tensor v_cache,
+                    at::Tensor positions, at::Tensor cos_sin,
+                    at::Tensor slot_mapping) {
+  CHECK_GPU(q);
+  CHECK_LASTDIM(q);
+  CHECK_LASTDIM(k);
+  TORCH_CHECK(q.dim() == 3 && k.dim() == 3, "q/k must be [T, H, D]");
+  TORCH_CHECK(cos_sin.scalar_type() == at::kFloat);
+  TORCH_CHECK(slot_mapping.scalar_type() == at::kLong);
+  const int T_ = q.size(0);
+  if (T_ == 0) return;
+  const int hq = q.size(1), hk = k.size(1), d = q.size(2);
+  const int bs = k_cache.size(2);
+  TORCH_CHECK(q.stride(1) == d && k.stride(1) == d, "head dim must be packed");
+  dispatch_dtype(q, "rope_and_cache", [&]<typename T>() {
+    hipLaunchKernelGGL(rope_and_cache_kernel<T>, dim3(T_), dim3(256), 0,
+                       stream(), reinterpret_cast<T*>(q.data_ptr()),
+                       reinterpret_cast<T*>(k.data_ptr()),
+                       reinterpret_cast<const T*>(value.data_ptr()),
+                       reinterpret_cast<T*>(k_cache.data_ptr()),
+                       reinterpret_cast<T*>(v_cache.data_ptr()),
+                       positions.data_ptr<long>(), cos_sin.data_ptr<float>(),
+                       slot_mapping.data_ptr<long>(), hq, hk, d, bs,
+                       q.stride(0), k.stride(0), value.stride(0));
   });
 }
 
@@ -363,6 +407,8 @@ TORCH_LIBRARY(llmq_amd, m) {
   m.def("paged_decode_attention(Tensor(a!) out, Tensor q, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor context_lens, float scale, float softcap, int window) -> ()");
   m.def("varlen_prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, int max_seqlen, float scale, float softcap, int window) -> ()");
   m.def("sample_gumbel_argmax(Tensor(a!) out, Tensor(b!) keys, Tensor logits, Tensor temps, int seed, int step) -> ()");
+  m.def("norm_add_norm(Tensor(a!) x, Tensor(b!) residual, Tensor w_post, Tensor w_pre, float eps, float offset) -> ()");
+  m.def("rope_and_cache(Tensor(a!) q, Tensor(b!) k, Tensor value, Tensor(c!) k_cache, Tensor(d!) v_cache, Tensor positions, Tensor cos_sin, Tensor slot_mapping) -> ()");
 }
 
 TORCH_LIBRARY_IMPL(llmq_amd, CUDA, m) {
@@ -375,4 +421,6 @@ TORCH_LIBRARY_IMPL(llmq_amd, CUDA, m) {
   m.impl("paged_decode_attention", &paged_decode_attention);
   m.impl("varlen_prefill_attention", &varlen_prefill_attention);
   m.impl("sample_gumbel_argmax", &sample_gumbel_argmax);
+  m.impl("norm_add_norm", &norm_add_norm);
+  m.impl("rope_and_cache", &rope_and_cache);
 }

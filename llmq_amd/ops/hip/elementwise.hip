@@ -146,3 +146,60 @@ __global__ void rope_kernel(T* __restrict__ q, T* __restrict__ k,
     base[half + d] = from_f32<T>(x2 * c + x1 * s);
   }
 }
+
+// Gemma-2 "sandwich": x = rmsnorm(residual + rmsnorm(x, w_post), w_pre);
+// residual += rmsnorm(x_in, w_post). One kernel replaces the post-block
+// rmsnorm + fused_add_rmsnorm pair (two of these per layer — the decode
+// step is launch-bound on these small norms at M=256).
+template <typename T>
+__global__ void norm_add_norm_kernel(T* __restrict__ x, T* __restrict__ residual,
+                                     const T* __restrict__ w_post,
+                                     const T* __restrict__ w_pre, int hidden,
+                                     float eps, float offset) {
+  constexpr int VE = Vec8<T>::kElems;
+  __shared__ float red[4];
+  const int row = blockIdx.x;
+  T* xr = x + (long)row * hidden;
+  T* rr = residual + (long)row * hidden;
+
+  float ss = 0.f;
+  for (int i = threadIdx.x * VE; i < hidden; i += blockDim.x * VE) {
+    Vec8<T> v = load16(xr + i);
+#pragma unroll
+    for (int j = 0; j < VE; ++j) {
+      float f = to_f32(v.data[j]);
+      ss += f * f;
+    }
+  }
+  ss = block_reduce_sum(ss, red);
+  const float inv1 = rsqrtf(ss / hidden + eps);
+
+  // residual += norm_post(x); accumulate the new residual's sumsq
+  float ss2 = 0.f;
+  for (int i = threadIdx.x * VE; i < hidden; i += blockDim.x * VE) {
+    Vec8<T> vx = load16(xr + i);
+    Vec8<T> vr = load16(rr + i);
+    Vec8<T> vw = load16(w_post + i);
+    Vec8<T> nr;
+#pragma unroll
+    for (int j = 0; j < VE; ++j) {
+      const float t = to_f32(vx.data[j]) * inv1 * (to_f32(vw.data[j]) + offset);
+      const float r = to_f32(vr.data[j]) + t;
+      nr.data[j] = from_f32<T>(r);
+      ss2 += r * r;
+    }
+    store16(rr + i, nr);
+  }
+  ss2 = block_reduce_sum(ss2, red);
+  const float inv2 = rsqrtf(ss2 / hidden + eps);
+
+  for (int i = threadIdx.x * VE; i < hidden; i += blockDim.x * VE) {
+    Vec8<T> vr = load16(rr + i);
+    Vec8<T> vw = load16(w_pre + i);
+    Vec8<T> o;
+#pragma unroll
+    for (int j = 0; j < VE; ++j)
+      o.data[j] = from_f32<T>(to_f32(vr.data[j]) * inv2 * (to_f32(vw.data[j]) + offset));
+    store16(xr + i, o);
+  }
+}

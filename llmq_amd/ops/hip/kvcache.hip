@@ -31,3 +31,71 @@ __global__ void reshape_and_cache_kernel(
     store16(v_cache + dst, vv);
   }
 }
+
+// Fused RoPE + cache write: rotates q and k in place AND scatters the
+// rotated k plus v into the paged cache in one launch (the decode step is
+// launch-bound on these per-layer elementwise kernels at M=256).
+// Grid: one workgroup per token.
+template <typename T>
+__global__ void rope_and_cache_kernel(
+    T* __restrict__ q,             // [T, Hq, D] (row stride q_stride)
+    T* __restrict__ k,             // [T, Hk, D]
+    const T* __restrict__ value,   // [T, Hk, D]
+    T* __restrict__ k_cache,       // [blocks, Hk, bs, D]
+    T* __restrict__ v_cache,
+    const long* __restrict__ positions,
+    const float* __restrict__ cos_sin,  // [max_pos, D] f32
+    const long* __restrict__ slots,
+    int num_q_heads, int num_k_heads, int head_dim, int block_size,
+    long q_stride, long k_stride, long v_stride) {
+  const int token = blockIdx.x;
+  const long pos = positions[token];
+  const float* cs = cos_sin + pos * head_dim;
+  const int half = head_dim / 2;
+  const long slot = slots[token];
+  const long block = slot / block_size;
+  const int off = (int)(slot % block_size);
+
+  // rotate q heads; rotate k heads and write both halves to the cache
+  const int total = (num_q_heads + num_k_heads) * half;
+  for (int i = threadIdx.x; i < total; i += blockDim.x) {
+    const int h = i / half;
+    const int d = i % half;
+    const float c = cs[d];
+    const float s = cs[half + d];
+    if (h < num_q_heads) {
+      T* base = q + (long)token * q_stride + (long)h * head_dim;
+      const float x1 = to_f32(base[d]);
+      const float x2 = to_f32(base[half + d]);
+      base[d] = from_f32<T>(x1 * c - x2 * s);
+      base[half + d] = from_f32<T>(x2 * c + x1 * s);
+    } else {
+      const int kh = h - num_q_heads;
+      T* base = k + (long)token * k_stride + (long)kh * head_dim;
+      const float x1 = to_f32(base[d]);
+      const float x2 = to_f32(base[half + d]);
+      const T r1 = from_f32<T>(x1 * c - x2 * s);
+      const T r2 = from_f32<T>(x2 * c + x1 * s);
+      base[d] = r1;
+      base[half + d] = r2;
+      if (slot >= 0) {
+        T* kdst = k_cache +
+            ((block * num_k_heads + kh) * (long)block_size + off) * head_dim;
+        kdst[d] = r1;
+        kdst[half + d] = r2;
+      }
+    }
+  }
+  if (slot < 0) return;
+  // copy v (vectorized; untouched by rope)
+  constexpr int VE = Vec8<T>::kElems;
+  const int vchunks = (num_k_heads * head_dim) / VE;
+  for (int i = threadIdx.x; i < vchunks; i += blockDim.x) {
+    const int h = (i * VE) / head_dim;
+    const int d = (i * VE) % head_dim;
+    Vec8<T> vv = load16(value + (long)token * v_stride + (long)h * head_dim + d);
+    store16(v_cache +
+                ((block * num_k_heads + h) * (long)block_size + off) * head_dim + d,
+            vv);
+  }
+}

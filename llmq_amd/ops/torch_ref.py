@@ -22,6 +22,14 @@ def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float, offset: float = 0
     return (xf * (weight.float() + offset)).to(dtype)
 
 
+def norm_add_norm(x, residual, w_post, w_pre, eps, offset=0.0):
+    """residual += rmsnorm(x, w_post); x = rmsnorm(residual, w_pre)."""
+    t = rmsnorm(x, w_post, eps, offset)
+    residual = (residual.float() + t.float()).to(x.dtype)
+    out = rmsnorm(residual, w_pre, eps, offset)
+    return out, residual
+
+
 def fused_add_rmsnorm(
     x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float, offset: float = 0.0
 ) -> Tuple[torch.Tensor, torch.Tensor]:

@@ -306,3 +306,49 @@ class TestFusedSampler:
         sigma = (expected.clamp_min(1.0)).sqrt()
         assert ((counts - expected).abs() < 5 * sigma + 10).all(), (
             counts, expected)
+
+
+class TestFusedNormSandwich:
+    @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+    def test_norm_add_norm(self, dtype):
+        hidden = 3584
+        x = torch.randn(65, hidden, device=DEV, dtype=dtype)
+        res = torch.randn(65, hidden, device=DEV, dtype=dtype)
+        w1 = torch.randn(hidden, device=DEV, dtype=dtype)
+        w2 = torch.randn(hidden, device=DEV, dtype=dtype)
+        ref_out, ref_res = torch_ref.norm_add_norm(
+            x.float().cpu(), res.float().cpu(), w1.float().cpu(),
+            w2.float().cpu(), 1e-6, 1.0,
+        )
+        out, new_res = ops.norm_add_norm(x, res, w1, w2, 1e-6, 1.0)
+        atol, rtol = TOL[dtype]
+        assert_close_to_f32_ref(new_res.cpu(), ref_res, atol, 2 * rtol)
+        assert_close_to_f32_ref(out.cpu(), ref_out, 2 * atol, 4 * rtol)
+
+
+class TestFusedRopeCache:
+    @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+    @pytest.mark.parametrize("head_dim", [128, 256])
+    def test_matches_separate_ops(self, dtype, head_dim):
+        T, HQ, HK, BS, NB = 37, 8, 2, 16, 8
+        cos_sin = torch_ref.build_rope_cache(512, head_dim, 500000.0, DEV)
+        qkv = torch.randn(T, (HQ + 2 * HK) * head_dim, device=DEV, dtype=dtype)
+        q = qkv[:, : HQ * head_dim].view(T, HQ, head_dim)
+        k = qkv[:, HQ * head_dim : (HQ + HK) * head_dim].view(T, HK, head_dim)
+        v = qkv[:, (HQ + HK) * head_dim :].view(T, HK, head_dim)
+        pos = torch.randint(0, 512, (T,), device=DEV)
+        slots = torch.randperm(NB * BS, device=DEV)[:T]
+        kc = torch.zeros(NB, HK, BS, head_dim, device=DEV, dtype=dtype)
+        vc = torch.zeros_like(kc)
+        # reference: separate ops on fp32 CPU copies
+        q_ref, k_ref = q.float().cpu().clone(), k.float().cpu().clone()
+        kc_ref = kc.float().cpu().clone()
+        vc_ref = vc.float().cpu().clone()
+        torch_ref.rope_inplace(q_ref, k_ref, pos.cpu(), cos_sin.cpu())
+        torch_ref.reshape_and_cache(k_ref, v.float().cpu(), kc_ref, vc_ref, slots.cpu())
+        ops.rope_and_cache(q, k, v, kc, vc, pos, cos_sin, slots)
+        atol, rtol = TOL[dtype]
+        assert_close_to_f32_ref(q.cpu(), q_ref, atol, rtol)
+        assert_close_to_f32_ref(k.cpu(), k_ref, atol, rtol)
+        assert_close_to_f32_ref(kc.cpu(), kc_ref, atol, rtol)
+        assert torch.equal(vc.cpu().float(), vc_ref)  # pure copy: bitwise
