@@ -174,3 +174,71 @@ def test_bridge_concurrent_requests():
         bridge.shutdown()
 
     run_async(main())
+
+
+def test_two_stage_engine_pipeline():
+    """BASELINE config #5 shape: two chained ENGINE stages (translate →
+    format), stage-2 template interpolating stage-1's result (the reference
+    never applied stage N>1 templates — broker.py:176-181)."""
+    import yaml
+
+    from llmq_amd.core.pipeline import PipelineConfig
+
+    pipeline = PipelineConfig.model_validate(yaml.safe_load("""
+name: twostage
+stages:
+  - name: translate
+    worker: engine
+    config:
+      model: tiny-llama
+      max_tokens: 4
+      temperature: 0.0
+  - name: format
+    worker: engine
+    config:
+      model: tiny-qwen2
+      template: "fmt: {translate_result}"
+      max_tokens: 4
+      temperature: 0.0
+"""))
+
+    async def main():
+        async with live_broker() as (server, config):
+            client = BrokerClient(config)
+            await client.connect()
+            await client.setup_pipeline_infrastructure(pipeline)
+
+            def stage_worker(stage):
+                cfg = pipeline.get_stage(stage).config
+                return EngineWorker(
+                    "ignored",
+                    model=cfg["model"],
+                    tensor_parallel_size=1,
+                    max_num_seqs=4,
+                    max_model_len=128,
+                    config=config,
+                    pipeline=pipeline,
+                    stage_name=stage,
+                    stage_config=cfg,
+                    engine_overrides=dict(TINY_OVERRIDES),
+                )
+
+            w1, w2 = stage_worker("translate"), stage_worker("format")
+            t1 = await _start_worker(w1)
+            t2 = await _start_worker(w2)
+            await client.publish_job(
+                pipeline.get_stage_queue_name("translate"),
+                Job(id="p1", prompt="hello"),
+            )
+            results = await _collect_results(
+                client, pipeline.get_pipeline_results_queue_name(), 1, timeout=90.0
+            )
+            assert results[0].id == "p1"
+            # stage-2 prompt was built from the stage-2 template with the
+            # stage-1 result interpolated
+            assert results[0].prompt.startswith("fmt: ")
+            await _stop_worker(w1, t1)
+            await _stop_worker(w2, t2)
+            await client.disconnect()
+
+    run_async(main())
