@@ -97,11 +97,11 @@ def _llama(name, vocab, hidden, inter, layers, heads, kv_heads, head_dim=None, *
     )
 
 
-def _qwen2(name, vocab, hidden, inter, layers, heads, kv_heads, **kw):
+def _qwen2(name, vocab, hidden, inter, layers, heads, kv_heads, head_dim=None, **kw):
     return ModelSpec(
         name=name, family="qwen2", vocab_size=vocab, hidden_size=hidden,
         intermediate_size=inter, num_layers=layers, num_heads=heads,
-        num_kv_heads=kv_heads, head_dim=hidden // heads, rms_eps=1e-6,
+        num_kv_heads=kv_heads, head_dim=head_dim or hidden // heads, rms_eps=1e-6,
         rope_theta=1000000.0, qkv_bias=True, **kw,
     )
 
@@ -120,11 +120,14 @@ def _gemma2(name, vocab, hidden, inter, layers, heads, kv_heads, head_dim, **kw)
 
 PRESETS = {
     # tiny configs for CPU tests
-    "tiny-llama": _llama("tiny-llama", 512, 64, 128, 2, 4, 2, rope_theta=10000.0,
+    # head_dim 64 (not hidden/heads = 16): the HIP attention kernels support
+    # D in {64, 128, 256}, and the GPU consistency tests run these presets.
+    "tiny-llama": _llama("tiny-llama", 512, 64, 128, 2, 4, 2, head_dim=64,
+                         rope_theta=10000.0,
                          max_position_embeddings=512, tied_embeddings=True),
-    "tiny-qwen2": _qwen2("tiny-qwen2", 512, 64, 128, 2, 4, 2,
+    "tiny-qwen2": _qwen2("tiny-qwen2", 512, 64, 128, 2, 4, 2, head_dim=64,
                          max_position_embeddings=512, tied_embeddings=True),
-    "tiny-gemma2": _gemma2("tiny-gemma2", 512, 64, 128, 2, 4, 2, 16,
+    "tiny-gemma2": _gemma2("tiny-gemma2", 512, 64, 128, 2, 4, 2, 64,
                            sliding_window=64, max_position_embeddings=512),
     # Llama 3.2 (vocab 128256, rope theta 500k)
     "llama-3.2-1b": _llama("llama-3.2-1b", 128256, 2048, 8192, 16, 32, 8,
