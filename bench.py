@@ -155,7 +155,11 @@ def main() -> None:
 
     if rank == 0:
         result = {
-            "metric": "output_tokens_per_sec (Tower-Plus-9B continuous-batching serving decode)",
+            "metric": (
+                "output_tokens_per_sec "
+                f"({'Tower-Plus-9B' if args.model == 'tower-plus-9b' else args.model}"
+                " continuous-batching serving decode)"
+            ),
             "value": round(value, 1),
             "unit": "tokens/s",
             "n_gpus": world,
