@@ -215,17 +215,24 @@ class LLMEngine:
         t0 = time.perf_counter()
         if batch.kind == "prefill":
             tokens = self.runner.execute_prefill(batch.seqs)
+        elif batch.kind == "mixed":
+            tokens = self.runner.execute_mixed(batch)
         else:
             tokens = self.runner.execute_decode(batch.seqs)
         token_list = tokens.tolist()
         now = time.time()
         self.steps += 1
         outputs: List[RequestOutput] = []
-        is_prefill = batch.kind == "prefill"
+        # rows that completed their prefill this step
+        first_prefill = (
+            0 if batch.kind == "prefill"
+            else batch.n_decode if batch.kind == "mixed"
+            else len(batch.seqs)
+        )
         eos_id = self.tokenizer.eos_token_id
         mml = self.max_model_len
-        for seq, tok in zip(batch.seqs, token_list):
-            if is_prefill:
+        for row, (seq, tok) in enumerate(zip(batch.seqs, token_list)):
+            if row >= first_prefill:
                 self._prefill_done_at[seq.request_id] = now
                 if seq.first_token_time is None:
                     seq.first_token_time = now

@@ -17,6 +17,7 @@ class PrefillMeta:
     slot_mapping: torch.Tensor  # [T] int64 global KV slots
 
     is_prefill: bool = True
+    is_mixed: bool = False
 
 
 @dataclass
@@ -28,3 +29,20 @@ class DecodeMeta:
     slot_mapping: torch.Tensor  # [B] int64
 
     is_prefill: bool = False
+    is_mixed: bool = False
+
+
+@dataclass
+class MixedMeta:
+    """One step carrying BOTH a decode batch (rows [0, n_decode)) and a
+    packed prefill batch (rows [n_decode, T)) — decode never stalls behind
+    admissions. Attention runs per segment; everything else (norms, GEMMs,
+    rope+cache) runs on the packed whole."""
+
+    n_decode: int
+    decode: "DecodeMeta"
+    prefill: "PrefillMeta"
+    slot_mapping: torch.Tensor  # [T] both segments, in row order
+
+    is_prefill: bool = False
+    is_mixed: bool = True

@@ -19,7 +19,7 @@ import torch
 
 from llmq_amd import ops
 from llmq_amd.engine.config import EngineConfig
-from llmq_amd.engine.forward_meta import DecodeMeta, PrefillMeta
+from llmq_amd.engine.forward_meta import DecodeMeta, MixedMeta, PrefillMeta
 from llmq_amd.engine.kv_cache import KVCache
 from llmq_amd.engine.models.llama import CausalLM
 from llmq_amd.engine.scheduler import ScheduledBatch, Sequence
@@ -161,6 +161,63 @@ class ModelRunner:
         last_hidden = hidden[torch.tensor(last_idx, dtype=torch.long, device=dev)]
         logits = self.model.compute_logits(last_hidden)
         return self._sample(logits, seqs)
+
+    # -- mixed (decode rows + ride-along prefill rows, eager) ------------
+
+    @torch.no_grad()
+    def execute_mixed(self, batch: "ScheduledBatch") -> torch.Tensor:
+        """Decode seqs[:n_decode] and prefill seqs[n_decode:] in ONE forward:
+        shared GEMMs/norms over the packed rows, per-segment attention.
+        Runs eager (prefill shapes vary); pure-decode steps keep hipGraphs."""
+        dev = self.device
+        dseqs = batch.seqs[: batch.n_decode]
+        pseqs = batch.seqs[batch.n_decode :]
+        bs = self.block_size
+        nd = len(dseqs)
+
+        ids = [s.token_ids[-1] for s in dseqs]
+        pos = [s.num_tokens - 1 for s in dseqs]
+        slots = [s.block_table[p // bs] * bs + (p % bs) for s, p in zip(dseqs, pos)]
+        ctx = [s.num_tokens for s in dseqs]
+        cu = [0]
+        last_idx = list(range(nd))
+        for seq in pseqs:
+            n = seq.num_tokens
+            ids.extend(seq.token_ids)
+            pos.extend(range(n))
+            bt = seq.block_table
+            slots.extend(bt[p // bs] * bs + (p % bs) for p in range(n))
+            cu.append(cu[-1] + n)
+            last_idx.append(nd + cu[-1] - 1)
+
+        block_tables = torch.zeros(
+            nd, max(len(s.block_table) for s in dseqs), dtype=torch.int32, device=dev
+        )
+        bt_np = np.zeros(tuple(block_tables.shape), dtype=np.int32)
+        for i, s in enumerate(dseqs):
+            bt_np[i, : len(s.block_table)] = s.block_table
+        block_tables.copy_(torch.from_numpy(bt_np))
+        slot_mapping = torch.tensor(slots, dtype=torch.long, device=dev)
+        meta = MixedMeta(
+            n_decode=nd,
+            decode=DecodeMeta(
+                block_tables=block_tables,
+                context_lens=torch.tensor(ctx, dtype=torch.int32, device=dev),
+                slot_mapping=slot_mapping[:nd],
+            ),
+            prefill=PrefillMeta(
+                cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+                max_seqlen=max(s.num_tokens for s in pseqs),
+                slot_mapping=slot_mapping[nd:],
+            ),
+            slot_mapping=slot_mapping,
+        )
+        input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
+        positions = torch.tensor(pos, dtype=torch.long, device=dev)
+        hidden = self.model.forward(input_ids, positions, self.kv_cache, meta)
+        last_hidden = hidden[torch.tensor(last_idx, dtype=torch.long, device=dev)]
+        logits = self.model.compute_logits(last_hidden)
+        return self._sample(logits, batch.seqs)
 
     # -- decode ----------------------------------------------------------
 

@@ -25,7 +25,7 @@ import torch
 import torch.nn.functional as F
 
 from llmq_amd import ops
-from llmq_amd.engine.forward_meta import DecodeMeta, PrefillMeta
+from llmq_amd.engine.forward_meta import DecodeMeta, MixedMeta, PrefillMeta
 from llmq_amd.engine.kv_cache import KVCache
 from llmq_amd.engine.model_specs import ModelSpec
 from llmq_amd.parallel import get_tp_group
@@ -194,7 +194,7 @@ class CausalLM:
         input_ids: torch.Tensor,  # [T]
         positions: torch.Tensor,  # [T]
         kv_cache: KVCache,
-        meta: Union[PrefillMeta, DecodeMeta],
+        meta: Union[PrefillMeta, DecodeMeta, MixedMeta],
     ) -> torch.Tensor:
         s = self.spec
         tp = get_tp_group() if self.tp_size > 1 else None
@@ -224,6 +224,19 @@ class CausalLM:
                     q, k, v, meta.cu_seqlens, meta.max_seqlen, s.scale,
                     s.attn_softcap, window,
                 )
+            elif meta.is_mixed:
+                # per-segment attention; GEMMs/norms already ran packed
+                nd = meta.n_decode
+                attn_d = ops.paged_decode_attention(
+                    q[:nd], kv_cache.k[i], kv_cache.v[i],
+                    meta.decode.block_tables, meta.decode.context_lens,
+                    s.scale, s.attn_softcap, window,
+                )
+                attn_p = ops.varlen_prefill_attention(
+                    q[nd:], k[nd:], v[nd:], meta.prefill.cu_seqlens,
+                    meta.prefill.max_seqlen, s.scale, s.attn_softcap, window,
+                )
+                attn = torch.cat([attn_d, attn_p], dim=0)
             else:
                 attn = ops.paged_decode_attention(
                     q, kv_cache.k[i], kv_cache.v[i], meta.block_tables,

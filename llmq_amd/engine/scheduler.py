@@ -69,8 +69,10 @@ class Sequence:
 
 @dataclass
 class ScheduledBatch:
-    kind: str  # "prefill" | "decode"
+    kind: str  # "prefill" | "decode" | "mixed"
     seqs: List[Sequence] = field(default_factory=list)
+    # mixed: seqs[:n_decode] are decoding, seqs[n_decode:] are new prefills
+    n_decode: int = 0
 
     @property
     def empty(self) -> bool:
@@ -93,6 +95,11 @@ class Scheduler:
         self.max_model_len = max_model_len
         self.waiting: Deque[Sequence] = deque()
         self.running: List[Sequence] = []
+        self._starve_ticks = 0
+
+    # steps a too-big-for-budget prompt may wait behind mixed steps before
+    # it gets a solo prefill batch
+    SOLO_PREFILL_AFTER = 8
 
     # -- public ----------------------------------------------------------
 
@@ -136,10 +143,52 @@ class Scheduler:
             pass
 
     def schedule(self) -> ScheduledBatch:
-        prefill = self._schedule_prefill()
-        if not prefill.empty:
-            return prefill
-        return self._schedule_decode()
+        if not self.running:
+            return self._schedule_prefill()
+        # Admissions that fit the prefill-token budget ride along with the
+        # decode batch (mixed step) so decode never stalls behind new
+        # prompts; a head-of-queue prompt larger than the budget gets a
+        # solo prefill step (preserves FCFS, bounds its wait).
+        decode = self._schedule_decode()
+        if not self.waiting:
+            return decode
+        head = self.waiting[0]
+        if head.num_tokens > self.max_prefill_tokens and decode.empty:
+            return self._schedule_prefill()
+        admits = self._admit_within_budget(len(decode.seqs))
+        if not admits:
+            if head.num_tokens > self.max_prefill_tokens:
+                # big prompt and decode still running: alternate one solo
+                # prefill step so it cannot starve
+                if self._starve_ticks >= self.SOLO_PREFILL_AFTER:
+                    self._starve_ticks = 0
+                    return self._schedule_prefill()
+                self._starve_ticks += 1
+            return decode
+        self._starve_ticks = 0
+        if decode.empty:
+            return ScheduledBatch("prefill", admits)
+        return ScheduledBatch("mixed", decode.seqs + admits, n_decode=len(decode.seqs))
+
+    def _admit_within_budget(self, seats_used: int) -> List[Sequence]:
+        admits: List[Sequence] = []
+        budget = self.max_prefill_tokens
+        while self.waiting and seats_used + len(admits) < self.max_num_seqs:
+            seq = self.waiting[0]
+            n = seq.num_tokens
+            if n > budget:
+                break
+            blocks = self.allocator.allocate(blocks_needed(n, self.block_size))
+            if blocks is None:
+                break
+            seq.block_table = blocks
+            seq.status = SeqStatus.RUNNING
+            self.waiting.popleft()
+            admits.append(seq)
+            budget -= n
+        if admits:
+            self.running.extend(admits)
+        return admits
 
     # -- internals -------------------------------------------------------
 
