@@ -93,6 +93,7 @@ class AsyncEngineBridge:
         self._ready = threading.Event()
         self._init_error: Optional[BaseException] = None
         self._stop = False
+        self._fatal: Optional[BaseException] = None
         self.engine: Any = None
         self._ctrl_group = None
 
@@ -113,6 +114,8 @@ class AsyncEngineBridge:
         self, request_id: str, prompt: str, params: SamplingParams
     ) -> _FinishedRequest:
         assert self._loop is not None, "bridge not started"
+        if self._fatal is not None:
+            raise RuntimeError("engine thread crashed") from self._fatal
         fut: asyncio.Future = self._loop.create_future()
         self._futures[request_id] = fut
         with self._pending_lock:
@@ -187,6 +190,14 @@ class AsyncEngineBridge:
                                 decode_ms=out.decode_ms,
                             ),
                         )
+        except BaseException as exc:  # noqa: BLE001 — engine died: fail fast
+            logger.exception("engine thread crashed; failing %d in-flight requests",
+                             len(self._futures))
+            self._fatal = exc
+            for request_id in list(self._futures):
+                self._resolve_error(
+                    request_id, RuntimeError(f"engine thread crashed: {exc}")
+                )
         finally:
             if self.tp_size > 1:
                 try:

@@ -242,3 +242,35 @@ stages:
             await client.disconnect()
 
     run_async(main())
+
+
+def test_bridge_engine_crash_fails_inflight():
+    """If the engine thread dies, in-flight generate() awaits must raise
+    (not hang) and later submissions are rejected."""
+
+    async def main():
+        class Boom:
+            tokenizer = None
+
+            def add_request(self, *a, **k):
+                pass
+
+            def has_unfinished(self):
+                return True
+
+            def step(self):
+                raise RuntimeError("kaboom")
+
+        bridge = AsyncEngineBridge(lambda: Boom())
+        await bridge.start()
+        from llmq_amd.engine.sampling_params import SamplingParams
+
+        with pytest.raises(RuntimeError):
+            await asyncio.wait_for(
+                bridge.generate("x", "p", SamplingParams(max_tokens=2)), 10
+            )
+        with pytest.raises(RuntimeError):
+            await bridge.generate("y", "p", SamplingParams(max_tokens=2))
+        bridge.shutdown()
+
+    run_async(main())
