@@ -264,10 +264,12 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
       };
       if (D == 128) {
         // NW=16 needs D/NW >= 16 (a full MFMA dim tile per wave)
-        if (nw_env == 4) lm.template operator()<128, 4>();
+        if (nw_env == 2) lm.template operator()<128, 2>();
+        else if (nw_env == 4) lm.template operator()<128, 4>();
         else lm.template operator()<128, 8>();
       } else {
-        if (nw_env == 4) lm.template operator()<256, 4>();
+        if (nw_env == 2) lm.template operator()<256, 2>();
+        else if (nw_env == 4) lm.template operator()<256, 4>();
         else if (nw_env == 16) lm.template operator()<256, 16>();
         else lm.template operator()<256, 8>();
       }
