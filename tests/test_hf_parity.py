@@ -19,8 +19,8 @@ pytestmark = pytest.mark.integration
 transformers = pytest.importorskip("transformers")
 
 
-def _save_tiny_llama(tmp_path, qwen=False):
-    if qwen:
+def _save_tiny(tmp_path, family="llama"):
+    if family == "qwen2":
         cfg = transformers.Qwen2Config(
             vocab_size=300, hidden_size=64, intermediate_size=128,
             num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
@@ -28,6 +28,16 @@ def _save_tiny_llama(tmp_path, qwen=False):
             tie_word_embeddings=False,
         )
         model = transformers.Qwen2ForCausalLM(cfg)
+    elif family == "gemma2":
+        cfg = transformers.Gemma2Config(
+            vocab_size=300, hidden_size=64, intermediate_size=128,
+            num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+            head_dim=32, max_position_embeddings=256, rope_theta=10000.0,
+            rms_norm_eps=1e-6, attn_logit_softcapping=50.0,
+            final_logit_softcapping=30.0, sliding_window=64,
+            query_pre_attn_scalar=32, tie_word_embeddings=True,
+        )
+        model = transformers.Gemma2ForCausalLM(cfg)
     else:
         cfg = transformers.LlamaConfig(
             vocab_size=300, hidden_size=64, intermediate_size=128,
@@ -43,9 +53,9 @@ def _save_tiny_llama(tmp_path, qwen=False):
     return model
 
 
-@pytest.mark.parametrize("qwen", [False, True])
-def test_logits_match_transformers(tmp_path, qwen):
-    hf = _save_tiny_llama(tmp_path, qwen=qwen)
+@pytest.mark.parametrize("family", ["llama", "qwen2", "gemma2"])
+def test_logits_match_transformers(tmp_path, family):
+    hf = _save_tiny(tmp_path, family=family)
 
     from llmq_amd.engine.config import EngineConfig
     from llmq_amd.engine.engine import LLMEngine
@@ -57,7 +67,8 @@ def test_logits_match_transformers(tmp_path, qwen):
     spec = engine.spec
     assert spec.num_layers == 2
     assert spec.num_heads == 4 and spec.num_kv_heads == 2
-    assert spec.qkv_bias == qwen
+    assert spec.qkv_bias == (family == "qwen2")
+    assert spec.post_norms == (family == "gemma2")
 
     token_ids = [1, 7, 42, 99, 123, 250, 3]
     # in-tree forward: run a prefill step and capture the logits
