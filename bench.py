@@ -38,6 +38,8 @@ def main() -> None:
     parser.add_argument("--max-model-len", type=int, default=4096)
     parser.add_argument("--temperature", type=float, default=0.7)
     parser.add_argument("--eager", action="store_true", help="disable hipGraphs")
+    parser.add_argument("--kv-dtype", default="auto", choices=["auto", "fp8"],
+                        help="KV cache storage dtype (fp8 = optional mode, NOT the headline)")
     parser.add_argument("--profile-steps", type=int, default=0,
                         help="extra untimed steps after the timed region (rocprof)")
     args = parser.parse_args()
@@ -80,6 +82,7 @@ def main() -> None:
         fast_init=True,
         device=device,
         enforce_eager=args.eager,
+        kv_cache_dtype=args.kv_dtype,
         hipgraph_max_batch=batch,
         seed=1234 + rank,
     )
@@ -169,13 +172,17 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if use_gpu else "fp32",
+            "dtype": (
+                ("bf16" if args.kv_dtype == "auto" else "bf16+fp8kv")
+                if use_gpu else "fp32"
+            ),
             "data": "synthetic",
             "config": {
                 "model": f"{model} ({engine.spec.name} arch, random init)",
                 "global_batch": world * batch,
                 "seq_len": prompt_len,
                 "parallelism": f"dp{world}",
+                "kv_cache_dtype": args.kv_dtype,
             },
         }
         # completed jobs/sec companion number: at steady-state decode one

@@ -17,6 +17,9 @@ class EngineConfig:
     max_prefill_tokens: int = 8192  # token budget per prefill step
     gpu_memory_utilization: float = 0.9
     kv_block_size: int = 16
+    # "auto" = model dtype; "fp8" = OCP e4m3 KV storage (bf16 compute) —
+    # halves the decode KV stream; opt-in, off by default.
+    kv_cache_dtype: str = "auto"
     enable_hipgraph: bool = True
     hipgraph_max_batch: int = 512
     tensor_parallel_size: int = 1
@@ -32,6 +35,16 @@ class EngineConfig:
         if self.device == "auto":
             return torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
         return torch.device(self.device)
+
+    def resolve_kv_dtype(self, device: torch.device) -> torch.dtype:
+        model_dtype = self.resolve_dtype(device)
+        if self.kv_cache_dtype == "auto":
+            return model_dtype
+        if self.kv_cache_dtype == "fp8":
+            if device.type != "cuda" or model_dtype != torch.bfloat16:
+                raise ValueError("kv_cache_dtype=fp8 requires GPU bf16 execution")
+            return torch.float8_e4m3fn
+        raise ValueError(f"unknown kv_cache_dtype {self.kv_cache_dtype!r}")
 
     def resolve_dtype(self, device: torch.device) -> torch.dtype:
         if device.type == "cpu":

@@ -100,11 +100,14 @@ class LLMEngine:
         self.tokenizer = load_tokenizer(
             config.model, self.spec.vocab_size, self.spec.bos_token_id, self.spec.eos_token_id
         )
+        self.kv_dtype = config.resolve_kv_dtype(self.device)
+        if self.kv_dtype == torch.float8_e4m3fn and self.spec.head_dim not in (128, 256):
+            raise ValueError("fp8 KV cache requires head_dim 128 or 256 (MFMA decode path)")
         num_blocks = self._size_kv_cache()
         kv_heads_local = self.spec.num_kv_heads // tp
         self.kv_cache = KVCache(
             self.spec.num_layers, num_blocks, kv_heads_local,
-            config.kv_block_size, self.spec.head_dim, self.device, self.dtype,
+            config.kv_block_size, self.spec.head_dim, self.device, self.kv_dtype,
         )
         self.allocator = BlockAllocator(num_blocks)
         self.scheduler = Scheduler(
@@ -126,7 +129,7 @@ class LLMEngine:
             num_blocks, num_blocks * config.kv_block_size,
             num_blocks * KVCache.block_bytes(
                 self.spec.num_layers, kv_heads_local, config.kv_block_size,
-                self.spec.head_dim, self.dtype,
+                self.spec.head_dim, self.kv_dtype,
             ) / 2 ** 30 * 1.0,
             config.max_num_seqs,
             self.runner.use_graphs,
@@ -161,7 +164,7 @@ class LLMEngine:
         budget = int(total * cfg.gpu_memory_utilization) - used - _RESERVE_BYTES
         num = KVCache.num_blocks_for_budget(
             budget, self.spec.num_layers, kv_heads_local, cfg.kv_block_size,
-            self.spec.head_dim, self.dtype,
+            self.spec.head_dim, self.kv_dtype,
         )
         # No point holding more KV than every admitted seq at full context.
         cap = (cfg.max_num_seqs * self.max_model_len) // cfg.kv_block_size + cfg.max_num_seqs

@@ -527,6 +527,23 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
 
 #define PD_MAX_BT 1024  // block-table entries staged in LDS (tail from global)
 
+
+// KV-cache load: 8 cache elements → bf16x8 fragment. bf16 caches are a raw
+// 16-byte load; fp8 (OCP e4m3) caches load 8 bytes and convert on the fly —
+// the fp8 KV option halves the decode KV stream.
+DEVINL bf16x8_t kv8_to_bf16(const __hip_bfloat16* p) {
+  return *reinterpret_cast<const bf16x8_t*>(p);
+}
+DEVINL bf16x8_t kv8_to_bf16(const __hip_fp8_e4m3* p) {
+  const uint2 raw = *reinterpret_cast<const uint2*>(p);
+  const __hip_fp8_e4m3* e = reinterpret_cast<const __hip_fp8_e4m3*>(&raw);
+  bf16x8_t out;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    out[j] = __bfloat16_as_short(__float2bfloat16(static_cast<float>(e[j])));
+  return out;
+}
+
 // NW = waves per workgroup (4 -> 256 threads/64-key chunks; 8 -> 512
 // threads/128-key chunks: same waves/SIMD at half the barriers per token).
 // With gridDim.z == NSPLIT > 1 (flash-decoding split-KV: small B×KVH, e.g.
@@ -534,12 +551,12 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
 // slice of the context and writes UNNORMALISED fp32 partials
 // (acc[G][D], m, l) to scratch[B][KVH][NSPLIT][G][D+2]; the
 // decode_splitkv_merge_kernel combines them. NSPLIT==1 writes out directly.
-template <int HEAD_DIM, int NW>
+template <int HEAD_DIM, int NW, typename TC = __hip_bfloat16>
 __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
     __hip_bfloat16* __restrict__ out,      // [B, H, D]
     const __hip_bfloat16* __restrict__ q,  // [B, H, D]
-    const __hip_bfloat16* __restrict__ k_cache,  // [nb, KVH, bs, D]
-    const __hip_bfloat16* __restrict__ v_cache,
+    const TC* __restrict__ k_cache,        // [nb, KVH, bs, D]
+    const TC* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [B, max_blocks]
     const int* __restrict__ context_lens,  // [B]
     float* __restrict__ scratch,           // [B,KVH,NSPLIT,G,D+2] (NSPLIT>1)
@@ -624,8 +641,7 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
         const long blk = (bidx < PD_MAX_BT) ? bt_lds[bidx] : bt_glob[bidx];
         const long rowoff =
             ((blk * num_kv_heads + kh) * block_size + gkey % block_size) * D + d8;
-        *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) =
-            *reinterpret_cast<const bf16x8_t*>(k_cache + rowoff);
+        *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) = kv8_to_bf16(k_cache + rowoff);
       } else {
         *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
       }
@@ -677,8 +693,7 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
         const long blk = (bidx < PD_MAX_BT) ? bt_lds[bidx] : bt_glob[bidx];
         const long rowoff =
             ((blk * num_kv_heads + kh) * block_size + gkey % block_size) * D + d8;
-        *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) =
-            *reinterpret_cast<const bf16x8_t*>(v_cache + rowoff);
+        *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) = kv8_to_bf16(v_cache + rowoff);
       } else {
         *reinterpret_cast<bf16x8_t*>(&kv_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
       }

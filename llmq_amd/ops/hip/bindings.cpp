@@ -9,34 +9,9 @@
 #include <hip/hip_bf16.h>
 #include <hip/hip_fp16.h>
 
-// kernel declarations (defined in the .hip TUs)
-template <typename T>
-__global__ void rmsnorm_kernel(T*, const T*, const T*, int, float, float);
-template <typename T>
-__global__ void fused_add_rmsnorm_kernel(T*, T*, const T*, int, float, float);
-template <typename T, bool GELU>
-__global__ void act_and_mul_kernel(T*, const T*, long, int);
-template <typename T>
-__global__ void rope_kernel(T*, T*, const long*, const float*, int, int, int,
-                            long, long);
-template <typename T>
-__global__ void reshape_and_cache_kernel(const T*, const T*, T*, T*,
-                                         const long*, int, int, int, long, long);
-template <typename T, int HEAD_DIM>
-__global__ void paged_decode_attention_kernel(T*, const T*, const T*, const T*,
-                                              const int*, const int*, int, int,
-                                              int, int, float, float, int, long,
-                                              long);
-template <typename T, int HEAD_DIM>
-__global__ void varlen_prefill_attention_kernel(T*, const T*, const T*, const T*,
-                                                const int*, int, int, float,
-                                                float, int, long, long, long,
-                                                long);
-template <int HEAD_DIM>
-__global__ void flash_prefill_bf16_kernel(__hip_bfloat16*, const __hip_bfloat16*,
-                                          const __hip_bfloat16*, const __hip_bfloat16*,
-                                          const int*, int, int, float, float, int,
-                                          long, long, long, long);
+// Kernel definitions are included BEFORE this file in the umbrella TU
+// (ops.hip) — no forward declarations needed (stale declarations with
+// narrower template parameter lists silently generate dead launch stubs).
 
 namespace {
 
@@ -167,7 +142,23 @@ void reshape_and_cache(at::Tensor key, at::Tensor value, at::Tensor k_cache,
   const int bs = k_cache.size(2);
   TORCH_CHECK(k_cache.is_contiguous() && v_cache.is_contiguous());
   TORCH_CHECK(slot_mapping.scalar_type() == at::kLong);
+  const bool fp8_cache = k_cache.scalar_type() == at::kFloat8_e4m3fn;
   dispatch_dtype(key, "reshape_and_cache", [&]<typename T>() {
+    if (fp8_cache) {
+      if constexpr (std::is_same_v<T, __hip_bfloat16>) {
+        hipLaunchKernelGGL((reshape_and_cache_kernel<T, __hip_fp8_e4m3>),
+                           dim3(T_), dim3(128), 0, stream(),
+                           reinterpret_cast<const T*>(key.data_ptr()),
+                           reinterpret_cast<const T*>(value.data_ptr()),
+                           reinterpret_cast<__hip_fp8_e4m3*>(k_cache.data_ptr()),
+                           reinterpret_cast<__hip_fp8_e4m3*>(v_cache.data_ptr()),
+                           slot_mapping.data_ptr<long>(), kvh, d, bs,
+                           key.stride(0), value.stride(0));
+      } else {
+        TORCH_CHECK(false, "fp8 KV cache requires bf16 activations");
+      }
+      return;
+    }
     hipLaunchKernelGGL(reshape_and_cache_kernel<T>, dim3(T_), dim3(128), 0,
                        stream(), reinterpret_cast<const T*>(key.data_ptr()),
                        reinterpret_cast<const T*>(value.data_ptr()),
@@ -193,7 +184,25 @@ void rope_and_cache(at::Tensor q, at::Tensor k, at::Tensor value,
   const int hq = q.size(1), hk = k.size(1), d = q.size(2);
   const int bs = k_cache.size(2);
   TORCH_CHECK(q.stride(1) == d && k.stride(1) == d, "head dim must be packed");
+  const bool fp8_cache = k_cache.scalar_type() == at::kFloat8_e4m3fn;
   dispatch_dtype(q, "rope_and_cache", [&]<typename T>() {
+    if (fp8_cache) {
+      if constexpr (std::is_same_v<T, __hip_bfloat16>) {
+        hipLaunchKernelGGL((rope_and_cache_kernel<T, __hip_fp8_e4m3>),
+                           dim3(T_), dim3(256), 0, stream(),
+                           reinterpret_cast<T*>(q.data_ptr()),
+                           reinterpret_cast<T*>(k.data_ptr()),
+                           reinterpret_cast<const T*>(value.data_ptr()),
+                           reinterpret_cast<__hip_fp8_e4m3*>(k_cache.data_ptr()),
+                           reinterpret_cast<__hip_fp8_e4m3*>(v_cache.data_ptr()),
+                           positions.data_ptr<long>(), cos_sin.data_ptr<float>(),
+                           slot_mapping.data_ptr<long>(), hq, hk, d, bs,
+                           q.stride(0), k.stride(0), value.stride(0));
+      } else {
+        TORCH_CHECK(false, "fp8 KV cache requires bf16 activations");
+      }
+      return;
+    }
     hipLaunchKernelGGL(rope_and_cache_kernel<T>, dim3(T_), dim3(256), 0,
                        stream(), reinterpret_cast<T*>(q.data_ptr()),
                        reinterpret_cast<T*>(k.data_ptr()),
@@ -244,13 +253,14 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
         scratch_ptr = scratch.data_ptr<float>();
       }
       dim3 sgrid(B, KVH, nsplit);
-      auto lm = [&]<int HD, int NW>() {
-        hipLaunchKernelGGL((paged_decode_mfma_kernel<HD, NW>), sgrid,
+      const bool fp8_cache = kc.scalar_type() == at::kFloat8_e4m3fn;
+      auto lm = [&]<int HD, int NW, typename TC>() {
+        hipLaunchKernelGGL((paged_decode_mfma_kernel<HD, NW, TC>), sgrid,
                            dim3(NW * 64), 0, stream(),
                            reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
                            reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
-                           reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()),
-                           reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()),
+                           reinterpret_cast<const TC*>(kc.data_ptr()),
+                           reinterpret_cast<const TC*>(vc.data_ptr()),
                            bt.data_ptr<int>(), cl.data_ptr<int>(), scratch_ptr,
                            H, KVH, bs, max_blocks, (float)scale,
                            (float)softcap, (int)window, q.stride(0),
@@ -262,16 +272,21 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
                              scratch_ptr, KVH, G, nsplit, out.stride(0));
         }
       };
-      if (D == 128) {
+      using BF = __hip_bfloat16;
+      using F8 = __hip_fp8_e4m3;
+      if (fp8_cache) {  // fp8 KV rides the default NW=8 configuration
+        if (D == 128) lm.template operator()<128, 8, F8>();
+        else lm.template operator()<256, 8, F8>();
+      } else if (D == 128) {
         // NW=16 needs D/NW >= 16 (a full MFMA dim tile per wave)
-        if (nw_env == 2) lm.template operator()<128, 2>();
-        else if (nw_env == 4) lm.template operator()<128, 4>();
-        else lm.template operator()<128, 8>();
+        if (nw_env == 2) lm.template operator()<128, 2, BF>();
+        else if (nw_env == 4) lm.template operator()<128, 4, BF>();
+        else lm.template operator()<128, 8, BF>();
       } else {
-        if (nw_env == 2) lm.template operator()<256, 2>();
-        else if (nw_env == 4) lm.template operator()<256, 4>();
-        else if (nw_env == 16) lm.template operator()<256, 16>();
-        else lm.template operator()<256, 8>();
+        if (nw_env == 2) lm.template operator()<256, 2, BF>();
+        else if (nw_env == 4) lm.template operator()<256, 4, BF>();
+        else if (nw_env == 16) lm.template operator()<256, 16, BF>();
+        else lm.template operator()<256, 8, BF>();
       }
       return;
     }

@@ -5,6 +5,9 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include <hip/hip_fp16.h>
+#include <hip/hip_fp8.h>
+
+#include <type_traits>
 
 #define WAVE 64
 #define DEVINL __device__ __forceinline__
@@ -17,6 +20,9 @@ template <> DEVINL float to_f32<__hip_bfloat16>(__hip_bfloat16 x) {
   return __bfloat162float(x);
 }
 template <> DEVINL float to_f32<_Float16>(_Float16 x) { return (float)x; }
+template <> DEVINL float to_f32<__hip_fp8_e4m3>(__hip_fp8_e4m3 x) {
+  return static_cast<float>(x);
+}
 
 template <typename T> DEVINL T from_f32(float x);
 template <> DEVINL float from_f32<float>(float x) { return x; }
@@ -24,6 +30,9 @@ template <> DEVINL __hip_bfloat16 from_f32<__hip_bfloat16>(float x) {
   return __float2bfloat16(x);
 }
 template <> DEVINL _Float16 from_f32<_Float16>(float x) { return (_Float16)x; }
+template <> DEVINL __hip_fp8_e4m3 from_f32<__hip_fp8_e4m3>(float x) {
+  return __hip_fp8_e4m3(x);
+}
 
 // ---- vectorized 16-byte access (8 bf16 / 8 fp16 / 4 f32) -------------------
 // G13: hipcc does not auto-vectorize bf16 loads; reinterpret as int4.

@@ -4,12 +4,12 @@
 
 #include "common.h"
 
-template <typename T>
+template <typename T, typename TC = T>
 __global__ void reshape_and_cache_kernel(
     const T* __restrict__ key,     // [T, kv_heads, D] (row stride key_stride)
     const T* __restrict__ value,
-    T* __restrict__ k_cache,       // [blocks, kv_heads, bs, D]
-    T* __restrict__ v_cache,
+    TC* __restrict__ k_cache,      // [blocks, kv_heads, bs, D]
+    TC* __restrict__ v_cache,
     const long* __restrict__ slots,  // [T] global slot id
     int kv_heads, int head_dim, int block_size,
     long key_stride, long val_stride) {
@@ -26,9 +26,17 @@ __global__ void reshape_and_cache_kernel(
     const long dst =
         ((block * kv_heads + h) * (long)block_size + off) * head_dim + d;
     Vec8<T> kv = load16(key + (long)token * key_stride + (long)h * head_dim + d);
-    store16(k_cache + dst, kv);
     Vec8<T> vv = load16(value + (long)token * val_stride + (long)h * head_dim + d);
-    store16(v_cache + dst, vv);
+    if constexpr (std::is_same_v<T, TC>) {
+      store16(k_cache + dst, kv);
+      store16(v_cache + dst, vv);
+    } else {  // quantizing cache write (e.g. bf16 → fp8)
+#pragma unroll
+      for (int j = 0; j < VE; ++j) {
+        k_cache[dst + j] = from_f32<TC>(to_f32(kv.data[j]));
+        v_cache[dst + j] = from_f32<TC>(to_f32(vv.data[j]));
+      }
+    }
   }
 }
 
@@ -36,13 +44,13 @@ __global__ void reshape_and_cache_kernel(
 // rotated k plus v into the paged cache in one launch (the decode step is
 // launch-bound on these per-layer elementwise kernels at M=256).
 // Grid: one workgroup per token.
-template <typename T>
+template <typename T, typename TC = T>
 __global__ void rope_and_cache_kernel(
     T* __restrict__ q,             // [T, Hq, D] (row stride q_stride)
     T* __restrict__ k,             // [T, Hk, D]
     const T* __restrict__ value,   // [T, Hk, D]
-    T* __restrict__ k_cache,       // [blocks, Hk, bs, D]
-    T* __restrict__ v_cache,
+    TC* __restrict__ k_cache,      // [blocks, Hk, bs, D]
+    TC* __restrict__ v_cache,
     const long* __restrict__ positions,
     const float* __restrict__ cos_sin,  // [max_pos, D] f32
     const long* __restrict__ slots,
@@ -79,10 +87,15 @@ __global__ void rope_and_cache_kernel(
       base[d] = r1;
       base[half + d] = r2;
       if (slot >= 0) {
-        T* kdst = k_cache +
+        TC* kdst = k_cache +
             ((block * num_k_heads + kh) * (long)block_size + off) * head_dim;
-        kdst[d] = r1;
-        kdst[half + d] = r2;
+        if constexpr (std::is_same_v<T, TC>) {
+          kdst[d] = r1;
+          kdst[half + d] = r2;
+        } else {
+          kdst[d] = from_f32<TC>(to_f32(r1));
+          kdst[half + d] = from_f32<TC>(to_f32(r2));
+        }
       }
     }
   }
@@ -94,8 +107,13 @@ __global__ void rope_and_cache_kernel(
     const int h = (i * VE) / head_dim;
     const int d = (i * VE) % head_dim;
     Vec8<T> vv = load16(value + (long)token * v_stride + (long)h * head_dim + d);
-    store16(v_cache +
-                ((block * num_k_heads + h) * (long)block_size + off) * head_dim + d,
-            vv);
+    TC* vdst = v_cache +
+        ((block * num_k_heads + h) * (long)block_size + off) * head_dim + d;
+    if constexpr (std::is_same_v<T, TC>) {
+      store16(vdst, vv);
+    } else {
+#pragma unroll
+      for (int j = 0; j < VE; ++j) vdst[j] = from_f32<TC>(to_f32(vv.data[j]));
+    }
   }
 }

@@ -392,3 +392,67 @@ class TestEngineFamiliesGPU:
         # model; require agreement on the first few steps for every prompt.
         for c, g in zip(cpu_tokens, gpu_tokens):
             assert c[:4] == g[:4], (c, g)
+
+
+class TestFp8KVCache:
+    def test_decode_matches_quantized_ref(self):
+        """fp8 KV decode vs an fp32 reference computed on the DEQUANTIZED
+        cache — isolates kernel correctness from quantization error."""
+        B, G, KVH, D, BS = 6, 2, 8, 256, 16
+        H = G * KVH
+        torch.manual_seed(5)
+        ctx = torch.tensor([1, 17, 64, 100, 150, 333], dtype=torch.int32, device=DEV)
+        kc8, vc8, bt = None, None, None
+        kc, vc, bt = _build_cache(B, KVH, D, BS, 333, torch.bfloat16)
+        kc8 = kc.to(torch.float8_e4m3fn)
+        vc8 = vc.to(torch.float8_e4m3fn)
+        q = torch.randn(B, H, D, device=DEV, dtype=torch.bfloat16)
+        scale = D ** -0.5
+        out = ops.paged_decode_attention(q, kc8, vc8, bt, ctx, scale)
+        ref = torch_ref.paged_decode_attention(
+            q.float().cpu(), kc8.float().cpu(), vc8.float().cpu(),
+            bt.cpu(), ctx.cpu(), scale,
+        )
+        assert_close_to_f32_ref(out.cpu(), ref, 2e-2, 1e-1)
+
+    def test_cache_write_roundtrip(self):
+        """rope_and_cache with an fp8 cache: written values must equal the
+        bf16-rotated values quantized to e4m3."""
+        T, HQ, HK, D, BS, NB = 16, 4, 2, 128, 16, 4
+        cos_sin = torch_ref.build_rope_cache(64, D, 10000.0, DEV)
+        qkv = torch.randn(T, (HQ + 2 * HK) * D, device=DEV, dtype=torch.bfloat16)
+        q = qkv[:, : HQ * D].view(T, HQ, D)
+        k = qkv[:, HQ * D : (HQ + HK) * D].view(T, HK, D)
+        v = qkv[:, (HQ + HK) * D :].view(T, HK, D)
+        pos = torch.randint(0, 64, (T,), device=DEV)
+        slots = torch.randperm(NB * BS, device=DEV)[:T]
+        kc8 = torch.zeros(NB, HK, BS, D, device=DEV, dtype=torch.float8_e4m3fn)
+        vc8 = torch.zeros_like(kc8)
+        k_before = k.clone()
+        ops.rope_and_cache(q, k, v, kc8, vc8, pos, cos_sin, slots)
+        # k was rotated in place (bf16); cache rows must equal fp8(k)
+        for t in range(T):
+            s = int(slots[t])
+            blk, off = s // BS, s % BS
+            got_k = kc8[blk, :, off].float()
+            want_k = k[t].to(torch.float8_e4m3fn).float()
+            assert torch.equal(got_k.cpu(), want_k.cpu())
+            got_v = vc8[blk, :, off].float()
+            want_v = v[t].to(torch.float8_e4m3fn).float()
+            assert torch.equal(got_v.cpu(), want_v.cpu())
+        assert not torch.equal(k_before, k)  # rope actually applied
+
+    def test_engine_runs_with_fp8_kv(self):
+        from llmq_amd.engine.config import EngineConfig
+        from llmq_amd.engine.engine import LLMEngine
+        from llmq_amd.engine.sampling_params import SamplingParams
+
+        eng = LLMEngine(EngineConfig(
+            model="llama-3.2-1b", max_num_seqs=4, max_model_len=256,
+            load_weights=False, num_kv_blocks=512, kv_cache_dtype="fp8",
+        ))
+        outs = eng.generate_batch(
+            ["hello", "world"], SamplingParams(temperature=0.0, max_tokens=8,
+                                               ignore_eos=True))
+        assert len(outs) == 2
+        assert eng.kv_cache.k[0].dtype == torch.float8_e4m3fn
