@@ -125,6 +125,9 @@ PRESETS = {
     "tiny-llama": _llama("tiny-llama", 512, 64, 128, 2, 4, 2, head_dim=64,
                          rope_theta=10000.0,
                          max_position_embeddings=512, tied_embeddings=True),
+    "tiny-llama-d128": _llama("tiny-llama-d128", 512, 64, 128, 2, 4, 2,
+                              head_dim=128, rope_theta=10000.0,
+                              max_position_embeddings=512, tied_embeddings=True),
     "tiny-qwen2": _qwen2("tiny-qwen2", 512, 64, 128, 2, 4, 2, head_dim=64,
                          max_position_embeddings=512, tied_embeddings=True),
     "tiny-gemma2": _gemma2("tiny-gemma2", 512, 64, 128, 2, 4, 2, 64,

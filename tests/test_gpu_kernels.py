@@ -448,7 +448,7 @@ class TestFp8KVCache:
         from llmq_amd.engine.sampling_params import SamplingParams
 
         eng = LLMEngine(EngineConfig(
-            model="llama-3.2-1b", max_num_seqs=4, max_model_len=256,
+            model="tiny-llama-d128", max_num_seqs=4, max_model_len=256,
             load_weights=False, num_kv_blocks=512, kv_cache_dtype="fp8",
         ))
         outs = eng.generate_batch(
