@@ -162,8 +162,21 @@ class AsyncEngineBridge:
             return
         self._ready.set()
         engine = self.engine
+        last_log = time.monotonic()
+        steps_since = 0
         try:
             while not self._stop:
+                now = time.monotonic()
+                if now - last_log >= 10.0:
+                    sched = engine.scheduler
+                    logger.info(
+                        "engine: %.1f steps/s, running=%d waiting=%d in_flight=%d "
+                        "free_blocks=%d",
+                        steps_since / (now - last_log), sched.num_running,
+                        sched.num_waiting, len(self._futures),
+                        engine.allocator.num_free,
+                    )
+                    last_log, steps_since = now, 0
                 subs = self._drain()
                 if not subs and not engine.has_unfinished():
                     self._wakeup.wait(timeout=self.IDLE_POLL_S)
@@ -177,6 +190,7 @@ class AsyncEngineBridge:
                         engine.add_request(s.request_id, prompt=s.prompt, params=s.params)
                     except ValueError as exc:
                         self._resolve_error(s.request_id, exc)
+                steps_since += 1
                 for out in engine.step():
                     if out.finished:
                         self._resolve(
