@@ -217,7 +217,7 @@ class LLMEngine:
             return []
         t0 = time.perf_counter()
         if batch.kind == "prefill":
-            tokens = self.runner.execute_prefill(batch.seqs)
+            tokens = self.runner.execute_prefill(batch.seqs, batch.chunks or None)
         elif batch.kind == "mixed":
             tokens = self.runner.execute_mixed(batch)
         else:
@@ -226,15 +226,25 @@ class LLMEngine:
         now = time.time()
         self.steps += 1
         outputs: List[RequestOutput] = []
-        # rows that completed their prefill this step
-        first_prefill = (
-            0 if batch.kind == "prefill"
-            else batch.n_decode if batch.kind == "mixed"
-            else len(batch.seqs)
-        )
+        # Only rows that SAMPLED get a token: every decode row, plus prefill
+        # rows whose context completed this step (mid-chunks of long prompts
+        # produce no token yet — chunked prefill).
+        if batch.kind == "decode":
+            sampled_seqs = batch.seqs
+            first_prefill = len(batch.seqs)
+        else:
+            pseqs = batch.seqs[batch.n_decode:]
+            chunks = batch.chunks or [(0, sq.num_tokens) for sq in pseqs]
+            final = []
+            for sq, (st, e) in zip(pseqs, chunks):
+                sq.prefilled = e
+                if e == sq.num_tokens:
+                    final.append(sq)
+            sampled_seqs = batch.seqs[: batch.n_decode] + final
+            first_prefill = batch.n_decode
         eos_id = self.tokenizer.eos_token_id
         mml = self.max_model_len
-        for row, (seq, tok) in enumerate(zip(batch.seqs, token_list)):
+        for row, (seq, tok) in enumerate(zip(sampled_seqs, token_list)):
             if row >= first_prefill:
                 self._prefill_done_at[seq.request_id] = now
                 if seq.first_token_time is None:

@@ -9,12 +9,27 @@ import torch
 
 
 @dataclass
-class PrefillMeta:
-    """Packed variable-length prompt batch."""
+class ChunkGather:
+    """Chunked-prefill context assembly: per layer, the attention K/V is the
+    GATHERED past (from the paged cache) interleaved with the fresh chunk
+    rows. Indices are precomputed once per step."""
 
-    cu_seqlens: torch.Tensor  # [B+1] int32
+    cu_seqlens_k: torch.Tensor   # [B+1] int32 key offsets (>= query lens)
+    blocks: torch.Tensor         # [P] int64 cache block index per past row
+    offs: torch.Tensor           # [P] int64 in-block offset per past row
+    past_dst: torch.Tensor       # [P] int64 destination rows in the packed K
+    fresh_dst: torch.Tensor      # [Tq] int64 destination rows in the packed K
+    total_k: int                 # packed K rows
+
+
+@dataclass
+class PrefillMeta:
+    """Packed variable-length prompt batch (whole prompts or chunks)."""
+
+    cu_seqlens: torch.Tensor  # [B+1] int32 (query offsets)
     max_seqlen: int
     slot_mapping: torch.Tensor  # [T] int64 global KV slots
+    gather: "ChunkGather | None" = None  # set when any entry is a chunk
 
     is_prefill: bool = True
     is_mixed: bool = False

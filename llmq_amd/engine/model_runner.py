@@ -19,7 +19,7 @@ import torch
 
 from llmq_amd import ops
 from llmq_amd.engine.config import EngineConfig
-from llmq_amd.engine.forward_meta import DecodeMeta, MixedMeta, PrefillMeta
+from llmq_amd.engine.forward_meta import ChunkGather, DecodeMeta, MixedMeta, PrefillMeta
 from llmq_amd.engine.kv_cache import KVCache
 from llmq_amd.engine.models.llama import CausalLM
 from llmq_amd.engine.scheduler import ScheduledBatch, Sequence
@@ -130,37 +130,83 @@ class ModelRunner:
                 return b
         return None
 
-    # -- prefill ---------------------------------------------------------
+    # -- prefill / mixed (chunk-aware, eager) ----------------------------
 
-    @torch.no_grad()
-    def execute_prefill(self, seqs: List[Sequence]) -> torch.Tensor:
-        """Run packed varlen prefill; returns sampled next tokens [B]."""
-        dev = self.device
-        all_ids: List[int] = []
-        all_pos: List[int] = []
-        all_slots: List[int] = []
+    def _build_prefill_meta(self, pseqs, chunks, dev) -> tuple:
+        """Packed ids/pos/slots for the prefill segment + optional gather
+        plan for chunk continuations. Returns (ids, pos, slots, meta,
+        rel_last_idx) where rel_last_idx are within-segment row indices of
+        FINAL chunks (the rows that sample)."""
+        bs = self.block_size
+        ids: List[int] = []
+        pos: List[int] = []
+        slots: List[int] = []
         cu = [0]
-        last_idx = []
-        for seq in seqs:
-            n = seq.num_tokens
-            all_ids.extend(seq.token_ids)
-            all_pos.extend(range(n))
+        cu_k = [0]
+        rel_last: List[int] = []
+        any_chunk = any(start > 0 for (start, _e) in chunks)
+        g_blocks: List[int] = []
+        g_offs: List[int] = []
+        past_dst: List[int] = []
+        fresh_dst: List[int] = []
+        krow = 0
+        for seq, (start, end) in zip(pseqs, chunks):
+            n = end - start
+            ids.extend(seq.token_ids[start:end])
+            pos.extend(range(start, end))
             bt = seq.block_table
-            bs = self.block_size
-            all_slots.extend(bt[p // bs] * bs + (p % bs) for p in range(n))
+            slots.extend(bt[p // bs] * bs + (p % bs) for p in range(start, end))
+            if end == seq.num_tokens:
+                rel_last.append(cu[-1] + n - 1)
             cu.append(cu[-1] + n)
-            last_idx.append(cu[-1] - 1)
-        input_ids = torch.tensor(all_ids, dtype=torch.long, device=dev)
-        positions = torch.tensor(all_pos, dtype=torch.long, device=dev)
+            if any_chunk:
+                for p in range(start):  # past rows gathered from the cache
+                    g_blocks.append(bt[p // bs])
+                    g_offs.append(p % bs)
+                    past_dst.append(krow)
+                    krow += 1
+                for i in range(cu[-2], cu[-1]):
+                    fresh_dst.append(krow)
+                    krow += 1
+            cu_k.append(cu_k[-1] + end)  # keys = full context so far
+        slot_mapping = torch.tensor(slots, dtype=torch.long, device=dev)
+        gather = None
+        if any_chunk:
+            gather = ChunkGather(
+                cu_seqlens_k=torch.tensor(cu_k, dtype=torch.int32, device=dev),
+                blocks=torch.tensor(g_blocks, dtype=torch.long, device=dev),
+                offs=torch.tensor(g_offs, dtype=torch.long, device=dev),
+                past_dst=torch.tensor(past_dst, dtype=torch.long, device=dev),
+                fresh_dst=torch.tensor(fresh_dst, dtype=torch.long, device=dev),
+                total_k=krow,
+            )
         meta = PrefillMeta(
             cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
-            max_seqlen=max(s.num_tokens for s in seqs),
-            slot_mapping=torch.tensor(all_slots, dtype=torch.long, device=dev),
+            max_seqlen=max(e - st for (st, e) in chunks),
+            slot_mapping=slot_mapping,
+            gather=gather,
         )
+        return ids, pos, slot_mapping, meta, rel_last
+
+    @torch.no_grad()
+    def execute_prefill(self, seqs: List[Sequence], chunks=None) -> torch.Tensor:
+        """Run a packed prefill segment; returns sampled next tokens for the
+        rows whose context completed this step (all rows when no chunking)."""
+        if chunks is None:
+            chunks = [(0, s.num_tokens) for s in seqs]
+        dev = self.device
+        ids, pos, slot_mapping, meta, rel_last = self._build_prefill_meta(
+            seqs, chunks, dev
+        )
+        input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
+        positions = torch.tensor(pos, dtype=torch.long, device=dev)
         hidden = self.model.forward(input_ids, positions, self.kv_cache, meta)
-        last_hidden = hidden[torch.tensor(last_idx, dtype=torch.long, device=dev)]
+        if not rel_last:
+            return torch.empty(0, dtype=torch.long, device=dev)
+        last_hidden = hidden[torch.tensor(rel_last, dtype=torch.long, device=dev)]
         logits = self.model.compute_logits(last_hidden)
-        return self._sample(logits, seqs)
+        sample_seqs = [s for s, (st, e) in zip(seqs, chunks) if e == s.num_tokens]
+        return self._sample(logits, sample_seqs)
 
     # -- mixed (decode rows + ride-along prefill rows, eager) ------------
 
@@ -168,10 +214,12 @@ class ModelRunner:
     def execute_mixed(self, batch: "ScheduledBatch") -> torch.Tensor:
         """Decode seqs[:n_decode] and prefill seqs[n_decode:] in ONE forward:
         shared GEMMs/norms over the packed rows, per-segment attention.
-        Runs eager (prefill shapes vary); pure-decode steps keep hipGraphs."""
+        Runs eager (prefill shapes vary); pure-decode steps keep hipGraphs.
+        Returns tokens for [decode rows] + [final-chunk prefill rows]."""
         dev = self.device
         dseqs = batch.seqs[: batch.n_decode]
         pseqs = batch.seqs[batch.n_decode :]
+        chunks = batch.chunks or [(0, s.num_tokens) for s in pseqs]
         bs = self.block_size
         nd = len(dseqs)
 
@@ -179,16 +227,11 @@ class ModelRunner:
         pos = [s.num_tokens - 1 for s in dseqs]
         slots = [s.block_table[p // bs] * bs + (p % bs) for s, p in zip(dseqs, pos)]
         ctx = [s.num_tokens for s in dseqs]
-        cu = [0]
-        last_idx = list(range(nd))
-        for seq in pseqs:
-            n = seq.num_tokens
-            ids.extend(seq.token_ids)
-            pos.extend(range(n))
-            bt = seq.block_table
-            slots.extend(bt[p // bs] * bs + (p % bs) for p in range(n))
-            cu.append(cu[-1] + n)
-            last_idx.append(nd + cu[-1] - 1)
+        p_ids, p_pos, p_slots, pmeta, rel_last = self._build_prefill_meta(
+            pseqs, chunks, dev
+        )
+        ids.extend(p_ids)
+        pos.extend(p_pos)
 
         block_tables = torch.zeros(
             nd, max(len(s.block_table) for s in dseqs), dtype=torch.int32, device=dev
@@ -197,27 +240,28 @@ class ModelRunner:
         for i, s in enumerate(dseqs):
             bt_np[i, : len(s.block_table)] = s.block_table
         block_tables.copy_(torch.from_numpy(bt_np))
-        slot_mapping = torch.tensor(slots, dtype=torch.long, device=dev)
+        d_slots = torch.tensor(slots, dtype=torch.long, device=dev)
+        slot_mapping = torch.cat([d_slots, p_slots])
         meta = MixedMeta(
             n_decode=nd,
             decode=DecodeMeta(
                 block_tables=block_tables,
                 context_lens=torch.tensor(ctx, dtype=torch.int32, device=dev),
-                slot_mapping=slot_mapping[:nd],
+                slot_mapping=d_slots,
             ),
-            prefill=PrefillMeta(
-                cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
-                max_seqlen=max(s.num_tokens for s in pseqs),
-                slot_mapping=slot_mapping[nd:],
-            ),
+            prefill=pmeta,
             slot_mapping=slot_mapping,
         )
         input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
         positions = torch.tensor(pos, dtype=torch.long, device=dev)
         hidden = self.model.forward(input_ids, positions, self.kv_cache, meta)
+        last_idx = list(range(nd)) + [nd + r for r in rel_last]
         last_hidden = hidden[torch.tensor(last_idx, dtype=torch.long, device=dev)]
         logits = self.model.compute_logits(last_hidden)
-        return self._sample(logits, batch.seqs)
+        sample_seqs = list(dseqs) + [
+            s for s, (st, e) in zip(pseqs, chunks) if e == s.num_tokens
+        ]
+        return self._sample(logits, sample_seqs)
 
     # -- decode ----------------------------------------------------------
 

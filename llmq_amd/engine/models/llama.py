@@ -188,6 +188,30 @@ class CausalLM:
     def _norm(self, x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
         return ops.rmsnorm(x, w, self.spec.rms_eps, self.norm_offset)
 
+    def _prefill_attn(self, q, k, v, pmeta, kv_cache, layer: int, window: int):
+        """Packed varlen attention; chunk continuations assemble their full
+        context (gathered past from the paged cache + fresh rows)."""
+        s = self.spec
+        g = pmeta.gather
+        if g is None:
+            return ops.varlen_prefill_attention(
+                q, k, v, pmeta.cu_seqlens, pmeta.max_seqlen, s.scale,
+                s.attn_softcap, window,
+            )
+        KVH, D = k.shape[1], k.shape[2]
+        k_att = q.new_empty(g.total_k, KVH, D)
+        v_att = q.new_empty(g.total_k, KVH, D)
+        past_k = kv_cache.k[layer][g.blocks, :, g.offs].to(q.dtype)
+        past_v = kv_cache.v[layer][g.blocks, :, g.offs].to(q.dtype)
+        k_att.index_copy_(0, g.past_dst, past_k)
+        v_att.index_copy_(0, g.past_dst, past_v)
+        k_att.index_copy_(0, g.fresh_dst, k)
+        v_att.index_copy_(0, g.fresh_dst, v)
+        return ops.varlen_prefill_attention(
+            q, k_att, v_att, pmeta.cu_seqlens, pmeta.max_seqlen, s.scale,
+            s.attn_softcap, window, cu_seqlens_k=g.cu_seqlens_k,
+        )
+
     @torch.no_grad()
     def forward(
         self,
@@ -220,10 +244,7 @@ class CausalLM:
             )
             window = s.sliding_window if s.layer_uses_sliding_window(i) else 0
             if meta.is_prefill:
-                attn = ops.varlen_prefill_attention(
-                    q, k, v, meta.cu_seqlens, meta.max_seqlen, s.scale,
-                    s.attn_softcap, window,
-                )
+                attn = self._prefill_attn(q, k, v, meta, kv_cache, i, window)
             elif meta.is_mixed:
                 # per-segment attention; GEMMs/norms already ran packed
                 nd = meta.n_decode
@@ -232,9 +253,8 @@ class CausalLM:
                     meta.decode.block_tables, meta.decode.context_lens,
                     s.scale, s.attn_softcap, window,
                 )
-                attn_p = ops.varlen_prefill_attention(
-                    q[nd:], k[nd:], v[nd:], meta.prefill.cu_seqlens,
-                    meta.prefill.max_seqlen, s.scale, s.attn_softcap, window,
+                attn_p = self._prefill_attn(
+                    q[nd:], k[nd:], v[nd:], meta.prefill, kv_cache, i, window
                 )
                 attn = torch.cat([attn_d, attn_p], dim=0)
             else:

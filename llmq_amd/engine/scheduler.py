@@ -38,7 +38,7 @@ class Sequence:
     __slots__ = (
         "request_id", "token_ids", "prompt_len", "params", "status",
         "block_table", "arrival_time", "first_token_time", "finish_reason",
-        "output_text", "num_preemptions",
+        "output_text", "num_preemptions", "prefilled",
     )
 
     def __init__(self, request_id: str, prompt_token_ids: List[int], params: SamplingParams,
@@ -54,6 +54,7 @@ class Sequence:
         self.finish_reason: Optional[str] = None
         self.output_text = None  # set only when a stop string truncates
         self.num_preemptions = 0
+        self.prefilled = 0  # context tokens whose KV is in the cache
 
     @property
     def num_tokens(self) -> int:
@@ -71,8 +72,11 @@ class Sequence:
 class ScheduledBatch:
     kind: str  # "prefill" | "decode" | "mixed"
     seqs: List[Sequence] = field(default_factory=list)
-    # mixed: seqs[:n_decode] are decoding, seqs[n_decode:] are new prefills
+    # mixed: seqs[:n_decode] are decoding, seqs[n_decode:] are prefilling
     n_decode: int = 0
+    # per prefill entry: (start, end) token range entering the cache this
+    # step; end < num_tokens ⇒ a chunk of a long prompt (no sampling yet)
+    chunks: List[tuple] = field(default_factory=list)
 
     @property
     def empty(self) -> bool:
@@ -95,11 +99,8 @@ class Scheduler:
         self.max_model_len = max_model_len
         self.waiting: Deque[Sequence] = deque()
         self.running: List[Sequence] = []
-        self._starve_ticks = 0
-
-    # steps a too-big-for-budget prompt may wait behind mixed steps before
-    # it gets a solo prefill batch
-    SOLO_PREFILL_AFTER = 8
+        # long prompts being prefilled chunk-by-chunk (not yet decodable)
+        self.prefilling: List[Sequence] = []
 
     # -- public ----------------------------------------------------------
 
@@ -109,10 +110,10 @@ class Scheduler:
 
     @property
     def num_running(self) -> int:
-        return len(self.running)
+        return len(self.running) + len(self.prefilling)
 
     def has_work(self) -> bool:
-        return bool(self.waiting or self.running)
+        return bool(self.waiting or self.running or self.prefilling)
 
     def add(self, seq: Sequence) -> None:
         if seq.num_tokens > self.max_model_len:
@@ -122,11 +123,12 @@ class Scheduler:
         self.waiting.append(seq)
 
     def abort(self, request_id: str) -> bool:
-        for i, seq in enumerate(self.running):
-            if seq.request_id == request_id:
-                self._release(seq)
-                del self.running[i]
-                return True
+        for pool in (self.running, self.prefilling):
+            for i, seq in enumerate(pool):
+                if seq.request_id == request_id:
+                    self._release(seq)
+                    del pool[i]
+                    return True
         for i, seq in enumerate(self.waiting):
             if seq.request_id == request_id:
                 del self.waiting[i]
@@ -143,52 +145,72 @@ class Scheduler:
             pass
 
     def schedule(self) -> ScheduledBatch:
-        if not self.running:
-            return self._schedule_prefill()
-        # Admissions that fit the prefill-token budget ride along with the
-        # decode batch (mixed step) so decode never stalls behind new
-        # prompts; a head-of-queue prompt larger than the budget gets a
-        # solo prefill step (preserves FCFS, bounds its wait).
-        decode = self._schedule_decode()
-        if not self.waiting:
+        # One step = the decode batch (all running seqs) + a prefill segment
+        # built from (a) continuing chunks of long prompts, (b) new
+        # admissions, within max_prefill_tokens — decode never stalls.
+        decode = self._schedule_decode() if self.running else ScheduledBatch("decode")
+        prefills, chunks = self._build_prefill_segment(len(decode.seqs))
+        if not prefills:
             return decode
-        head = self.waiting[0]
-        if head.num_tokens > self.max_prefill_tokens and decode.empty:
-            return self._schedule_prefill()
-        admits = self._admit_within_budget(len(decode.seqs))
-        if not admits:
-            if head.num_tokens > self.max_prefill_tokens:
-                # big prompt and decode still running: alternate one solo
-                # prefill step so it cannot starve
-                if self._starve_ticks >= self.SOLO_PREFILL_AFTER:
-                    self._starve_ticks = 0
-                    return self._schedule_prefill()
-                self._starve_ticks += 1
-            return decode
-        self._starve_ticks = 0
         if decode.empty:
-            return ScheduledBatch("prefill", admits)
-        return ScheduledBatch("mixed", decode.seqs + admits, n_decode=len(decode.seqs))
+            return ScheduledBatch("prefill", prefills, chunks=chunks)
+        return ScheduledBatch(
+            "mixed", decode.seqs + prefills, n_decode=len(decode.seqs),
+            chunks=chunks,
+        )
 
-    def _admit_within_budget(self, seats_used: int) -> List[Sequence]:
-        admits: List[Sequence] = []
+    def _build_prefill_segment(self, seats_used: int):
+        prefills: List[Sequence] = []
+        chunks: List[tuple] = []
         budget = self.max_prefill_tokens
-        while self.waiting and seats_used + len(admits) < self.max_num_seqs:
+        bs = self.block_size
+        # (a) continue chunked prompts first (FCFS among them)
+        for seq in list(self.prefilling):
+            if budget <= 0:
+                break
+            start = seq.prefilled
+            end = min(seq.num_tokens, start + budget)
+            need = blocks_needed(end, bs) - len(seq.block_table)
+            if need > 0:
+                blocks = self.allocator.allocate(need)
+                if blocks is None:
+                    break
+                seq.block_table.extend(blocks)
+            prefills.append(seq)
+            chunks.append((start, end))
+            budget -= end - start
+        # (b) admit new sequences
+        while (
+            self.waiting and budget > 0
+            and seats_used + len(prefills) < self.max_num_seqs
+        ):
             seq = self.waiting[0]
             n = seq.num_tokens
-            if n > budget:
-                break
-            blocks = self.allocator.allocate(blocks_needed(n, self.block_size))
+            end = min(n, budget)
+            if end < n and prefills:
+                break  # start a long prompt's first chunk only at segment head
+            blocks = self.allocator.allocate(blocks_needed(end, bs))
             if blocks is None:
                 break
             seq.block_table = blocks
             seq.status = SeqStatus.RUNNING
             self.waiting.popleft()
-            admits.append(seq)
-            budget -= n
-        if admits:
-            self.running.extend(admits)
-        return admits
+            prefills.append(seq)
+            chunks.append((0, end))
+            budget -= end
+            if end < n:
+                break  # chunked head consumed the budget
+        # move bookkeeping: where does each prefill seq live after this step?
+        for seq, (start, end) in zip(prefills, chunks):
+            if seq in self.prefilling:
+                if end == seq.num_tokens:
+                    self.prefilling.remove(seq)
+                    self.running.append(seq)
+            elif end < seq.num_tokens:
+                self.prefilling.append(seq)
+            else:
+                self.running.append(seq)
+        return prefills, chunks
 
     # -- internals -------------------------------------------------------
 
@@ -197,26 +219,6 @@ class Scheduler:
             self.allocator.free(seq.block_table)
             seq.block_table = []
 
-    def _schedule_prefill(self) -> ScheduledBatch:
-        batch = ScheduledBatch("prefill")
-        budget = self.max_prefill_tokens
-        while self.waiting and len(self.running) + len(batch.seqs) < self.max_num_seqs:
-            seq = self.waiting[0]
-            n = seq.num_tokens
-            if batch.seqs and n > budget:
-                break
-            need = blocks_needed(n, self.block_size)
-            blocks = self.allocator.allocate(need)
-            if blocks is None:
-                break
-            seq.block_table = blocks
-            seq.status = SeqStatus.RUNNING
-            self.waiting.popleft()
-            batch.seqs.append(seq)
-            budget -= n
-        if batch.seqs:
-            self.running.extend(batch.seqs)
-        return batch
 
     def _schedule_decode(self) -> ScheduledBatch:
         batch = ScheduledBatch("decode")
@@ -258,5 +260,6 @@ class Scheduler:
         self._release(seq)
         seq.status = SeqStatus.WAITING
         seq.num_preemptions += 1
+        seq.prefilled = 0  # cache gone: the whole context re-prefills
         self.running.remove(seq)
         self.waiting.appendleft(seq)

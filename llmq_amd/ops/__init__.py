@@ -201,14 +201,21 @@ def varlen_prefill_attention(
     scale: float,
     softcap: float = 0.0,
     window: int = 0,
+    cu_seqlens_k: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
+    """Packed causal attention; cu_seqlens_k (key offsets per seq, >= query
+    lengths) enables chunked prefill: Q is the last Lq positions of a
+    Lk-long context."""
+    cu_k = cu_seqlens if cu_seqlens_k is None else cu_seqlens_k
     if _use_hip(q):
         out = torch.empty_like(q)
         _EXT.varlen_prefill_attention(
-            out, q, k, v, cu_seqlens, int(max_seqlen), scale, softcap, window
+            out, q, k, v, cu_seqlens, cu_k, int(max_seqlen), scale, softcap, window
         )
         return out
-    return torch_ref.varlen_prefill_attention(q, k, v, cu_seqlens, scale, softcap, window)
+    return torch_ref.varlen_prefill_attention(
+        q, k, v, cu_seqlens, scale, softcap, window, cu_seqlens_k=cu_k
+    )
 
 
 # ------------------------------------------------------------- sampling --

@@ -218,11 +218,12 @@ __global__ __launch_bounds__(256) void paged_decode_attention_kernel(
 
 template <typename T, int HEAD_DIM>
 __global__ __launch_bounds__(256) void varlen_prefill_attention_kernel(
-    T* __restrict__ out,            // [T, H, D]
-    const T* __restrict__ q,        // [T, H, D] (row strides below)
-    const T* __restrict__ k,        // [T, KVH, D]
+    T* __restrict__ out,            // [Tq, H, D]
+    const T* __restrict__ q,        // [Tq, H, D] (row strides below)
+    const T* __restrict__ k,        // [Tk, KVH, D]
     const T* __restrict__ v,
-    const int* __restrict__ cu_seqlens,  // [B+1]
+    const int* __restrict__ cu_seqlens,    // [B+1] query offsets
+    const int* __restrict__ cu_seqlens_k,  // [B+1] key offsets (chunked)
     int num_heads, int num_kv_heads, float scale, float softcap, int window,
     long q_stride, long k_stride, long v_stride, long o_stride) {
   constexpr int D = HEAD_DIM;
@@ -233,6 +234,9 @@ __global__ __launch_bounds__(256) void varlen_prefill_attention_kernel(
   const int kvh = h / (num_heads / num_kv_heads);
   const int s0 = cu_seqlens[seq], s1 = cu_seqlens[seq + 1];
   const int L = s1 - s0;
+  const int s0k = cu_seqlens_k[seq];
+  const int Lk = cu_seqlens_k[seq + 1] - s0k;
+  const int off = Lk - L;  // abs position of q row 0
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
 
@@ -249,12 +253,13 @@ __global__ __launch_bounds__(256) void varlen_prefill_attention_kernel(
 #pragma unroll
     for (int t = 0; t < DPT; ++t) acc[t] = 0.f;
 
-    const int kstart = (window > 0 && i + 1 > window) ? (i + 1 - window) : 0;
-    for (int base = (kstart / WAVE) * WAVE; base <= i; base += WAVE) {
-      const int j = base + lane;  // this lane's key
+    const int iabs = off + i;  // absolute position of this query row
+    const int kstart = (window > 0 && iabs + 1 > window) ? (iabs + 1 - window) : 0;
+    for (int base = (kstart / WAVE) * WAVE; base <= iabs; base += WAVE) {
+      const int j = base + lane;  // this lane's key (absolute)
       float score = -1e30f;
-      if (j <= i && j >= kstart && j < L) {
-        const T* krow = k + (long)(s0 + j) * k_stride + (long)kvh * D;
+      if (j <= iabs && j >= kstart && j < Lk) {
+        const T* krow = k + (long)(s0k + j) * k_stride + (long)kvh * D;
         float dot = 0.f;
 #pragma unroll 4
         for (int d = 0; d < D; d += VE) {
@@ -273,11 +278,11 @@ __global__ __launch_bounds__(256) void varlen_prefill_attention_kernel(
       m_run = m_new;
 #pragma unroll
       for (int t = 0; t < DPT; ++t) acc[t] *= alpha;
-      const int jn = min(i - base + 1, WAVE);
+      const int jn = min(iabs - base + 1, WAVE);
       for (int jj = 0; jj < jn; ++jj) {
         const float pj = __shfl(p, jj, WAVE);
         if (pj != 0.f) {
-          const T* vrow = v + (long)(s0 + base + jj) * v_stride + (long)kvh * D;
+          const T* vrow = v + (long)(s0k + base + jj) * v_stride + (long)kvh * D;
 #pragma unroll
           for (int t = 0; t < DPT; ++t)
             acc[t] += pj * to_f32(vrow[lane * DPT + t]);
@@ -316,13 +321,18 @@ DEVINL int swz(int row, int d) {  // element-index XOR swizzle (16B granules)
   return d ^ ((row & 7) << 3);
 }
 
+// cu_seqlens_k: key-side offsets — equal to cu_seqlens for plain prefill;
+// for CHUNKED prefill a sequence's K/V covers its full context so far
+// (gathered past + fresh chunk) while Q is only the new chunk, and q row i
+// sits at absolute position (Lk - Lq) + i.
 template <int HEAD_DIM>
 __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
-    __hip_bfloat16* __restrict__ out,      // [T, H, D]
-    const __hip_bfloat16* __restrict__ q,  // [T, H, D]
-    const __hip_bfloat16* __restrict__ k,  // [T, KVH, D]
+    __hip_bfloat16* __restrict__ out,      // [Tq, H, D]
+    const __hip_bfloat16* __restrict__ q,  // [Tq, H, D]
+    const __hip_bfloat16* __restrict__ k,  // [Tk, KVH, D]
     const __hip_bfloat16* __restrict__ v,
-    const int* __restrict__ cu_seqlens, int num_heads, int num_kv_heads,
+    const int* __restrict__ cu_seqlens, const int* __restrict__ cu_seqlens_k,
+    int num_heads, int num_kv_heads,
     float scale, float softcap, int window, long q_stride, long k_stride,
     long v_stride, long o_stride) {
   constexpr int D = HEAD_DIM;
@@ -330,7 +340,10 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
   const int h = blockIdx.y;
   const int kvh = h / (num_heads / num_kv_heads);
   const int s0 = cu_seqlens[seq];
-  const int L = cu_seqlens[seq + 1] - s0;
+  const int L = cu_seqlens[seq + 1] - s0;       // query rows
+  const int s0k = cu_seqlens_k[seq];
+  const int Lk = cu_seqlens_k[seq + 1] - s0k;   // key rows (>= L)
+  const int off = Lk - L;                       // abs position of q row 0
   const int q_base = blockIdx.z * PF_QT;
   if (q_base >= L) return;
 
@@ -366,12 +379,12 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
 #pragma unroll
   for (int dt = 0; dt < DT; ++dt) ot[dt] = {0.f, 0.f, 0.f, 0.f};
 
-  const int wave_max_row = min(q_base + wid * 16 + 15, L - 1);
-  const int block_max_row = min(q_base + PF_QT - 1, L - 1);
-  const int kv_end = block_max_row + 1;  // causal bound for the workgroup
+  const int wave_max_row = off + min(q_base + wid * 16 + 15, L - 1);
+  const int block_max_row = off + min(q_base + PF_QT - 1, L - 1);
+  const int kv_end = block_max_row + 1;  // causal bound (absolute keys)
   int kv_begin = 0;
   if (window > 0) {
-    const int wave_min_needed = q_base + 1 - window;  // earliest key any row sees
+    const int wave_min_needed = off + q_base + 1 - window;
     kv_begin = max(0, (wave_min_needed / PF_KT) * PF_KT);
   }
 
@@ -385,13 +398,13 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
         const int d8 = (c % (D / CHW)) * CHW;
         const int dst = key * D + swz(key, d8);
         const int gkey = kt + key;
-        if (gkey < L) {
+        if (gkey < Lk) {
           *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) =
               *reinterpret_cast<const bf16x8_t*>(
-                  k + (long)(s0 + gkey) * k_stride + (long)kvh * D + d8);
+                  k + (long)(s0k + gkey) * k_stride + (long)kvh * D + d8);
           *reinterpret_cast<bf16x8_t*>(&v_lds[dst]) =
               *reinterpret_cast<const bf16x8_t*>(
-                  v + (long)(s0 + gkey) * v_stride + (long)kvh * D + d8);
+                  v + (long)(s0k + gkey) * v_stride + (long)kvh * D + d8);
         } else {
           *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
           *reinterpret_cast<bf16x8_t*>(&v_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
@@ -419,14 +432,15 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
       float m_tile[4], l_tile[4], p[4][4];
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        const int row = q_base + wid * 16 + kgrp * 4 + reg;
+        const int qrow = q_base + wid * 16 + kgrp * 4 + reg;
+        const int row = off + qrow;  // absolute position
         float mx = -1e30f;
 #pragma unroll
         for (int ct = 0; ct < 4; ++ct) {
           const int key = kt + ct * 16 + col;
           float x = s[ct][reg] * scale;
           if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
-          const bool dead = key > row || key >= L || row >= L ||
+          const bool dead = key > row || key >= Lk || qrow >= L ||
                             (window > 0 && key <= row - window);
           x = dead ? -1e30f : x;
           p[reg][ct] = x;

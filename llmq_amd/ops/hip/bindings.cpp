@@ -329,7 +329,8 @@ void paged_decode_attention(at::Tensor out, at::Tensor q, at::Tensor k_cache,
 
 template <typename T>
 void launch_prefill(at::Tensor& out, const at::Tensor& q, const at::Tensor& k,
-                    const at::Tensor& v, const at::Tensor& cu, int B,
+                    const at::Tensor& v, const at::Tensor& cu,
+                    const at::Tensor& cu_k, int B,
                     double scale, double softcap, long window, long max_seqlen) {
   const int H = q.size(1), D = q.size(2);
   const int KVH = k.size(1);
@@ -344,9 +345,9 @@ void launch_prefill(at::Tensor& out, const at::Tensor& q, const at::Tensor& k,
                          reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                          reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
                          reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
-                         cu.data_ptr<int>(), H, KVH, (float)scale,
-                         (float)softcap, (int)window, q.stride(0), k.stride(0),
-                         v.stride(0), out.stride(0));
+                         cu.data_ptr<int>(), cu_k.data_ptr<int>(), H, KVH,
+                         (float)scale, (float)softcap, (int)window,
+                         q.stride(0), k.stride(0), v.stride(0), out.stride(0));
     };
     switch (D) {
       case 64: lf.template operator()<64>(); return;
@@ -363,9 +364,9 @@ void launch_prefill(at::Tensor& out, const at::Tensor& q, const at::Tensor& k,
                        reinterpret_cast<const T*>(q.data_ptr()),
                        reinterpret_cast<const T*>(k.data_ptr()),
                        reinterpret_cast<const T*>(v.data_ptr()),
-                       cu.data_ptr<int>(), H, KVH, (float)scale, (float)softcap,
-                       (int)window, q.stride(0), k.stride(0), v.stride(0),
-                       out.stride(0));
+                       cu.data_ptr<int>(), cu_k.data_ptr<int>(), H, KVH,
+                       (float)scale, (float)softcap, (int)window,
+                       q.stride(0), k.stride(0), v.stride(0), out.stride(0));
   };
   switch (D) {
     case 64: l.template operator()<64>(); break;
@@ -377,15 +378,17 @@ void launch_prefill(at::Tensor& out, const at::Tensor& q, const at::Tensor& k,
 
 void varlen_prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,
                               at::Tensor v, at::Tensor cu_seqlens,
+                              at::Tensor cu_seqlens_k,
                               long max_seqlen, double scale, double softcap,
                               long window) {
   CHECK_GPU(q);
   CHECK_LASTDIM(q);
   TORCH_CHECK(cu_seqlens.scalar_type() == at::kInt);
+  TORCH_CHECK(cu_seqlens_k.scalar_type() == at::kInt);
   const int B = cu_seqlens.size(0) - 1;
   dispatch_dtype(q, "varlen_prefill_attention", [&]<typename T>() {
-    launch_prefill<T>(out, q, k, v, cu_seqlens, B, scale, softcap, window,
-                      max_seqlen);
+    launch_prefill<T>(out, q, k, v, cu_seqlens, cu_seqlens_k, B, scale,
+                      softcap, window, max_seqlen);
   });
 }
 
@@ -422,7 +425,7 @@ TORCH_LIBRARY(llmq_amd, m) {
   m.def("rope_inplace(Tensor(a!) q, Tensor(b!) k, Tensor positions, Tensor cos_sin) -> ()");
   m.def("reshape_and_cache(Tensor key, Tensor value, Tensor(a!) k_cache, Tensor(b!) v_cache, Tensor slot_mapping) -> ()");
   m.def("paged_decode_attention(Tensor(a!) out, Tensor q, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor context_lens, float scale, float softcap, int window) -> ()");
-  m.def("varlen_prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, int max_seqlen, float scale, float softcap, int window) -> ()");
+  m.def("varlen_prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, Tensor cu_seqlens_k, int max_seqlen, float scale, float softcap, int window) -> ()");
   m.def("sample_gumbel_argmax(Tensor(a!) out, Tensor(b!) keys, Tensor logits, Tensor temps, int seed, int step) -> ()");
   m.def("norm_add_norm(Tensor(a!) x, Tensor(b!) residual, Tensor w_post, Tensor w_pre, float eps, float offset) -> ()");
   m.def("rope_and_cache(Tensor(a!) q, Tensor(b!) k, Tensor value, Tensor(c!) k_cache, Tensor(d!) v_cache, Tensor positions, Tensor cos_sin, Tensor slot_mapping) -> ()");

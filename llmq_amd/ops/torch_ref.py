@@ -138,29 +138,37 @@ def paged_decode_attention(
 
 
 def varlen_prefill_attention(
-    q: torch.Tensor,  # [T, num_heads, head_dim]
-    k: torch.Tensor,  # [T, kv_heads, head_dim]
+    q: torch.Tensor,  # [Tq, num_heads, head_dim]
+    k: torch.Tensor,  # [Tk, kv_heads, head_dim]
     v: torch.Tensor,
-    cu_seqlens: torch.Tensor,  # [B+1]
+    cu_seqlens: torch.Tensor,  # [B+1] query offsets
     scale: float,
     softcap: float = 0.0,
     window: int = 0,
+    cu_seqlens_k: "Optional[torch.Tensor]" = None,  # [B+1] key offsets
 ) -> torch.Tensor:
-    """Causal self-attention over packed variable-length sequences (reference)."""
+    """Causal attention over packed variable-length sequences (reference).
+
+    With cu_seqlens_k, each sequence's K/V may be LONGER than its Q (chunked
+    prefill: the queries are the last Tq positions of a Tk-long context);
+    query row i has absolute position (Tk - Tq) + i."""
     T, H, D = q.shape
     KVH = k.shape[1]
     group = H // KVH
+    cu_k = cu_seqlens if cu_seqlens_k is None else cu_seqlens_k
     out = torch.empty_like(q)
     for b in range(len(cu_seqlens) - 1):
         s, e = int(cu_seqlens[b]), int(cu_seqlens[b + 1])
-        L = e - s
-        qb = q[s:e].float()  # [L, H, D]
-        kb = k[s:e].float().repeat_interleave(group, dim=1)  # [L, H, D]
-        vb = v[s:e].float().repeat_interleave(group, dim=1)
+        sk, ek = int(cu_k[b]), int(cu_k[b + 1])
+        Lq, Lk = e - s, ek - sk
+        off = Lk - Lq  # absolute position of q row 0
+        qb = q[s:e].float()  # [Lq, H, D]
+        kb = k[sk:ek].float().repeat_interleave(group, dim=1)  # [Lk, H, D]
+        vb = v[sk:ek].float().repeat_interleave(group, dim=1)
         scores = torch.einsum("ihd,jhd->hij", qb, kb) * scale
         scores = _softcap(scores, softcap)
-        i = torch.arange(L, device=q.device).view(-1, 1)
-        j = torch.arange(L, device=q.device).view(1, -1)
+        i = torch.arange(Lq, device=q.device).view(-1, 1) + off
+        j = torch.arange(Lk, device=q.device).view(1, -1)
         mask = j > i
         if window:
             mask = mask | (j <= i - window)

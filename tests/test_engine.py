@@ -214,3 +214,46 @@ class TestMixedSteps:
         # execute it through the runner to cover the mixed forward
         tokens = engine.runner.execute_mixed(batch)
         assert tokens.shape[0] == 2
+
+
+class TestChunkedPrefill:
+    PROMPT = "a quick brown fox jumps over the lazy dog " * 4  # ~170 byte-tokens
+
+    def _greedy(self, **kw):
+        eng = make_engine(max_model_len=256, **kw)
+        return eng.generate_batch(
+            [self.PROMPT], SamplingParams(temperature=0.0, max_tokens=8,
+                                          ignore_eos=True))[0]
+
+    def test_long_prompt_chunks_match_whole(self):
+        """A prompt above max_prefill_tokens prefills in chunks (gathered
+        past + fresh rows) and must produce identical greedy output."""
+        whole = self._greedy(max_prefill_tokens=4096)
+        chunked = self._greedy(max_prefill_tokens=48)  # ~4 chunks
+        assert chunked == whole
+
+    def test_chunks_interleave_with_decode(self):
+        """Decode of a running sequence keeps stepping while a long prompt
+        prefills chunk-by-chunk (mixed steps carrying mid-prompt chunks),
+        and the chunked long prompt decodes exactly like an unchunked run."""
+        greedy = SamplingParams(temperature=0.0, max_tokens=12, ignore_eos=True)
+        solo = make_engine(max_prefill_tokens=4096, max_model_len=256)
+        expect_long = solo.generate_batch([self.PROMPT], greedy)[0]
+
+        eng = make_engine(max_prefill_tokens=48, max_model_len=256)
+        eng.add_request("short", prompt="hello", params=greedy)
+        eng.step()  # prefill short
+        eng.add_request("long", prompt=self.PROMPT, params=greedy)
+        saw_midchunk_mixed = False
+        res = {}
+        while eng.has_unfinished():
+            nrun = eng.scheduler.num_running
+            npre = len(eng.scheduler.prefilling)
+            if npre > 0 and nrun > npre:
+                saw_midchunk_mixed = True  # decode + chunk in flight together
+            for out in eng.step():
+                if out.finished:
+                    res[out.request_id] = out.text
+        assert saw_midchunk_mixed
+        assert res["long"] == expect_long
+        assert "short" in res
