@@ -456,3 +456,50 @@ class TestFp8KVCache:
                                                ignore_eos=True))
         assert len(outs) == 2
         assert eng.kv_cache.k[0].dtype == torch.float8_e4m3fn
+
+
+class TestChunkedPrefillAttention:
+    @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+    def test_chunk_queries_vs_ref(self, dtype):
+        """Q = last Lq rows of an Lk-long context (cu_seqlens_k path)."""
+        H, KVH, D = 8, 2, 128
+        torch.manual_seed(6)
+        q_lens = [16, 33, 1]
+        k_lens = [80, 33, 129]  # seq 2 is a pure continuation (Lq=1)
+        cu_q = torch.tensor([0, 16, 49, 50], dtype=torch.int32, device=DEV)
+        cu_k = torch.tensor([0, 80, 113, 242], dtype=torch.int32, device=DEV)
+        Tq, Tk = 50, 242
+        q = torch.randn(Tq, H, D, device=DEV, dtype=dtype)
+        k = torch.randn(Tk, KVH, D, device=DEV, dtype=dtype)
+        v = torch.randn(Tk, KVH, D, device=DEV, dtype=dtype)
+        scale = D ** -0.5
+        out = ops.varlen_prefill_attention(
+            q, k, v, cu_q, max(q_lens), scale, cu_seqlens_k=cu_k
+        )
+        ref = torch_ref.varlen_prefill_attention(
+            q.float().cpu(), k.float().cpu(), v.float().cpu(), cu_q.cpu(),
+            scale, cu_seqlens_k=cu_k.cpu(),
+        )
+        atol, rtol = TOL[dtype]
+        assert_close_to_f32_ref(out.cpu(), ref, 2 * atol, 5 * rtol)
+
+    def test_engine_chunked_matches_whole_gpu(self):
+        from llmq_amd.engine.config import EngineConfig
+        from llmq_amd.engine.engine import LLMEngine
+        from llmq_amd.engine.sampling_params import SamplingParams
+
+        prompt = "the quick brown fox " * 20  # ~400 byte tokens
+        greedy = SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True)
+
+        def run(budget):
+            eng = LLMEngine(EngineConfig(
+                model="tiny-llama-d128", max_num_seqs=4, max_model_len=512,
+                load_weights=False, num_kv_blocks=256,
+                max_prefill_tokens=budget, enforce_eager=True,
+            ))
+            out = eng.generate_batch([prompt], greedy)[0]
+            del eng
+            torch.cuda.empty_cache()
+            return out
+
+        assert run(4096) == run(96)
