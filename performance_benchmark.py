@@ -77,11 +77,23 @@ def gpu_info() -> Dict[str, Any]:
     return {"tool": None, "raw": {}}
 
 
-def synthetic_jobs(n: int, prompt_tokens: int) -> List[Dict[str, Any]]:
-    """Deterministic synthetic prompts of ~prompt_tokens byte-tokens."""
+def synthetic_jobs(
+    n: int, prompt_tokens: int, long_frac: float = 0.0, long_tokens: int = 0
+) -> List[Dict[str, Any]]:
+    """Deterministic synthetic prompts of ~prompt_tokens byte-tokens;
+    every ⌈1/long_frac⌉-th job gets a long_tokens prompt (mixed workloads —
+    exercises chunked prefill + mixed steps)."""
     base = "the quick brown fox jumps over the lazy dog. "
-    body = (base * (prompt_tokens // len(base) + 1))[:prompt_tokens]
-    return [{"id": f"bench-{i:08d}", "prompt": f"[{i}] {body}"} for i in range(n)]
+
+    def body(tokens: int) -> str:
+        return (base * (tokens // len(base) + 1))[:tokens]
+
+    stride = int(1 / long_frac) if long_frac > 0 else 0
+    out = []
+    for i in range(n):
+        tokens = long_tokens if (stride and i % stride == 0) else prompt_tokens
+        out.append({"id": f"bench-{i:08d}", "prompt": f"[{i}] {body(tokens)}"})
+    return out
 
 
 class Orchestrator:
@@ -199,7 +211,10 @@ class Orchestrator:
         workers = self.start_workers(max_num_seqs, queue)
         self.wait_worker_ready()
 
-        jobs_data = synthetic_jobs(self.args.samples, self.args.prompt_tokens)
+        jobs_data = synthetic_jobs(
+            self.args.samples, self.args.prompt_tokens,
+            self.args.long_frac, self.args.long_prompt_tokens,
+        )
         jobs = [
             Job(max_tokens=self.args.max_tokens, temperature=self.args.temperature, **j)
             for j in jobs_data
@@ -273,6 +288,9 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=0, help="GPUs to spread workers over (0=CPU)")
     ap.add_argument("--max-tokens", type=int, default=128)
     ap.add_argument("--prompt-tokens", type=int, default=512)
+    ap.add_argument("--long-frac", type=float, default=0.0,
+                    help="fraction of jobs given a long prompt (mixed workload)")
+    ap.add_argument("--long-prompt-tokens", type=int, default=8192)
     ap.add_argument("--temperature", type=float, default=0.7)
     ap.add_argument("--max-model-len", type=int, default=None)
     ap.add_argument("--prefetch", type=int, default=1250,
