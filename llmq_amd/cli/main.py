@@ -16,7 +16,7 @@ Operation modules are imported lazily (reference main.py:102,290-294) so
 from __future__ import annotations
 
 import sys
-from typing import Optional, Tuple
+from typing import Tuple
 
 import click
 

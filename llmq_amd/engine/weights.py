@@ -8,7 +8,6 @@ GEMM tensors. Tensor-parallel loads take each rank's shard only.
 
 from __future__ import annotations
 
-import json
 import logging
 from pathlib import Path
 from typing import Dict, List

@@ -7,7 +7,6 @@ They are deliberately simple; nothing here runs in the GPU hot loop.
 
 from __future__ import annotations
 
-import math
 from typing import Optional, Tuple
 
 import torch
