@@ -69,8 +69,12 @@ class ModelSpec:
         return 1.0 / math.sqrt(self.head_dim)
 
     def layer_uses_sliding_window(self, layer_idx: int) -> bool:
-        # Gemma-2: layers 0,2,4,... are sliding-window (local) layers.
-        return self.sliding_window > 0 and layer_idx % 2 == 0
+        if self.sliding_window <= 0:
+            return False
+        # Gemma-2 alternates local/global layers; Mistral windows every layer.
+        if self.family == "gemma2":
+            return layer_idx % 2 == 0
+        return True
 
     def param_count(self) -> int:
         embed = self.vocab_size * self.hidden_size
@@ -151,6 +155,10 @@ PRESETS = {
     "qwen2.5-72b": _qwen2("qwen2.5-72b", 152064, 8192, 29568, 80, 64, 8,
                           max_position_embeddings=32768,
                           eos_token_id=151645, bos_token_id=151643),
+    # Mistral-7B (llama layout, 4096-token sliding window on every layer)
+    "mistral-7b": _llama("mistral-7b", 32000, 4096, 14336, 32, 32, 8,
+                         rope_theta=10000.0, max_position_embeddings=32768,
+                         sliding_window=4096, eos_token_id=2, bos_token_id=1),
     # Gemma-2 (Tower-Plus-9B is built on Gemma-2-9B)
     "gemma-2-9b": _gemma2("gemma-2-9b", 256000, 3584, 14336, 42, 16, 8, 256,
                           max_position_embeddings=8192),
@@ -183,6 +191,8 @@ def spec_from_hf_config(path: Path, name: str) -> ModelSpec:
         family = "gemma2"
     elif "qwen2" in arch or model_type == "qwen2":
         family = "qwen2"
+    elif "mistral" in arch or model_type == "mistral":
+        family = "mistral"  # llama layout + sliding window on every layer
     else:
         family = "llama"
     heads = cfg["num_attention_heads"]
@@ -207,7 +217,9 @@ def spec_from_hf_config(path: Path, name: str) -> ModelSpec:
         post_norms=family == "gemma2",
         attn_softcap=cfg.get("attn_logit_softcapping") or 0.0,
         final_softcap=cfg.get("final_logit_softcapping") or 0.0,
-        sliding_window=(cfg.get("sliding_window") or 0) if family == "gemma2" else 0,
+        sliding_window=(
+            (cfg.get("sliding_window") or 0) if family in ("gemma2", "mistral") else 0
+        ),
         embedding_scale=family == "gemma2",
         attn_scale=(
             cfg.get("query_pre_attn_scalar") and 1.0 / math.sqrt(cfg["query_pre_attn_scalar"])

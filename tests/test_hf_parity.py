@@ -28,6 +28,14 @@ def _save_tiny(tmp_path, family="llama"):
             tie_word_embeddings=False,
         )
         model = transformers.Qwen2ForCausalLM(cfg)
+    elif family == "mistral":
+        cfg = transformers.MistralConfig(
+            vocab_size=300, hidden_size=64, intermediate_size=128,
+            num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+            max_position_embeddings=256, rope_theta=10000.0, rms_norm_eps=1e-6,
+            sliding_window=32, tie_word_embeddings=False,
+        )
+        model = transformers.MistralForCausalLM(cfg)
     elif family == "gemma2":
         cfg = transformers.Gemma2Config(
             vocab_size=300, hidden_size=64, intermediate_size=128,
@@ -53,7 +61,7 @@ def _save_tiny(tmp_path, family="llama"):
     return model
 
 
-@pytest.mark.parametrize("family", ["llama", "qwen2", "gemma2"])
+@pytest.mark.parametrize("family", ["llama", "qwen2", "gemma2", "mistral"])
 def test_logits_match_transformers(tmp_path, family):
     hf = _save_tiny(tmp_path, family=family)
 
@@ -69,6 +77,9 @@ def test_logits_match_transformers(tmp_path, family):
     assert spec.num_heads == 4 and spec.num_kv_heads == 2
     assert spec.qkv_bias == (family == "qwen2")
     assert spec.post_norms == (family == "gemma2")
+    if family == "mistral":
+        assert spec.sliding_window == 32
+        assert spec.layer_uses_sliding_window(0) and spec.layer_uses_sliding_window(1)
 
     token_ids = [1, 7, 42, 99, 123, 250, 3]
     # in-tree forward: run a prefill step and capture the logits
