@@ -237,6 +237,8 @@ class LLMEngine:
             chunks = batch.chunks or [(0, sq.num_tokens) for sq in pseqs]
             final = []
             for sq, (st, e) in zip(pseqs, chunks):
+                if st == 0:
+                    sq.prefill_start_time = now
                 sq.prefilled = e
                 if e == sq.num_tokens:
                     final.append(sq)
@@ -274,7 +276,9 @@ class LLMEngine:
                 out.finish_reason = reason
                 out.text = self._final_text(seq)
                 prefill_at = self._prefill_done_at.pop(seq.request_id, now)
-                out.queue_wait_ms = (prefill_at - seq.arrival_time) * 1000.0
+                started = seq.prefill_start_time or prefill_at
+                out.queue_wait_ms = (started - seq.arrival_time) * 1000.0
+                out.prefill_ms = (prefill_at - started) * 1000.0
                 out.decode_ms = (now - prefill_at) * 1000.0
                 self._seqs.pop(seq.request_id, None)
             outputs.append(out)

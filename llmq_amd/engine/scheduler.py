@@ -38,7 +38,7 @@ class Sequence:
     __slots__ = (
         "request_id", "token_ids", "prompt_len", "params", "status",
         "block_table", "arrival_time", "first_token_time", "finish_reason",
-        "output_text", "num_preemptions", "prefilled",
+        "output_text", "num_preemptions", "prefilled", "prefill_start_time",
     )
 
     def __init__(self, request_id: str, prompt_token_ids: List[int], params: SamplingParams,
@@ -55,6 +55,7 @@ class Sequence:
         self.output_text = None  # set only when a stop string truncates
         self.num_preemptions = 0
         self.prefilled = 0  # context tokens whose KV is in the cache
+        self.prefill_start_time: Optional[float] = None
 
     @property
     def num_tokens(self) -> int:

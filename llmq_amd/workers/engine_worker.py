@@ -53,6 +53,7 @@ class _FinishedRequest:
     output_tokens: int
     finish_reason: Optional[str]
     queue_wait_ms: Optional[float]
+    prefill_ms: Optional[float]
     decode_ms: Optional[float]
 
 
@@ -201,6 +202,7 @@ class AsyncEngineBridge:
                                 output_tokens=out.output_tokens,
                                 finish_reason=out.finish_reason,
                                 queue_wait_ms=out.queue_wait_ms,
+                                prefill_ms=out.prefill_ms,
                                 decode_ms=out.decode_ms,
                             ),
                         )
@@ -462,5 +464,6 @@ class EngineWorker(BaseWorker):
             result.output_tokens = stats.output_tokens
             result.finish_reason = stats.finish_reason
             result.queue_wait_ms = stats.queue_wait_ms
+            result.prefill_ms = stats.prefill_ms
             result.decode_ms = stats.decode_ms
         return result
