@@ -52,6 +52,12 @@ class ResultReceiver:
             await delivery.nack(requeue=False, error=f"bad result: {exc}")
             return
         self._last_seen = time.time()
+        if self.limit is not None and self.received >= self.limit:
+            # over-delivered (prefetch window): leave it queued for the next
+            # `llmq receive` — results are durable/resumable
+            await delivery.nack(requeue=True)
+            self._stop.set()
+            return
         if self.skip_filtered and getattr(result, "filtered", False):
             await delivery.ack()
             return

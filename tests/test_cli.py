@@ -9,6 +9,7 @@ from __future__ import annotations
 
 import asyncio
 import json
+import os
 import threading
 
 import pytest
@@ -179,3 +180,46 @@ def test_submit_local_dataset_with_map(broker_env, tmp_path):
     jobs = asyncio.new_event_loop().run_until_complete(drain())
     prompts = sorted(j.get_formatted_prompt() for j in jobs)
     assert prompts[0] == "Translate from nl: zin 0"
+
+
+def test_submit_stream_echoes_results(broker_env, tmp_path):
+    """submit --stream consumes results inline (reference submit.py:266-305)."""
+    jobs = tmp_path / "j.jsonl"
+    jobs.write_text(json.dumps({"id": "s1", "text": "ping"}) + "\n")
+
+    # a worker must answer: run a dummy worker thread on the broker loop
+    import subprocess
+    import sys
+    worker = subprocess.Popen(
+        [sys.executable, "-m", "llmq_amd", "worker", "dummy", "cls", "--delay", "0"],
+        env={**os.environ, "LLMQ_BROKER_URL": broker_env},
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+    )
+    try:
+        runner = CliRunner()
+        res = runner.invoke(cli, ["submit", "cls", str(jobs),
+                                  "--template", "say {text}", "--stream"])
+        assert res.exit_code == 0, res.output
+        lines = [json.loads(l) for l in res.output.splitlines() if l.startswith("{")]
+        assert any(r["id"] == "s1" and r["result"] == "echo say ping" for r in lines), res.output
+    finally:
+        worker.terminate()
+        worker.wait(timeout=10)
+
+
+def test_receive_limit(broker_env):
+    async def seed():
+        c = _client(broker_env)
+        await c.connect()
+        await c.setup_queue_infrastructure("clim")
+        for i in range(5):
+            await c.publish_result("clim", Result(
+                id=f"m{i}", prompt="p", result="r", worker_id="w", duration_ms=1.0))
+        await c.disconnect()
+
+    asyncio.new_event_loop().run_until_complete(seed())
+    runner = CliRunner()
+    res = runner.invoke(cli, ["receive", "clim", "--timeout", "3", "--limit", "2"])
+    assert res.exit_code == 0, res.output
+    lines = [l for l in res.output.splitlines() if l.startswith("{")]
+    assert len(lines) == 2
