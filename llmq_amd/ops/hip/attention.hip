@@ -754,9 +754,11 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
       const int dtile = (wid * D4) / 16 + dt;
 #pragma unroll
       for (int ks = 0; ks < PD_KT / 32; ++ks) {
-        // two transpose reads give the full 8-key A-fragment: uniform base,
-        // hardware distributes [4-key][16-dim] blocks across lane groups
-        const int sub = (ks * (D / 16) + dtile) * 8 * 64;
+        // two transpose reads give the full 8-key A-fragment: each lane
+        // reads 4 contiguous bf16 at its own 8-byte slot and the hardware
+        // redistributes so lane l receives column (l&15) of its 16-lane
+        // group's [4-key][16-dim] block
+        const int sub = (ks * (D / 16) + dtile) * 8 * 64 + lane * 4;
         bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
             (lds_bf16x4*)&kv_lds[sub]);
         bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
