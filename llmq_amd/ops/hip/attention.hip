@@ -399,17 +399,24 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
         const int key = c / (D / CHW);
         const int d8 = (c % (D / CHW)) * CHW;
         const int dst = key * D + swz(key, d8);
+        // V uses the tr-read subtile image (quad-permuted 32-key × 16-dim
+        // blocks) so PV A-fragments load via ds_read_b64_tr_b16
+        const int dtile = d8 / 16, col0 = d8 & 15;
+        const int qq = (key & 31) >> 2;
+        const int bpos = ((qq & 1) << 2) + (qq >> 1);
+        const int vdst = (((key >> 5) * (D / 16) + dtile) * 8 + bpos) * 64 +
+                         (key & 3) * 16 + col0;
         const int gkey = kt + key;
         if (gkey < Lk) {
           *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) =
               *reinterpret_cast<const bf16x8_t*>(
                   k + (long)(s0k + gkey) * k_stride + (long)kvh * D + d8);
-          *reinterpret_cast<bf16x8_t*>(&v_lds[dst]) =
+          *reinterpret_cast<bf16x8_t*>(&v_lds[vdst]) =
               *reinterpret_cast<const bf16x8_t*>(
                   v + (long)(s0k + gkey) * v_stride + (long)kvh * D + d8);
         } else {
           *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
-          *reinterpret_cast<bf16x8_t*>(&v_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+          *reinterpret_cast<bf16x8_t*>(&v_lds[vdst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
         }
       }
     }
@@ -486,12 +493,16 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
         ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
 #pragma unroll
         for (int ks = 0; ks < PF_KT / 32; ++ks) {
+          const int sub = (ks * (D / 16) + dt) * 8 * 64 + lane * 4;
+          bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (lds_bf16x4*)&v_lds[sub]);
+          bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (lds_bf16x4*)&v_lds[sub + 4 * 64]);
           bf16x8_t a;
-          const int dim = dt * 16 + col;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int key = ks * 32 + kgrp * 8 + j;
-            a[j] = v_lds[key * D + swz(key, dim & ~7) + (dim & 7)];
+          for (int j = 0; j < 4; ++j) {
+            a[j] = __bfloat16_as_short((__hip_bfloat16)lo[j]);
+            a[4 + j] = __bfloat16_as_short((__hip_bfloat16)hi[j]);
           }
           bf16x8_t b = *reinterpret_cast<const bf16x8_t*>(
               &p_lds[wid][col * PF_KT + ks * 32 + kgrp * 8]);
