@@ -357,7 +357,8 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
 
   // LDS: K tile | V tile | per-wave P tiles | per-wave row stats
   __shared__ __attribute__((aligned(16))) short k_lds[PF_KT * D];
-  __shared__ __attribute__((aligned(16))) short v_lds[PF_KT * D];
+  // V: tr-subtile image, 528-elem stride per subtile (bank rotation)
+  __shared__ __attribute__((aligned(16))) short v_lds[(PF_KT / 32) * (D / 16) * 528];
   __shared__ __attribute__((aligned(16))) short p_lds[4][16 * PF_KT];
   __shared__ float stat_lds[4][2][16];  // [wave][alpha|inv_l][row]
 
@@ -404,7 +405,7 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
         const int dtile = d8 / 16, col0 = d8 & 15;
         const int qq = (key & 31) >> 2;
         const int bpos = ((qq & 1) << 2) + (qq >> 1);
-        const int vdst = (((key >> 5) * (D / 16) + dtile) * 8 + bpos) * 64 +
+        const int vdst = ((key >> 5) * (D / 16) + dtile) * 528 + bpos * 64 +
                          (key & 3) * 16 + col0;
         const int gkey = kt + key;
         if (gkey < Lk) {
@@ -493,7 +494,7 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
         ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
 #pragma unroll
         for (int ks = 0; ks < PF_KT / 32; ++ks) {
-          const int sub = (ks * (D / 16) + dt) * 8 * 64 + lane * 4;
+          const int sub = (ks * (D / 16) + dt) * 528 + lane * 4;
           bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
               (lds_bf16x4*)&v_lds[sub]);
           bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
@@ -607,7 +608,12 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
   const int col = lane & 15;   // MFMA col lane (key for S, qrow for OT)
   const int kgrp = lane >> 4;  // 0..3
 
-  __shared__ __attribute__((aligned(16))) short kv_lds[PD_KT * D];  // K then V
+  // V's tr-subtile image strides subtiles by 528 elems (512 + 16): the
+  // unpadded 1 KB stride lands every subtile on the same banks (16-way
+  // staging-write conflicts). Sized for max(K rows, padded V image).
+  constexpr int NSUB = (PD_KT / 32) * (D / 16);
+  constexpr int KV_ELEMS = (PD_KT * D > NSUB * 528) ? PD_KT * D : NSUB * 528;
+  __shared__ __attribute__((aligned(16))) short kv_lds[KV_ELEMS];  // K then V
   __shared__ __attribute__((aligned(16))) short p_lds[16 * PD_KT];  // [qrow][key]
   __shared__ float mpart_lds[NW][16];  // per-wave row-max partials
   __shared__ float alpha_lds[16];      // per-row rescale for the OT lanes
@@ -720,7 +726,7 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
       const int dt = d8 / 16, col0 = d8 & 15;
       const int qq = (key & 31) >> 2;
       const int block_pos = ((qq & 1) << 2) + (qq >> 1);
-      const int dst = (((key >> 5) * (D / 16) + dt) * 8 + block_pos) * 64 +
+      const int dst = ((key >> 5) * (D / 16) + dt) * 528 + block_pos * 64 +
                       (key & 3) * 16 + col0;
       const int gkey = base + key;
       if (gkey < L) {
@@ -769,7 +775,7 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
         // reads 4 contiguous bf16 at its own 8-byte slot and the hardware
         // redistributes so lane l receives column (l&15) of its 16-lane
         // group's [4-key][16-dim] block
-        const int sub = (ks * (D / 16) + dtile) * 8 * 64 + lane * 4;
+        const int sub = (ks * (D / 16) + dtile) * 528 + lane * 4;
         bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
             (lds_bf16x4*)&kv_lds[sub]);
         bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
