@@ -104,14 +104,18 @@ def main() -> None:
         engine.add_request(f"bench-{rank}-{i}", prompt_token_ids=ids, params=params)
 
     t_prefill = time.perf_counter()
-    while engine.scheduler.num_waiting > 0:
-        engine.step()  # prefill batches
+    while engine.scheduler.num_waiting > 0 or engine.scheduler.prefilling:
+        engine.step()  # admission steps (prefill + ride-along decode)
     prefill_s = time.perf_counter() - t_prefill
     prefill_tokens = batch * prompt_len
+    # mixed steps decode already-admitted seqs while later ones prefill, so
+    # the phase also produced decode tokens — count both for the rate.
+    decoded = sum(sq.num_tokens for sq in engine.scheduler.running) - prefill_tokens
     if rank == 0:
         print(
-            f"[bench] prefill {prefill_tokens} tokens in {prefill_s:.2f}s "
-            f"({prefill_tokens / prefill_s:.0f} tok/s/gpu)",
+            f"[bench] admission phase: {prefill_tokens} prefill + {decoded} decode "
+            f"tokens in {prefill_s:.2f}s "
+            f"({(prefill_tokens + decoded) / prefill_s:.0f} tok/s/gpu)",
             file=sys.stderr,
         )
 
