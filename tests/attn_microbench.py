@@ -36,8 +36,12 @@ def main() -> None:
     ap.add_argument("--softcap", type=float, default=0.0)
     ap.add_argument("--window", type=int, default=0)
     ap.add_argument("--check", action="store_true", help="verify vs torch ref")
+    ap.add_argument("--prefill", action="store_true",
+                    help="time varlen flash prefill instead of decode")
     args = ap.parse_args()
 
+    if args.prefill:
+        return prefill_bench(args)
     use_gpu = torch.cuda.is_available()
     if use_gpu:
         assert ops.has_hip_ext()
@@ -94,6 +98,44 @@ def main() -> None:
         err = (out.float() - ref).abs().max().item()
         print(f"max|err| vs f32 ref: {err:.4e}")
         assert err < 0.05
+
+
+def prefill_bench(args) -> None:
+    use_gpu = torch.cuda.is_available()
+    dev = torch.device("cuda:0" if use_gpu else "cpu")
+    dtype = torch.bfloat16 if use_gpu else torch.float32
+    B, L, H, KVH, D = 8, args.ctx, args.heads, args.kv_heads, args.head_dim
+    torch.manual_seed(0)
+    T = B * L
+    q = torch.randn(T, H, D, device=dev, dtype=dtype)
+    k = torch.randn(T, KVH, D, device=dev, dtype=dtype)
+    v = torch.randn(T, KVH, D, device=dev, dtype=dtype)
+    cu = torch.arange(0, T + 1, L, dtype=torch.int32, device=dev)
+    scale = D ** -0.5
+
+    def run():
+        return ops.varlen_prefill_attention(q, k, v, cu, L, scale)
+
+    for _ in range(args.warmup):
+        run()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.iters):
+        run()
+    if use_gpu:
+        torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / args.iters
+    flops = 4.0 * B * H * D * (L * (L + 1) / 2)  # causal QK^T + PV
+    print(f"prefill B={B} L={L} H={H} KVH={KVH} D={D}")
+    print(f"{dt * 1e3:.3f} ms/iter   {flops / dt / 1e12:.1f} TF/s")
+    if args.check:
+        from llmq_amd.ops import torch_ref
+
+        ref = torch_ref.varlen_prefill_attention(
+            q.float().cpu(), k.float().cpu(), v.float().cpu(), cu.cpu(), scale)
+        err = (run().float().cpu() - ref).abs().max().item()
+        print(f"max|err|: {err:.4e}")
 
 
 if __name__ == "__main__":
