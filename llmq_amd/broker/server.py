@@ -299,6 +299,16 @@ class BrokerServer:
                 self._ack_internal(q, msg)
                 self._dead_letter(q, msg, "expired: TTL exceeded")
                 continue
+            if msg.attempts > q.max_retries:
+                # poison job that kills its consumer: disconnect-requeued
+                # deliveries count too, so the retry cap holds even when no
+                # explicit nack ever arrives
+                self._ack_internal(q, msg)
+                self._dead_letter(
+                    q, msg, f"max retries exceeded ({q.max_retries}): "
+                    "consumer lost repeatedly",
+                )
+                continue
             tag = target.conn.next_tag()
             msg.attempts += 1
             q.unacked[(target.conn.id, tag)] = msg

@@ -342,3 +342,41 @@ def test_worker_heartbeat_registry():
             await client.disconnect()
 
     run_async(main())
+
+
+def test_crash_looping_consumer_dead_letters():
+    """A job whose consumer DISCONNECTS (no nack) every delivery must hit
+    the DLQ after max_retries — crash-loop poison jobs cannot cycle forever."""
+
+    async def main():
+        async with live_broker(max_retries=2) as (server, config):
+            pub = BrokerClient(config)
+            await pub.connect()
+            await pub.setup_queue_infrastructure("crashq")
+            await pub.publish_jobs("crashq", [Job(id="poison", prompt="x")])
+
+            for _ in range(4):  # a few crash-loops
+                victim = BrokerClient(config)
+                await victim.connect()
+                got = asyncio.Event()
+
+                async def cb(delivery):
+                    got.set()  # neither ack nor nack: simulate crash
+
+                await victim.consume_jobs("crashq", cb, prefetch=1)
+                try:
+                    await asyncio.wait_for(got.wait(), 5)
+                except asyncio.TimeoutError:
+                    await victim.disconnect()
+                    break  # no more deliveries: already dead-lettered
+                await victim.disconnect()  # unacked → requeue, attempts += 1
+                await asyncio.sleep(0.05)
+
+            failed = await pub.get_failed_messages("crashq", limit=10)
+            assert len(failed) == 1
+            assert failed[0].job_id == "poison"
+            stats = await pub.get_queue_stats("crashq")
+            assert stats.message_count == 0  # gone from the main queue
+            await pub.disconnect()
+
+    run_async(main())
