@@ -533,7 +533,4 @@ class TestPreemptionGPU:
         assert len(outs) == 8
         for o in outs.values():
             assert o.output_tokens == 64
-        preempts = sum(
-            1 for _ in []
-        )  # engine frees Sequence objects; assert via allocator instead
         assert eng.allocator.num_free == eng.allocator.num_blocks
