@@ -328,7 +328,7 @@ DEVINL int swz(int row, int d) {  // element-index XOR swizzle (16B granules)
 // (gathered past + fresh chunk) while Q is only the new chunk, and q row i
 // sits at absolute position (Lk - Lq) + i.
 template <int HEAD_DIM>
-__global__ __launch_bounds__(256, 2) void flash_prefill_bf16_kernel(
+__global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
     __hip_bfloat16* __restrict__ out,      // [Tq, H, D]
     const __hip_bfloat16* __restrict__ q,  // [Tq, H, D]
     const __hip_bfloat16* __restrict__ k,  // [Tk, KVH, D]
@@ -391,56 +391,37 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_bf16_kernel(
     kv_begin = max(0, (wave_min_needed / PF_KT) * PF_KT);
   }
 
-  // ---- async staging: each tile's K/V global loads are issued one tile
-  // AHEAD into registers (they fly under the previous tile's S/softmax/PV
-  // MFMA work — the guide's async-STAGE idiom) and land in LDS after the
-  // barrier. NPC 16-byte pieces per thread per tile.
-  constexpr int CHW = 8;
-  constexpr int NPC = PF_KT * D / CHW / 256;
-  bf16x8_t kreg[NPC], vreg[NPC];
-
-  auto issue_tile = [&](int kt) {
-#pragma unroll
-    for (int pc = 0; pc < NPC; ++pc) {
-      const int c = pc * 256 + tid;
-      const int key = c / (D / CHW);
-      const int d8 = (c % (D / CHW)) * CHW;
-      const int gkey = kt + key;
-      if (gkey < Lk) {
-        kreg[pc] = *reinterpret_cast<const bf16x8_t*>(
-            k + (long)(s0k + gkey) * k_stride + (long)kvh * D + d8);
-        vreg[pc] = *reinterpret_cast<const bf16x8_t*>(
-            v + (long)(s0k + gkey) * v_stride + (long)kvh * D + d8);
-      } else {
-        kreg[pc] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
-        vreg[pc] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+  for (int kt = kv_begin; kt < kv_end; kt += PF_KT) {
+    // ---- stage K/V tile (zero-fill beyond L so garbage never reaches MFMA)
+    {
+      constexpr int CHW = 8;  // elems per 16B chunk
+      const int chunks = PF_KT * D / CHW;
+      for (int c = tid; c < chunks; c += 256) {
+        const int key = c / (D / CHW);
+        const int d8 = (c % (D / CHW)) * CHW;
+        const int dst = key * D + swz(key, d8);
+        // V uses the tr-read subtile image (quad-permuted 32-key × 16-dim
+        // blocks) so PV A-fragments load via ds_read_b64_tr_b16
+        const int dtile = d8 / 16, col0 = d8 & 15;
+        const int qq = (key & 31) >> 2;
+        const int bpos = ((qq & 1) << 2) + (qq >> 1);
+        const int vdst = ((key >> 5) * (D / 16) + dtile) * 528 + bpos * 64 +
+                         (key & 3) * 16 + col0;
+        const int gkey = kt + key;
+        if (gkey < Lk) {
+          *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) =
+              *reinterpret_cast<const bf16x8_t*>(
+                  k + (long)(s0k + gkey) * k_stride + (long)kvh * D + d8);
+          *reinterpret_cast<bf16x8_t*>(&v_lds[vdst]) =
+              *reinterpret_cast<const bf16x8_t*>(
+                  v + (long)(s0k + gkey) * v_stride + (long)kvh * D + d8);
+        } else {
+          *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+          *reinterpret_cast<bf16x8_t*>(&v_lds[vdst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+        }
       }
     }
-  };
-  auto write_tile = [&]() {
-#pragma unroll
-    for (int pc = 0; pc < NPC; ++pc) {
-      const int c = pc * 256 + tid;
-      const int key = c / (D / CHW);
-      const int d8 = (c % (D / CHW)) * CHW;
-      const int dst = key * D + swz(key, d8);
-      // V uses the tr-read subtile image (quad-permuted 32-key × 16-dim
-      // blocks) so PV A-fragments load via ds_read_b64_tr_b16
-      const int dtile = d8 / 16, col0 = d8 & 15;
-      const int qq = (key & 31) >> 2;
-      const int bpos = ((qq & 1) << 2) + (qq >> 1);
-      const int vdst = ((key >> 5) * (D / 16) + dtile) * 528 + bpos * 64 +
-                       (key & 3) * 16 + col0;
-      *reinterpret_cast<bf16x8_t*>(&k_lds[dst]) = kreg[pc];
-      *reinterpret_cast<bf16x8_t*>(&v_lds[vdst]) = vreg[pc];
-    }
-  };
-
-  if (kv_begin < kv_end) issue_tile(kv_begin);
-  for (int kt = kv_begin; kt < kv_end; kt += PF_KT) {
-    write_tile();
     __syncthreads();
-    if (kt + PF_KT < kv_end) issue_tile(kt + PF_KT);
 
     if (kt <= wave_max_row) {  // this wave has rows that see this tile
       // ---- S = Q K^T for 4 column tiles of 16 keys
