@@ -223,3 +223,32 @@ def test_receive_limit(broker_env):
     assert res.exit_code == 0, res.output
     lines = [l for l in res.output.splitlines() if l.startswith("{")]
     assert len(lines) == 2
+
+
+def test_pipeline_status_view(broker_env, tmp_path):
+    """`llmq status -p pipeline.yaml` renders the per-stage flow view."""
+    cfg = tmp_path / "p.yaml"
+    cfg.write_text("""
+name: demo
+stages:
+  - name: first
+    worker: dummy
+  - name: second
+    worker: dummy
+""")
+
+    async def seed():
+        from llmq_amd.core.pipeline import PipelineConfig
+        c = _client(broker_env)
+        await c.connect()
+        pipeline = PipelineConfig.from_yaml_file(str(cfg))
+        await c.setup_pipeline_infrastructure(pipeline)
+        await c.publish_job(pipeline.get_stage_queue_name("first"),
+                            Job(id="x", prompt="p"))
+        await c.disconnect()
+
+    asyncio.new_event_loop().run_until_complete(seed())
+    runner = CliRunner()
+    res = runner.invoke(cli, ["status", "-p", str(cfg)])
+    assert res.exit_code == 0, res.output
+    assert "first" in res.output and "second" in res.output

@@ -274,3 +274,18 @@ def test_bridge_engine_crash_fails_inflight():
         bridge.shutdown()
 
     run_async(main())
+
+
+def test_resolve_tp_precedence(monkeypatch):
+    """auto-TP = all visible GPUs (reference vllm_worker.py:62-89); explicit
+    and stage-config values take precedence."""
+    w = EngineWorker.__new__(EngineWorker)
+    w.stage_config = {}
+    w._tp = 4
+    assert w._resolve_tp() == 4
+    w._tp = None
+    w.stage_config = {"tensor_parallel_size": 2}
+    assert w._resolve_tp() == 2
+    w.stage_config = {}
+    # no GPU in this container → 1
+    assert w._resolve_tp() == 1
