@@ -172,7 +172,10 @@ class LLMEngine:
         if num < 16:
             raise RuntimeError(
                 f"KV budget too small: {budget / 2**30:.1f} GB free for KV "
-                f"(gpu_memory_utilization={cfg.gpu_memory_utilization})"
+                f"(gpu_memory_utilization={cfg.gpu_memory_utilization}; the "
+                "utilization caps TOTAL device usage incl. other processes — "
+                "when sharing a GPU, raise it and bound KV via max_model_len/"
+                "max_num_seqs instead)"
             )
         return num
 

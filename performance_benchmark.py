@@ -178,6 +178,14 @@ class Orchestrator:
                 with open(p, "rb") as fh:
                     if b"starting to consume from queue" in fh.read():
                         ready.add(p)
+            # fail FAST if a worker process already died (e.g. KV budget
+            # misconfiguration) instead of burning the full timeout
+            for proc in self.procs:
+                if proc.poll() not in (None, 0):
+                    raise RuntimeError(
+                        f"worker exited with {proc.returncode} before ready "
+                        f"(see {self.args.log_dir}/worker*.log)"
+                    )
             time.sleep(0.5)
         if len(ready) < len(paths):
             raise RuntimeError(f"only {len(ready)}/{len(paths)} workers became ready")
