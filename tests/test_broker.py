@@ -380,3 +380,30 @@ def test_crash_looping_consumer_dead_letters():
             await pub.disconnect()
 
     run_async(main())
+
+
+def test_bodies_with_newlines_and_unicode_roundtrip():
+    """Frames are newline-delimited JSON: bodies containing raw newlines,
+    unicode, and JSON metacharacters must survive the wire intact."""
+
+    async def main():
+        async with live_broker() as (server, config):
+            c = BrokerClient(config)
+            await c.connect()
+            await c.setup_queue_infrastructure("wire")
+            nasty = 'line1\nline2\t"quoted" \\backslash 雪 🚀 \r\n end'
+            await c.publish_jobs("wire", [Job(id="n1", prompt=nasty)])
+            got = []
+            done = asyncio.Event()
+
+            async def cb(delivery):
+                got.append(Job.model_validate_json(delivery.body))
+                await delivery.ack()
+                done.set()
+
+            await c.consume_jobs("wire", cb, prefetch=1)
+            await asyncio.wait_for(done.wait(), 5)
+            assert got[0].prompt == nasty
+            await c.disconnect()
+
+    run_async(main())
