@@ -1,0 +1,92 @@
+"""Tensor-parallel NUMERICS: a 2-way sharded model (column/row-sharded
+QKV/O/MLP + all-reduce over gloo) must produce the same logits as the
+unsharded model loaded from the same HF checkpoint.
+
+This is the correctness proof behind `worker run -tp N` — the TP worker
+tests cover lockstep plumbing; this covers the sharded math itself
+(weights.py shard slicing + the two per-layer all-reduces).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+
+import pytest
+import torch
+
+pytestmark = [pytest.mark.integration, pytest.mark.slow]
+
+transformers = pytest.importorskip("transformers")
+
+
+def _save_checkpoint(tmp_path):
+    cfg = transformers.LlamaConfig(
+        vocab_size=320, hidden_size=128, intermediate_size=256,
+        num_hidden_layers=2, num_attention_heads=8, num_key_value_heads=4,
+        max_position_embeddings=256, rope_theta=10000.0, rms_norm_eps=1e-6,
+        tie_word_embeddings=False,
+    )
+    model = transformers.LlamaForCausalLM(cfg).eval().float()
+    model.save_pretrained(tmp_path, safe_serialization=True)
+    (tmp_path / "tokenizer_config.json").write_text(json.dumps({}))
+    return model
+
+
+def _forward_logits(engine, token_ids):
+    from llmq_amd.engine.forward_meta import PrefillMeta
+
+    dev = engine.device
+    blocks = engine.allocator.allocate(4)
+    bs = engine.config.kv_block_size
+    slots = [blocks[p // bs] * bs + p % bs for p in range(len(token_ids))]
+    meta = PrefillMeta(
+        cu_seqlens=torch.tensor([0, len(token_ids)], dtype=torch.int32, device=dev),
+        max_seqlen=len(token_ids),
+        slot_mapping=torch.tensor(slots, dtype=torch.long, device=dev),
+    )
+    ids = torch.tensor(token_ids, dtype=torch.long, device=dev)
+    pos = torch.arange(len(token_ids), dtype=torch.long, device=dev)
+    hidden = engine.model.forward(ids, pos, engine.kv_cache, meta)
+    return engine.model.compute_logits(hidden)
+
+
+def _rank_main(rank, world, ckpt, out_file):
+    import torch.distributed as dist
+
+    from llmq_amd.engine.config import EngineConfig
+    from llmq_amd.engine.engine import LLMEngine
+    from llmq_amd.parallel import init_tp
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29591"
+    init_tp(world, rank=rank, backend="gloo")
+    engine = LLMEngine(EngineConfig(
+        model=ckpt, device="cpu", enforce_eager=True,
+        max_num_seqs=2, max_model_len=128, num_kv_blocks=64,
+    ), tp_rank=rank, tp_size=world)
+    token_ids = [1, 9, 77, 123, 200, 314, 5, 42]
+    logits = _forward_logits(engine, token_ids)
+    if rank == 0:
+        torch.save(logits, out_file)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp2_logits_match_unsharded(tmp_path):
+    hf = _save_checkpoint(tmp_path)
+
+    import torch.multiprocessing as mp
+
+    out_file = str(tmp_path / "tp_logits.pt")
+    mp.spawn(_rank_main, args=(2, str(tmp_path), out_file), nprocs=2, join=True)
+    tp_logits = torch.load(out_file)
+
+    token_ids = [1, 9, 77, 123, 200, 314, 5, 42]
+    with torch.no_grad():
+        ref = hf(torch.tensor([token_ids])).logits[0]
+
+    diff = (tp_logits - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert diff < 1e-3 * max(1.0, scale), f"TP logits diverge: {diff}"
