@@ -67,8 +67,20 @@ def _rank_main(rank, world, ckpt, out_file):
     ), tp_rank=rank, tp_size=world)
     token_ids = [1, 9, 77, 123, 200, 314, 5, 42]
     logits = _forward_logits(engine, token_ids)
+
+    # end-to-end greedy decode in lockstep (sharded KV heads through the
+    # paged decode path; every rank samples identically)
+    from llmq_amd.engine.sampling_params import SamplingParams
+
+    engine.add_request("g", prompt_token_ids=token_ids,
+                       params=SamplingParams(temperature=0.0, max_tokens=6,
+                                             ignore_eos=True))
+    text_tokens = []
+    while engine.has_unfinished():
+        for out in engine.step():
+            text_tokens.extend(out.new_token_ids)
     if rank == 0:
-        torch.save(logits, out_file)
+        torch.save({"logits": logits, "decode_tokens": text_tokens}, out_file)
     dist.barrier()
     dist.destroy_process_group()
 
@@ -81,7 +93,8 @@ def test_tp2_logits_match_unsharded(tmp_path):
 
     out_file = str(tmp_path / "tp_logits.pt")
     mp.spawn(_rank_main, args=(2, str(tmp_path), out_file), nprocs=2, join=True)
-    tp_logits = torch.load(out_file)
+    saved = torch.load(out_file)
+    tp_logits = saved["logits"]
 
     token_ids = [1, 9, 77, 123, 200, 314, 5, 42]
     with torch.no_grad():
@@ -90,3 +103,22 @@ def test_tp2_logits_match_unsharded(tmp_path):
     diff = (tp_logits - ref).abs().max().item()
     scale = ref.abs().max().item()
     assert diff < 1e-3 * max(1.0, scale), f"TP logits diverge: {diff}"
+
+    # decode tokens must match a single-process engine on the same checkpoint
+    from llmq_amd.engine.config import EngineConfig
+    from llmq_amd.engine.engine import LLMEngine
+    from llmq_amd.engine.sampling_params import SamplingParams
+
+    solo = LLMEngine(EngineConfig(
+        model=str(tmp_path), device="cpu", enforce_eager=True,
+        max_num_seqs=2, max_model_len=128, num_kv_blocks=64,
+    ))
+    solo.add_request("g", prompt_token_ids=token_ids,
+                     params=SamplingParams(temperature=0.0, max_tokens=6,
+                                           ignore_eos=True))
+    solo_tokens = []
+    while solo.has_unfinished():
+        for out in solo.step():
+            solo_tokens.extend(out.new_token_ids)
+    assert saved["decode_tokens"] == solo_tokens, (
+        saved["decode_tokens"], solo_tokens)
