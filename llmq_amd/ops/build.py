@@ -38,15 +38,26 @@ def build(verbose: bool = False, force: bool = False) -> Path:
     from torch.utils.cpp_extension import load
 
     BUILD_DIR.mkdir(exist_ok=True)
-    load(
-        name="llmq_amd_hip_ops",
-        sources=[str(SRC)],
-        extra_cflags=["-O3", "-std=c++20"],
-        extra_cuda_cflags=["-O3", "-std=c++20"],
-        build_directory=str(BUILD_DIR),
-        is_python_module=False,
-        verbose=verbose,
-    )
+
+    def _load() -> None:
+        load(
+            name="llmq_amd_hip_ops",
+            sources=[str(SRC)],
+            extra_cflags=["-O3", "-std=c++20"],
+            extra_cuda_cflags=["-O3", "-std=c++20"],
+            build_directory=str(BUILD_DIR),
+            is_python_module=False,
+            verbose=verbose,
+        )
+
+    try:
+        _load()
+    except Exception:
+        # A stale incremental object (e.g. after a template-signature change)
+        # can link a .so with dangling device stubs — rebuild clean once.
+        shutil.rmtree(BUILD_DIR, ignore_errors=True)
+        BUILD_DIR.mkdir(exist_ok=True)
+        _load()
     built = BUILD_DIR / "llmq_amd_hip_ops.so"
     if not built.is_file():
         candidates = list(BUILD_DIR.glob("*.so"))
