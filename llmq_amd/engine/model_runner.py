@@ -64,6 +64,10 @@ class ModelRunner:
         self._sample_keys: Optional[torch.Tensor] = None
         self._sample_temps: Optional[torch.Tensor] = None
         self._sample_temps_h: Optional[torch.Tensor] = None
+        self._sample_seeds: Optional[torch.Tensor] = None
+        self._sample_seeds_h: Optional[torch.Tensor] = None
+        self._sample_pos: Optional[torch.Tensor] = None
+        self._sample_pos_h: Optional[torch.Tensor] = None
         if self.use_graphs:
             self._alloc_static_buffers()
 
@@ -359,16 +363,45 @@ class ModelRunner:
                 self._sample_keys = torch.empty(cap, dtype=torch.int64, device=dev)
                 self._sample_temps = torch.empty(cap, dtype=torch.float32, device=dev)
                 self._sample_temps_h = torch.empty(cap, dtype=torch.float32, pin_memory=True)
+                self._sample_seeds = torch.empty(cap, dtype=torch.int32, device=dev)
+                self._sample_seeds_h = torch.empty(cap, dtype=torch.int32, pin_memory=True)
+                self._sample_pos = torch.empty(cap, dtype=torch.int32, device=dev)
+                self._sample_pos_h = torch.empty(cap, dtype=torch.int32, pin_memory=True)
             th = self._sample_temps_h
+            sh = self._sample_seeds_h
+            ph = self._sample_pos_h
             for i, s in enumerate(seqs):
                 th[i] = s.params.temperature
+                rs = s.params.seed
+                if rs is None:
+                    sh[i] = 0  # unseeded sentinel
+                else:
+                    v = rs & 0x7FFFFFFF
+                    sh[i] = v if v else 0x1E3779B9  # remap literal seed 0
+                ph[i] = s.output_len
             self._sample_temps[:B].copy_(th[:B], non_blocking=True)
+            self._sample_seeds[:B].copy_(sh[:B], non_blocking=True)
+            self._sample_pos[:B].copy_(ph[:B], non_blocking=True)
             ops.sample_gumbel_argmax(
                 self._sample_out[:B], self._sample_keys[:B], logits.float(),
-                self._sample_temps[:B], self.config.seed, self._sample_step,
+                self._sample_temps[:B], self._sample_seeds[:B],
+                self._sample_pos[:B], self.config.seed, self._sample_step,
             )
             return self._sample_out[:B]
         temps = torch.tensor([s.params.temperature for s in seqs], dtype=torch.float32, device=dev)
         tps = torch.tensor([s.params.top_p for s in seqs], dtype=torch.float32, device=dev)
         tks = torch.tensor([s.params.top_k for s in seqs], dtype=torch.int64, device=dev)
-        return ops.sample_tokens(logits, temps, tps, tks, self.sampling_generator)
+        out = ops.sample_tokens(logits, temps, tps, tks, self.sampling_generator)
+        # Per-request seeds (reproducible sampling): re-draw those rows with
+        # a generator keyed on (seed, output position) — batch-independent.
+        for i, s in enumerate(seqs):
+            if s.params.seed is not None and s.params.temperature > 0:
+                g = torch.Generator(device="cpu")
+                g.manual_seed((s.params.seed << 20) ^ s.output_len)
+                row = logits[i : i + 1].float().cpu()
+                tok = ops.sample_tokens(
+                    row, temps[i : i + 1].cpu(), tps[i : i + 1].cpu(),
+                    tks[i : i + 1].cpu(), g,
+                )
+                out[i] = tok.to(out.device)
+        return out

@@ -222,19 +222,23 @@ def varlen_prefill_attention(
 
 
 def sample_gumbel_argmax(
-    out: torch.Tensor,      # [B] int64
-    keys: torch.Tensor,     # [B] int64 scratch (packed value|~index)
-    logits: torch.Tensor,   # [B, V] fp32
-    temps: torch.Tensor,    # [B] fp32 (<= 0 → greedy row)
+    out: torch.Tensor,        # [B] int64
+    keys: torch.Tensor,       # [B] int64 scratch (packed value|~index)
+    logits: torch.Tensor,     # [B, V] fp32
+    temps: torch.Tensor,      # [B] fp32 (<= 0 → greedy row)
+    req_seeds: torch.Tensor,  # [B] int32 (0 = unseeded)
+    req_pos: torch.Tensor,    # [B] int32 output position (seeded rows)
     seed: int,
     step: int,
 ) -> None:
     """Fused one-pass sampler: per-row Gumbel-max (== softmax sampling) or
-    argmax for greedy rows. Counter-based RNG keyed on (seed, step) so
-    results are launch-geometry- and graph-independent."""
+    argmax for greedy rows. Unseeded rows draw from the counter-based
+    (engine seed, step, row) stream; rows with a request seed draw from
+    (request_seed, output_position) — reproducible for a given request
+    regardless of batch placement."""
     assert _use_hip(logits)
     keys.zero_()
-    _EXT.sample_gumbel_argmax(out, keys, logits, temps, seed, step)
+    _EXT.sample_gumbel_argmax(out, keys, logits, temps, req_seeds, req_pos, seed, step)
 
 
 def sample_tokens(

@@ -395,11 +395,14 @@ void varlen_prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,
 // ------------------------------------------------------------- sampling --
 
 void sample_gumbel_argmax(at::Tensor out, at::Tensor keys, at::Tensor logits,
-                          at::Tensor temps, long seed, long step) {
+                          at::Tensor temps, at::Tensor req_seeds,
+                          at::Tensor req_pos, long seed, long step) {
   CHECK_GPU(logits);
   TORCH_CHECK(logits.scalar_type() == at::kFloat && logits.dim() == 2);
   TORCH_CHECK(out.scalar_type() == at::kLong);
   TORCH_CHECK(keys.scalar_type() == at::kLong && keys.numel() == logits.size(0));
+  TORCH_CHECK(req_seeds.scalar_type() == at::kInt);
+  TORCH_CHECK(req_pos.scalar_type() == at::kInt);
   const int B = logits.size(0), V = logits.size(1);
   // enough splits to fill the chip at small B
   int nsplit = 1;
@@ -407,8 +410,10 @@ void sample_gumbel_argmax(at::Tensor out, at::Tensor keys, at::Tensor logits,
   hipLaunchKernelGGL(sample_argmax_kernel, dim3(B, nsplit), dim3(256), 0,
                      stream(),
                      reinterpret_cast<unsigned long long*>(keys.data_ptr()),
-                     logits.data_ptr<float>(), temps.data_ptr<float>(), V,
-                     (unsigned int)seed, (unsigned int)step);
+                     logits.data_ptr<float>(), temps.data_ptr<float>(),
+                     reinterpret_cast<const unsigned int*>(req_seeds.data_ptr()),
+                     reinterpret_cast<const unsigned int*>(req_pos.data_ptr()),
+                     V, (unsigned int)seed, (unsigned int)step);
   hipLaunchKernelGGL(unpack_keys_kernel, dim3((B + 255) / 256), dim3(256), 0,
                      stream(), out.data_ptr<long>(),
                      reinterpret_cast<const unsigned long long*>(keys.data_ptr()),
@@ -426,7 +431,7 @@ TORCH_LIBRARY(llmq_amd, m) {
   m.def("reshape_and_cache(Tensor key, Tensor value, Tensor(a!) k_cache, Tensor(b!) v_cache, Tensor slot_mapping) -> ()");
   m.def("paged_decode_attention(Tensor(a!) out, Tensor q, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor context_lens, float scale, float softcap, int window) -> ()");
   m.def("varlen_prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, Tensor cu_seqlens_k, int max_seqlen, float scale, float softcap, int window) -> ()");
-  m.def("sample_gumbel_argmax(Tensor(a!) out, Tensor(b!) keys, Tensor logits, Tensor temps, int seed, int step) -> ()");
+  m.def("sample_gumbel_argmax(Tensor(a!) out, Tensor(b!) keys, Tensor logits, Tensor temps, Tensor req_seeds, Tensor req_pos, int seed, int step) -> ()");
   m.def("norm_add_norm(Tensor(a!) x, Tensor(b!) residual, Tensor w_post, Tensor w_pre, float eps, float offset) -> ()");
   m.def("rope_and_cache(Tensor(a!) q, Tensor(b!) k, Tensor value, Tensor(c!) k_cache, Tensor(d!) v_cache, Tensor positions, Tensor cos_sin, Tensor slot_mapping) -> ()");
 }

@@ -43,10 +43,16 @@ DEVINL unsigned long long pack_key(float val, int idx) {
   return ((unsigned long long)u << 32) | (unsigned int)(~idx);
 }
 
+// Per-row RNG keying: rows with a user seed (req_seeds[b] != 0) draw from
+// (request_seed, output_position) — REPRODUCIBLE for a given request
+// regardless of batch placement or engine step. Unseeded rows draw from
+// (engine seed, step, row).
 __global__ __launch_bounds__(256) void sample_argmax_kernel(
     unsigned long long* __restrict__ out_keys,  // [B], pre-zeroed
     const float* __restrict__ logits,           // [B, V]
     const float* __restrict__ temps,            // [B]
+    const unsigned int* __restrict__ req_seeds, // [B] (0 = unseeded)
+    const unsigned int* __restrict__ req_pos,   // [B] output position
     int V, unsigned int seed, unsigned int step) {
   const int b = blockIdx.x;
   const int nsplit = gridDim.y;
@@ -57,13 +63,17 @@ __global__ __launch_bounds__(256) void sample_argmax_kernel(
   const bool greedy = temp <= 0.f;
   const float inv_t = greedy ? 1.0f : 1.0f / temp;
   const float* row = logits + (long)b * V;
+  const unsigned int rs = req_seeds[b];
+  const unsigned int kseed = rs ? rs : seed;
+  const unsigned int kstep = rs ? req_pos[b] : step;
+  const unsigned int kb = rs ? 0u : (unsigned int)b;
 
   float best = -3e38f;
   int best_i = v0;
   for (int v = v0 + threadIdx.x; v < v1; v += 256) {
     float x = row[v] * inv_t;
     if (!greedy) {
-      const float u = uniform01(seed, step, b, v);
+      const float u = uniform01(kseed, kstep, kb, v);
       x += -__logf(-__logf(u));  // Gumbel(0,1)
     }
     if (x > best) { best = x; best_i = v; }

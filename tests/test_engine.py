@@ -257,3 +257,25 @@ class TestChunkedPrefill:
         assert saw_midchunk_mixed
         assert res["long"] == expect_long
         assert "short" in res
+
+
+def test_per_request_seed_reproducible():
+    """The same prompt + the same request seed must sample the same output
+    across separate engines and different batch compositions (CPU path)."""
+    params = SamplingParams(temperature=0.9, max_tokens=8, seed=1234, ignore_eos=True)
+
+    e1 = make_engine()
+    a = e1.generate_batch(["seeded prompt"], params)[0]
+
+    e2 = make_engine()
+    # different batchmates → different rows/steps, same seeded stream
+    e2.add_request("other", prompt="some other work",
+                   params=SamplingParams(temperature=0.7, max_tokens=12, ignore_eos=True))
+    e2.step()
+    e2.add_request("s", prompt="seeded prompt", params=params)
+    res = {}
+    while e2.has_unfinished():
+        for out in e2.step():
+            if out.finished:
+                res[out.request_id] = out.text
+    assert res["s"] == a

@@ -265,11 +265,16 @@ class TestEngineGPU:
 
 
 class TestFusedSampler:
-    def _run(self, logits, temps, seed=7, step=3):
+    def _run(self, logits, temps, seed=7, step=3, req_seeds=None, req_pos=None):
         B = logits.shape[0]
         out = torch.empty(B, dtype=torch.int64, device=DEV)
         keys = torch.empty(B, dtype=torch.int64, device=DEV)
-        ops.sample_gumbel_argmax(out, keys, logits, temps, seed, step)
+        if req_seeds is None:
+            req_seeds = torch.zeros(B, dtype=torch.int32, device=DEV)
+        if req_pos is None:
+            req_pos = torch.zeros(B, dtype=torch.int32, device=DEV)
+        ops.sample_gumbel_argmax(out, keys, logits, temps, req_seeds, req_pos,
+                                 seed, step)
         return out
 
     def test_greedy_matches_argmax(self):
@@ -288,6 +293,23 @@ class TestFusedSampler:
         c = self._run(logits, temps, seed=1, step=6)
         assert torch.equal(a, b)
         assert not torch.equal(a, c)  # different step → different draws
+
+    def test_request_seed_reproducible_and_position_keyed(self):
+        """Rows with a request seed draw from (seed, position): identical
+        across engine steps and batch rows; rows differ by position."""
+        torch.manual_seed(0)
+        logits = torch.randn(4, 4096, device=DEV)
+        same_logits = logits[0:1].expand(4, -1).contiguous()
+        temps = torch.full((4,), 0.8, device=DEV)
+        seeds = torch.full((4,), 99, dtype=torch.int32, device=DEV)
+        pos0 = torch.zeros(4, dtype=torch.int32, device=DEV)
+        a = self._run(same_logits, temps, seed=1, step=5, req_seeds=seeds, req_pos=pos0)
+        b = self._run(same_logits, temps, seed=2, step=9, req_seeds=seeds, req_pos=pos0)
+        assert torch.equal(a, b)  # engine step/seed irrelevant for seeded rows
+        assert (a == a[0]).all()  # batch row irrelevant
+        pos1 = torch.ones(4, dtype=torch.int32, device=DEV)
+        c = self._run(same_logits, temps, seed=1, step=5, req_seeds=seeds, req_pos=pos1)
+        assert not torch.equal(a, c)  # position advances the stream
 
     def test_distribution_matches_softmax(self):
         """Gumbel-max sampling must match the softmax distribution."""
