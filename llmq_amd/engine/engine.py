@@ -31,6 +31,11 @@ logger = logging.getLogger(__name__)
 # Memory the engine leaves free on top of weights+KV (workspace, graphs).
 _RESERVE_BYTES = 4 << 30
 
+# Mixed steps fuse decode+prefill into one forward (shared GEMMs); below
+# this many prefill tokens the fusion is not worth losing the decode
+# hipGraph, so the two segments run as separate forwards.
+_MIXED_FUSE_MIN_TOKENS = 128
+
 _TUNED_GEMMS_DONE = False
 
 
@@ -222,7 +227,18 @@ class LLMEngine:
         if batch.kind == "prefill":
             tokens = self.runner.execute_prefill(batch.seqs, batch.chunks or None)
         elif batch.kind == "mixed":
-            tokens = self.runner.execute_mixed(batch)
+            ptoks = sum(e - st for st, e in batch.chunks)
+            if ptoks < _MIXED_FUSE_MIN_TOKENS and self.runner.use_graphs:
+                # Tiny admission riding a big decode batch: the shared-GEMM
+                # win is small but fusing forces the whole decode EAGER —
+                # replay the decode hipGraph and prefill separately instead.
+                d_tokens = self.runner.execute_decode(batch.seqs[: batch.n_decode])
+                p_tokens = self.runner.execute_prefill(
+                    batch.seqs[batch.n_decode :], batch.chunks
+                )
+                tokens = torch.cat([d_tokens, p_tokens])
+            else:
+                tokens = self.runner.execute_mixed(batch)
         else:
             tokens = self.runner.execute_decode(batch.seqs)
         token_list = tokens.tolist()

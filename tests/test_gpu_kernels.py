@@ -556,3 +556,40 @@ class TestPreemptionGPU:
         for o in outs.values():
             assert o.output_tokens == 64
         assert eng.allocator.num_free == eng.allocator.num_blocks
+
+
+class TestMixedStepGPU:
+    def test_tiny_admission_uses_graph_split_path(self):
+        """A 1-seq admission riding a running decode batch takes the
+        graph-replay + separate-prefill path; outputs must be complete and
+        greedy-consistent with a fused-only engine."""
+        from llmq_amd.engine.config import EngineConfig
+        from llmq_amd.engine.engine import LLMEngine
+        from llmq_amd.engine.sampling_params import SamplingParams
+
+        def run(enforce_eager):
+            eng = LLMEngine(EngineConfig(
+                model="tiny-llama-d128", max_num_seqs=8, max_model_len=256,
+                load_weights=False, num_kv_blocks=512,
+                enforce_eager=enforce_eager,
+            ))
+            greedy = SamplingParams(temperature=0.0, max_tokens=24, ignore_eos=True)
+            for i in range(4):
+                eng.add_request(f"a{i}", prompt=f"base {i}", params=greedy)
+            eng.step()  # prefill the base batch
+            for _ in range(3):
+                eng.step()  # decoding
+            eng.add_request("late", prompt="xy", params=greedy)  # tiny admission
+            outs = {}
+            while eng.has_unfinished():
+                for out in eng.step():
+                    if out.finished:
+                        outs[out.request_id] = out.text
+            del eng
+            torch.cuda.empty_cache()
+            return outs
+
+        graphs = run(False)
+        eager = run(True)
+        assert set(graphs) == set(eager) == {"a0", "a1", "a2", "a3", "late"}
+        assert graphs == eager
