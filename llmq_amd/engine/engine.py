@@ -232,7 +232,11 @@ class LLMEngine:
                 # Tiny admission riding a big decode batch: the shared-GEMM
                 # win is small but fusing forces the whole decode EAGER —
                 # replay the decode hipGraph and prefill separately instead.
-                d_tokens = self.runner.execute_decode(batch.seqs[: batch.n_decode])
+                # .clone(): the fused sampler returns a VIEW of a persistent
+                # output buffer that the second _sample call would overwrite
+                d_tokens = self.runner.execute_decode(
+                    batch.seqs[: batch.n_decode]
+                ).clone()
                 p_tokens = self.runner.execute_prefill(
                     batch.seqs[batch.n_decode :], batch.chunks
                 )
