@@ -2,13 +2,17 @@
 
 The in-tree equivalent of the capability the reference buys from vLLM
 (admit up to max_num_seqs sequences per step, prefill/decode
-interleave, paged-KV admission control). Policy (vLLM-v0-style):
+interleave, paged-KV admission control). Policy:
 
-- FCFS waiting queue; a step is either one PREFILL batch (packed varlen,
-  bounded by max_prefill_tokens) or one DECODE batch (all running seqs).
-- Prefill has priority: new sequences join as soon as slots + KV blocks
-  allow, keeping the batch full (the broker-side prefetch ≫ batch trick,
-  SURVEY §3.1, keeps this queue fed).
+- One step = the DECODE batch (every running sequence) plus a PREFILL
+  segment built within the max_prefill_tokens budget ("mixed" step, one
+  forward with per-segment attention) — decode never stalls behind new
+  prompts. The broker-side prefetch ≫ batch trick keeps the
+  waiting queue fed.
+- The prefill segment first CONTINUES chunked long prompts, then admits
+  new sequences FCFS; a prompt larger than the budget prefills chunk by
+  chunk against the paged cache (mid-chunk sequences live in
+  ``prefilling`` and sample no tokens until their context completes).
 - Preemption by recompute: if a decode step cannot allocate the next KV
   block, the youngest running sequence is evicted, its blocks freed, and it
   re-enters the waiting queue with its generated tokens as prompt.
