@@ -252,3 +252,36 @@ stages:
     res = runner.invoke(cli, ["status", "-p", str(cfg)])
     assert res.exit_code == 0, res.output
     assert "first" in res.output and "second" in res.output
+
+
+def test_receive_pipeline_final_results(broker_env, tmp_path):
+    """`llmq receive -p pipeline.yaml` drains the pipeline's final results
+    queue (reference receive.py:147-283)."""
+    cfg = tmp_path / "p.yaml"
+    cfg.write_text("""
+name: recv
+stages:
+  - name: only
+    worker: dummy
+""")
+
+    async def seed():
+        from llmq_amd.core.pipeline import PipelineConfig
+        c = _client(broker_env)
+        await c.connect()
+        pipeline = PipelineConfig.from_yaml_file(str(cfg))
+        await c.setup_pipeline_infrastructure(pipeline)
+        await c.publish_to_queue(
+            pipeline.get_pipeline_results_queue_name(),
+            Result(id="pr1", prompt="p", result="done", worker_id="w",
+                   duration_ms=1.0).model_dump_json(),
+            "pr1",
+        )
+        await c.disconnect()
+
+    asyncio.new_event_loop().run_until_complete(seed())
+    runner = CliRunner()
+    res = runner.invoke(cli, ["receive", "-p", str(cfg), "--timeout", "2"])
+    assert res.exit_code == 0, res.output
+    lines = [json.loads(l) for l in res.output.splitlines() if l.startswith("{")]
+    assert lines and lines[0]["id"] == "pr1"
