@@ -85,7 +85,14 @@ def load_tokenizer(model: str, vocab_size: int, bos: int, eos: int):
         (path / "tokenizer.json").is_file() or (path / "tokenizer.model").is_file()
     ):
         try:
-            return HFTokenizer(path)
+            tok = HFTokenizer(path)
+            # some checkpoints ship no eos/bos in tokenizer_config — fall
+            # back to the model spec's ids so finish detection still works
+            if tok.eos_token_id is None:
+                tok.eos_token_id = eos
+            if tok.bos_token_id is None:
+                tok.bos_token_id = bos
+            return tok
         except Exception as exc:  # pragma: no cover
             logger.warning("HF tokenizer load failed (%s); using byte tokenizer", exc)
     return ByteTokenizer(vocab_size, bos, eos)
