@@ -190,22 +190,27 @@ def worker() -> None:
 @click.option("--max-num-seqs", type=int, default=None)
 @click.option("--max-model-len", type=int, default=None)
 @click.option("--prefetch", type=int, default=None)
+@click.option("--kv-cache-dtype", type=click.Choice(["auto", "fp8"]), default="auto",
+              help="KV storage dtype; fp8 (OCP e4m3) doubles KV capacity")
 @click.option("--engine-overrides", default=None,
               help='JSON dict of EngineConfig overrides, e.g. \'{"enforce_eager": true}\'')
 def worker_run(model, queue_name, tensor_parallel_size, max_num_seqs, max_model_len,
-               prefetch, engine_overrides):
+               prefetch, kv_cache_dtype, engine_overrides):
     """GPU inference worker (in-tree MI355X engine)."""
     import json as _json
 
     from llmq_amd.cli.worker import run_engine_worker
 
+    overrides = _json.loads(engine_overrides) if engine_overrides else {}
+    if kv_cache_dtype != "auto":
+        overrides.setdefault("kv_cache_dtype", kv_cache_dtype)
     run_engine_worker(
         model, queue_name,
         tensor_parallel_size=tensor_parallel_size,
         max_num_seqs=max_num_seqs,
         max_model_len=max_model_len,
         prefetch=prefetch,
-        engine_overrides=_json.loads(engine_overrides) if engine_overrides else None,
+        engine_overrides=overrides or None,
     )
 
 
