@@ -289,3 +289,66 @@ def test_resolve_tp_precedence(monkeypatch):
     w.stage_config = {}
     # no GPU in this container → 1
     assert w._resolve_tp() == 1
+
+
+def test_bridge_high_concurrency_fake_engine():
+    """500 concurrent requests through the bridge against a fake engine
+    ('add a FakeEngine echoing tokens') — ordering, accounting
+    and completion under prefetch ≫ batch churn."""
+
+    class FakeOut:
+        __slots__ = ("request_id", "finished", "text", "prompt_tokens",
+                     "output_tokens", "finish_reason", "queue_wait_ms",
+                     "prefill_ms", "decode_ms", "new_token_ids")
+
+    class FakeEngine:
+        def __init__(self):
+            self.pending = {}
+            self.tokenizer = None
+            self.steps = 0
+
+        def add_request(self, rid, prompt=None, params=None):
+            if rid in self.pending:
+                raise ValueError("dup")
+            self.pending[rid] = (prompt, 2 + (hash(rid) % 3))  # 2-4 steps
+
+        def has_unfinished(self):
+            return bool(self.pending)
+
+        def step(self):
+            self.steps += 1
+            outs = []
+            for rid in list(self.pending)[:64]:  # batch cap of 64
+                prompt, left = self.pending[rid]
+                left -= 1
+                if left <= 0:
+                    del self.pending[rid]
+                    o = FakeOut()
+                    o.request_id = rid
+                    o.finished = True
+                    o.text = f"echo {prompt}"
+                    o.prompt_tokens = len(prompt)
+                    o.output_tokens = 3
+                    o.finish_reason = "eos"
+                    o.queue_wait_ms = o.prefill_ms = o.decode_ms = 0.0
+                    outs.append(o)
+                else:
+                    self.pending[rid] = (prompt, left)
+            return outs
+
+    async def main():
+        bridge = AsyncEngineBridge(lambda: FakeEngine())
+        await bridge.start()
+        from llmq_amd.engine.sampling_params import SamplingParams
+
+        params = SamplingParams(max_tokens=4)
+        results = await asyncio.gather(*[
+            bridge.generate(f"r{i}", f"p{i}", params) for i in range(500)
+        ])
+        assert len(results) == 500
+        for i, r in enumerate(results):
+            assert r.text == f"echo p{i}"  # each future got ITS OWN result
+        assert bridge.num_in_flight == 0
+        bridge.shutdown()
+
+    run_async(main())
