@@ -168,6 +168,7 @@ class BrokerServer:
         self.default_ttl_ms = default_ttl_ms
         self.queues: Dict[str, Queue] = {}
         self.workers: Dict[str, Dict[str, Any]] = {}  # worker_id -> health blob
+        self._send_tasks: set = set()  # strong refs: bare ensure_future may be GC'd
         self._server: Optional[asyncio.AbstractServer] = None
         self._dirty_journals: set[str] = set()
         self._flusher_task: Optional[asyncio.Task] = None
@@ -314,7 +315,7 @@ class BrokerServer:
             q.unacked[(target.conn.id, tag)] = msg
             target.inflight += 1
             target.conn.tags[tag] = (q, target)
-            asyncio.ensure_future(
+            t = asyncio.ensure_future(
                 target.conn.send(
                     {
                         "push": "deliver",
@@ -326,6 +327,8 @@ class BrokerServer:
                     }
                 )
             )
+            self._send_tasks.add(t)
+            t.add_done_callback(self._send_tasks.discard)
 
     def _ack_internal(self, q: Queue, msg: Message) -> None:
         if q.journal:
