@@ -71,6 +71,7 @@ class BrokerClient:
         self._queue_callbacks: Dict[str, DeliverCallback] = {}
         self._consumer_queue: Dict[int, str] = {}  # consumer_id -> queue
         self._send_lock = asyncio.Lock()
+        self._cb_tasks: set = set()  # strong refs: bare ensure_future may be GC'd
         self._closed = True
         self.default_prefetch = self.config.queue_prefetch
 
@@ -156,10 +157,12 @@ class BrokerClient:
         delivery = Delivery(self, frame)
         cb = self._queue_callbacks.get(delivery.queue)
         if cb is not None:
-            asyncio.ensure_future(self._run_cb(cb, delivery))
+            t = asyncio.ensure_future(self._run_cb(cb, delivery))
         else:
             # No consumer (e.g. cancelled): requeue.
-            asyncio.ensure_future(delivery.nack(requeue=True))
+            t = asyncio.ensure_future(delivery.nack(requeue=True))
+        self._cb_tasks.add(t)
+        t.add_done_callback(self._cb_tasks.discard)
 
     @staticmethod
     async def _run_cb(cb: DeliverCallback, delivery: Delivery) -> None:
