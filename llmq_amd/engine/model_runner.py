@@ -95,8 +95,14 @@ class ModelRunner:
             B, self.max_blocks_per_seq, dtype=torch.int32, pin_memory=True
         )
         self.h_context_lens = torch.ones(B, dtype=torch.int32, pin_memory=True)
-        # (request_id, staged_len) per row — incremental block-table staging
-        self._staged_rows: List[Tuple[str, int]] = [("", 0)] * B
+        # (seq uid, preemption count, staged_len) per row — incremental
+        # block-table staging. Keyed on Sequence.uid (monotonic, never
+        # reused), NOT request_id: callers may reuse request ids across
+        # sequence lifetimes, and a reused id at the same row with an equal
+        # table length but different blocks would leave the hipGraph reading
+        # the old sequence's block ids. num_preemptions guards the same-life
+        # case where preemption-by-recompute reassigns the whole table.
+        self._staged_rows: List[Tuple[int, int, int]] = [(-1, -1, 0)] * B
 
     def capture_graphs(self) -> None:
         """Capture the decode forward for each batch-size bucket."""
@@ -322,9 +328,9 @@ class ModelRunner:
             n = len(s.block_table)
             if n > max_w:
                 max_w = n
-            if self._staged_rows[i] != (s.request_id, n):
+            if self._staged_rows[i] != (s.uid, s.num_preemptions, n):
                 bt_np[i, :n] = s.block_table
-                self._staged_rows[i] = (s.request_id, n)
+                self._staged_rows[i] = (s.uid, s.num_preemptions, n)
         if B < bucket:
             # pad rows: context_len 1 pointing at block 0 (defined garbage,
             # their logits are never read)
@@ -332,9 +338,9 @@ class ModelRunner:
             self.h_ids[B:bucket] = 0
             self.h_pos[B:bucket] = 0
             for i in range(B, bucket):
-                if self._staged_rows[i] != ("", 0):
+                if self._staged_rows[i] != (-1, -1, 0):
                     bt_np[i, :1] = 0
-                    self._staged_rows[i] = ("", 0)
+                    self._staged_rows[i] = (-1, -1, 0)
         self.in_ids[:bucket].copy_(self.h_ids[:bucket], non_blocking=True)
         self.in_pos[:bucket].copy_(self.h_pos[:bucket], non_blocking=True)
         self.in_slots[:bucket].copy_(self.h_slots[:bucket], non_blocking=True)
@@ -394,14 +400,24 @@ class ModelRunner:
         out = ops.sample_tokens(logits, temps, tps, tks, self.sampling_generator)
         # Per-request seeds (reproducible sampling): re-draw those rows with
         # a generator keyed on (seed, output position) — batch-independent.
-        for i, s in enumerate(seqs):
-            if s.params.seed is not None and s.params.temperature > 0:
+        # Gathered in ONE device→host transfer (a per-row .cpu() would sync
+        # the stream once per seeded row per step).
+        seeded = [
+            i for i, s in enumerate(seqs)
+            if s.params.seed is not None and s.params.temperature > 0
+        ]
+        if seeded:
+            idx = torch.tensor(seeded, dtype=torch.long, device=dev)
+            rows = logits[idx].float().cpu()
+            temps_c, tps_c, tks_c = temps[idx].cpu(), tps[idx].cpu(), tks[idx].cpu()
+            redraw = torch.empty(len(seeded), dtype=out.dtype)
+            for j, i in enumerate(seeded):
+                s = seqs[i]
                 g = torch.Generator(device="cpu")
                 g.manual_seed((s.params.seed << 20) ^ s.output_len)
-                row = logits[i : i + 1].float().cpu()
-                tok = ops.sample_tokens(
-                    row, temps[i : i + 1].cpu(), tps[i : i + 1].cpu(),
-                    tks[i : i + 1].cpu(), g,
-                )
-                out[i] = tok.to(out.device)
+                redraw[j] = ops.sample_tokens(
+                    rows[j : j + 1], temps_c[j : j + 1], tps_c[j : j + 1],
+                    tks_c[j : j + 1], g,
+                )[0]
+            out[idx] = redraw.to(out.device)
         return out

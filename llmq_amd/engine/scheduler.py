@@ -20,6 +20,7 @@ interleave, paged-KV admission control). Policy:
 
 from __future__ import annotations
 
+import itertools
 import logging
 from collections import deque
 from dataclasses import dataclass, field
@@ -38,15 +39,24 @@ class SeqStatus(Enum):
     FINISHED = "finished"
 
 
+_SEQ_UID = itertools.count()
+
+
 class Sequence:
     __slots__ = (
         "request_id", "token_ids", "prompt_len", "params", "status",
         "block_table", "arrival_time", "first_token_time", "finish_reason",
         "output_text", "num_preemptions", "prefilled", "prefill_start_time",
+        "uid",
     )
 
     def __init__(self, request_id: str, prompt_token_ids: List[int], params: SamplingParams,
                  arrival_time: float = 0.0):
+        # Monotonic per-process id: request_ids may be REUSED across sequence
+        # lifetimes (generate_batch reuses "batch-{i}"), so anything caching
+        # per-sequence state (e.g. the hipGraph block-table staging rows)
+        # must key on uid, not request_id.
+        self.uid = next(_SEQ_UID)
         self.request_id = request_id
         self.token_ids: List[int] = list(prompt_token_ids)
         self.prompt_len = len(prompt_token_ids)
@@ -184,10 +194,16 @@ class Scheduler:
             prefills.append(seq)
             chunks.append((start, end))
             budget -= end - start
-        # (b) admit new sequences
+        # (b) admit new sequences. Seats are held not just by this step's
+        # prefill segment but also by chunked sequences STRANDED this step
+        # (budget/allocation ran out above) — they still occupy KV blocks
+        # and will rejoin a later segment, so count them or running+prefilling
+        # can exceed max_num_seqs (oversized decode batches then silently
+        # fall off the hipGraph path).
+        stranded = sum(1 for s in self.prefilling if s not in prefills)
         while (
             self.waiting and budget > 0
-            and seats_used + len(prefills) < self.max_num_seqs
+            and seats_used + len(prefills) + stranded < self.max_num_seqs
         ):
             seq = self.waiting[0]
             n = seq.num_tokens

@@ -291,6 +291,11 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
       return;
     }
   }
+  // The VALU fallback reads the cache as T; an fp8 cache reaching it would
+  // be silently reinterpreted as bf16 (garbage, no error).
+  TORCH_CHECK(kc.scalar_type() != at::kFloat8_e4m3fn,
+              "fp8 KV cache requires the MFMA decode path (bf16 q, head_dim "
+              "128/256, GQA group <= 16); got head_dim=", D, " group=", G);
   TORCH_CHECK(G <= 8, "GQA group must be <= 8 for the VALU decode kernel");
   const int subs = 256 / (32 * G);
   const int lds = (G * D + DECODE_CHUNK * G + subs * G * (D + 2)) * sizeof(float) +

@@ -40,6 +40,13 @@ def check_invariants(eng: LLMEngine) -> None:
     assert alloc.num_free + len(held) == alloc.num_blocks, (
         f"leak: free={alloc.num_free} held={len(held)} total={alloc.num_blocks}"
     )
+    # seat cap: running + mid-prefill sequences can never exceed
+    # max_num_seqs (admission must count STRANDED prefilling seqs too —
+    # otherwise oversized decode batches fall off the hipGraph path)
+    assert len(sched.running) + len(sched.prefilling) <= sched.max_num_seqs, (
+        f"seats over cap: running={len(sched.running)} "
+        f"prefilling={len(sched.prefilling)} cap={sched.max_num_seqs}"
+    )
     # chunk bookkeeping
     for seq in sched.prefilling:
         assert 0 < seq.prefilled < seq.num_tokens
@@ -50,6 +57,21 @@ def check_invariants(eng: LLMEngine) -> None:
     bs = sched.block_size
     for seq in list(sched.running) + list(sched.prefilling):
         assert len(seq.block_table) * bs >= seq.prefilled
+
+
+def test_sequence_uid_unique_across_request_id_reuse():
+    """request_ids may be reused across sequence lifetimes; the uid used to
+    key per-sequence caches (hipGraph block-table staging rows) must not be.
+    """
+    from llmq_amd.engine.scheduler import Sequence
+
+    params = SamplingParams(temperature=0.0, max_tokens=4)
+    a = Sequence("batch-0", [1, 2, 3], params)
+    b = Sequence("batch-0", [4, 5, 6], params)
+    assert a.request_id == b.request_id
+    assert a.uid != b.uid
+    # the staging key distinguishes same-id same-length different-life rows
+    assert (a.uid, a.num_preemptions, 3) != (b.uid, b.num_preemptions, 3)
 
 
 @settings(max_examples=20, deadline=None)
