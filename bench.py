@@ -42,6 +42,11 @@ def main() -> None:
                         help="KV cache storage dtype (fp8 = optional mode, NOT the headline)")
     parser.add_argument("--profile-steps", type=int, default=0,
                         help="extra untimed steps after the timed region (rocprof)")
+    parser.add_argument("--tp", type=int, default=0,
+                        help="tensor-parallel degree: all ranks serve ONE "
+                        "model sharded over RCCL/xGMI (BASELINE config #4, "
+                        "e.g. --tp 8 --model tower-plus-72b under torchrun "
+                        "--nproc-per-node 8). Default 0 = data-parallel.")
     args = parser.parse_args()
 
     import torch
@@ -52,6 +57,12 @@ def main() -> None:
     distributed = world > 1
 
     use_gpu = torch.cuda.is_available()
+    tp_mode = args.tp > 1
+    if tp_mode and world != args.tp:
+        raise SystemExit(
+            f"--tp {args.tp} needs exactly that many ranks (WORLD_SIZE={world}); "
+            "launch via torchrun --nproc-per-node N bench.py --tp N"
+        )
     if distributed:
         import torch.distributed as dist
 
@@ -59,6 +70,10 @@ def main() -> None:
             torch.cuda.set_device(local_rank)
         # "nccl" IS RCCL on ROCm; gloo covers the CPU test path.
         dist.init_process_group(backend="nccl" if use_gpu else "gloo")
+        if tp_mode:
+            from llmq_amd.parallel import TPGroup, set_tp_group
+
+            set_tp_group(TPGroup(rank, world, None))
 
     device = f"cuda:{local_rank}" if use_gpu else "cpu"
 
@@ -84,24 +99,30 @@ def main() -> None:
         enforce_eager=args.eager,
         kv_cache_dtype=args.kv_dtype,
         hipgraph_max_batch=batch,
-        seed=1234 + rank,
+        # TP replicas must make IDENTICAL scheduler/sampler decisions —
+        # same seed everywhere; DP ranks decorrelate instead.
+        seed=1234 if tp_mode else 1234 + rank,
     )
     t_init = time.perf_counter()
-    engine = LLMEngine(cfg)
+    if tp_mode:
+        engine = LLMEngine(cfg, tp_rank=rank, tp_size=world)
+    else:
+        engine = LLMEngine(cfg)
     if rank == 0:
         print(f"[bench] engine init {time.perf_counter() - t_init:.1f}s", file=sys.stderr)
 
     # --- load the resident batch (synthetic prompts, random token ids)
     import numpy as np
 
-    rng = np.random.default_rng(42 + rank)
+    rng = np.random.default_rng(42 if tp_mode else 42 + rank)
     vocab = engine.spec.vocab_size
     params = SamplingParams(
         temperature=args.temperature, max_tokens=args.max_model_len, ignore_eos=True
     )
+    req_tag = "tp" if tp_mode else str(rank)
     for i in range(batch):
         ids = rng.integers(0, vocab, size=prompt_len).tolist()
-        engine.add_request(f"bench-{rank}-{i}", prompt_token_ids=ids, params=params)
+        engine.add_request(f"bench-{req_tag}-{i}", prompt_token_ids=ids, params=params)
 
     t_prefill = time.perf_counter()
     while engine.scheduler.num_waiting > 0 or engine.scheduler.prefilling:
@@ -151,7 +172,9 @@ def main() -> None:
         f"batch decayed during timing: {engine.scheduler.num_running} != {batch}"
     )
 
-    total_tokens = world * batch * args.steps
+    # TP shards ONE model over all ranks (whole-job tokens = one engine's);
+    # DP runs an independent engine per rank.
+    total_tokens = (1 if tp_mode else world) * batch * args.steps
     value = total_tokens / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
@@ -183,9 +206,9 @@ def main() -> None:
             "data": "synthetic",
             "config": {
                 "model": f"{model} ({engine.spec.name} arch, random init)",
-                "global_batch": world * batch,
+                "global_batch": batch if tp_mode else world * batch,
                 "seq_len": prompt_len,
-                "parallelism": f"dp{world}",
+                "parallelism": f"tp{world}" if tp_mode else f"dp{world}",
                 "kv_cache_dtype": args.kv_dtype,
             },
         }

@@ -109,6 +109,20 @@ class LLMEngine:
         if self.kv_dtype == torch.float8_e4m3fn and self.spec.head_dim not in (128, 256):
             raise ValueError("fp8 KV cache requires head_dim 128 or 256 (MFMA decode path)")
         num_blocks = self._size_kv_cache()
+        if tp > 1:
+            # Deterministic lockstep requires an IDENTICAL KV pool on every
+            # rank: mem_get_info varies per GPU, and different block counts
+            # mean divergent preemption/admission → silent token divergence
+            # between TP replicas. Take the min so every rank fits.
+            from llmq_amd.parallel import get_tp_group
+
+            synced = get_tp_group().min_scalar(num_blocks)
+            if synced != num_blocks:
+                logger.info(
+                    "KV blocks synced across TP ranks: local %d -> min %d",
+                    num_blocks, synced,
+                )
+            num_blocks = synced
         kv_heads_local = self.spec.num_kv_heads // tp
         self.kv_cache = KVCache(
             self.spec.num_layers, num_blocks, kv_heads_local,

@@ -42,6 +42,27 @@ class TPGroup:
         if self.world_size > 1:
             dist.barrier(group=self.group)
 
+    def min_scalar(self, v: int) -> int:
+        """All-reduce MIN of a host scalar across the TP group.
+
+        Used for deterministic KV-pool sizing: every rank must allocate the
+        SAME number of KV blocks or scheduler decisions (preemption,
+        admission) diverge between lockstep replicas and generated tokens
+        silently differ per rank. Rank-varying ``torch.cuda.mem_get_info``
+        is the hazard; min() keeps every rank within its own budget.
+        """
+        if self.world_size == 1:
+            return v
+        backend = dist.get_backend(self.group)
+        dev = (
+            torch.device("cuda", torch.cuda.current_device())
+            if backend == "nccl"
+            else torch.device("cpu")
+        )
+        t = torch.tensor([v], dtype=torch.int64, device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MIN, group=self.group)
+        return int(t.item())
+
 
 def init_tp(
     tp_size: int,

