@@ -844,6 +844,423 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
   }
 }
 
+// -------------------------------------------- pipelined MFMA paged decode --
+// Software-pipelined variant of paged_decode_mfma_kernel (round-2 item from
+// profiles/r1_step6_decode_tuning_story.md: the stage→S→stats→PV phase
+// serialization left ~30% of the KV stream rate on the table, and both
+// register-prefetch and occupancy-costing fixes measured WORSE — the fix
+// has to add zero VGPRs and keep the 2-workgroups/CU residency).
+//
+// Structure (per CDNA4 guide §5 "Pipelining across barriers" + §5.5 T3/T4):
+//   - TWO chunk buffers A/B; chunk i's K lives in buf[i&1], V overwrites the
+//     same buffer after S (the r1 shared-buffer trick), K(i+1) prefetches
+//     into buf[(i+1)&1].
+//   - ALL staging is `global_load_lds` (no VGPR round-trip, no staging
+//     instructions on the VALU): the LDS image is lane-linear per wave
+//     instruction, so K's XOR swizzle and V's tr16 subtile permutation move
+//     to the per-lane SOURCE address (guide §5.4 rule 21).
+//   - Raw `s_barrier` + counted `s_waitcnt vmcnt(N)`: V(i)+K(i+1) issue
+//     back-to-back right after the post-S barrier; the pre-PV barrier waits
+//     vmcnt(NI_K) (V landed, K still streaming); the loop-top barrier waits
+//     vmcnt(0) (K landed). A `__syncthreads()` would drain the glds queue
+//     at every barrier (the documented HIP-compiler ceiling).
+//   - The chunk containing L falls back to plain staged writes (glds cannot
+//     zero-fill; garbage V rows would reach the PV MFMA as NaN×0).
+// Per-workgroup timeline: the only window with no HBM traffic in flight is
+// the S phase (~hundreds of cycles vs ~6 µs of stream per chunk).
+
+// 16-B-per-lane global→LDS DMA, HIDDEN from hipcc in inline asm: with the
+// builtin form the compiler conservatively re-inserts `s_waitcnt vmcnt(0)`
+// before the first ds_read that may alias the DMA destination (verified in
+// the .s: it landed ahead of the PV tr16 reads, draining the K prefetch
+// every chunk). Hand-counted vmcnt in pipe_barrier_vm is the only wait
+// discipline for these, per guide §5.7 (its LDS-DMA recipe: M0 carries the
+// wave-uniform LDS byte address, saved/written/restored in ONE statement;
+// the s_nop is the SALU-write→M0-read wait state).
+DEVINL void glds16(const void* src, void* lds_dst_uniform) {
+  unsigned lds_off = (unsigned)(unsigned long)(
+      (__attribute__((address_space(3))) char*)lds_dst_uniform);
+  lds_off = __builtin_amdgcn_readfirstlane(lds_off);
+  unsigned keep;
+  asm volatile(
+      "s_mov_b32 %0, m0\n\t"
+      "s_mov_b32 m0, %2\n\t"
+      "s_nop 0\n\t"
+      "global_load_lds_dwordx4 %1, off\n\t"
+      "s_mov_b32 m0, %0"
+      : "=&s"(keep)
+      : "v"(src), "s"(lds_off)
+      : "memory");
+}
+
+DEVINL void pipe_barrier() {  // lgkm drain + raw barrier (glds may span)
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+}
+
+template <int N>
+DEVINL void pipe_barrier_vm() {  // + counted vmcnt: N glds may stay in flight
+  asm volatile("s_waitcnt vmcnt(%0) lgkmcnt(0)" :: "i"(N) : "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+}
+
+DEVINL void reg_fence(bf16x8_t& v) {  // force materialisation (wait) HERE
+  asm volatile("" : "+v"(v));
+}
+
+// min waves/SIMD: the 64-key chunk fits 2 workgroups/CU by LDS (4 waves/
+// SIMD) — force the register allocator to <=128 VGPRs so VGPRs don't cap
+// occupancy below that (the r1 lesson: this kernel is governed by
+// waves/SIMD, every latency trick that cost occupancy lost). The 128-key
+// chunk is 1 workgroup/CU by LDS (2 waves/SIMD) — let it use 196.
+template <int HEAD_DIM, int NW, int PD_KT>
+__global__ __launch_bounds__(NW * WAVE, (PD_KT <= 64 ? 4 : 2)) void paged_decode_pipe_kernel(
+    __hip_bfloat16* __restrict__ out,      // [B, H, D]
+    const __hip_bfloat16* __restrict__ q,  // [B, H, D]
+    const __hip_bfloat16* __restrict__ k_cache,  // [nb, KVH, bs, D]
+    const __hip_bfloat16* __restrict__ v_cache,
+    const int* __restrict__ block_tables,  // [B, max_blocks<=PD_MAX_BT]
+    const int* __restrict__ context_lens,  // [B]
+    float* __restrict__ scratch,           // [B,KVH,NSPLIT,G,D+2] (NSPLIT>1)
+    int num_heads, int num_kv_heads, int block_size, int max_blocks,
+    float scale, float softcap, int window, long q_stride, long out_stride) {
+  constexpr int D = HEAD_DIM;
+  constexpr int KS = D / 32;        // MFMA K-steps over the head dim
+  constexpr int NT = NW * WAVE;     // threads per workgroup
+  constexpr int SLABS = PD_KT / 16; // 16-key S slabs (<= NW; waves duplicate)
+  constexpr int D4 = D / NW;        // dim slab per wave (PV)
+  constexpr int DT = D4 / 16;       // 16-dim MFMA tiles per wave
+  constexpr int CPK = D / 8;        // 16B granules per key row
+  constexpr int NI_K = PD_KT * CPK / NT;          // K glds per wave per chunk
+  constexpr int NSUB = (PD_KT / 32) * (D / 16);   // V tr-subtiles per chunk
+  constexpr int NI_V = NSUB / NW;                 // V glds per wave per chunk
+  constexpr int KV_ELEMS = (PD_KT * D > NSUB * 528) ? PD_KT * D : NSUB * 528;
+  static_assert(PD_KT % 32 == 0 && SLABS <= NW && NSUB % NW == 0, "");
+  static_assert(PD_KT * CPK % NT == 0 && D % NW == 0 && D4 % 16 == 0, "");
+
+  const int b = blockIdx.x;
+  const int kh = blockIdx.y;
+  const int G = num_heads / num_kv_heads;  // <= 16
+  const int L = context_lens[b];
+  if (L <= 0) return;
+
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid & (WAVE - 1);
+  const int col = lane & 15;
+  const int kgrp = lane >> 4;
+  const int slab = wid % SLABS;
+
+  // ONE shared array (guide §5 trap 4a: a second __shared__ object makes
+  // hipcc drain vmcnt(0) before every ds_read while a glds is in flight).
+  constexpr int P_OFF = 2 * KV_ELEMS;                  // shorts
+  constexpr int MPART_OFF = P_OFF + 16 * PD_KT;        // floats from here
+  constexpr int F_BASE = (MPART_OFF * 2 + 15) / 16 * 4;  // float index base
+  constexpr int ALPHA_F = F_BASE + SLABS * 16;
+  constexpr int LPART_F = ALPHA_F + 16;
+  constexpr int BT_I = LPART_F + SLABS * 16;           // int index base
+  constexpr int TOTAL_BYTES = BT_I * 4 + PD_MAX_BT * 4;
+  __shared__ __attribute__((aligned(16))) char smem[TOTAL_BYTES];
+  short* const kv0 = reinterpret_cast<short*>(smem);
+  short* const kv1 = kv0 + KV_ELEMS;
+  short* const p_lds2 = kv0 + P_OFF;                   // [16][PD_KT] bf16
+  float* const fbase = reinterpret_cast<float*>(smem);
+  float* const mpart = fbase + F_BASE;                 // [SLABS][16]
+  float* const alpha_s = fbase + ALPHA_F;              // [16]
+  float* const l_part = fbase + LPART_F;               // [SLABS][16]
+  int* const bt_l = reinterpret_cast<int*>(smem) + BT_I;  // [PD_MAX_BT]
+
+  // ---- stage the block table once (plain loads; nb <= PD_MAX_BT is a
+  // dispatch precondition for this kernel)
+  const int* bt_glob = block_tables + (long)b * max_blocks;
+  {
+    const int nb = (L + block_size - 1) / block_size;
+    for (int i = tid; i < nb; i += NT) bt_l[i] = bt_glob[i];
+  }
+
+  // ---- Q fragments (padded GQA rows; row = lane&15)
+  bf16x8_t qfrag[KS];
+  {
+    const int qrow = (col < G) ? col : 0;
+    const __hip_bfloat16* qp = q + (long)b * q_stride + (long)(kh * G + qrow) * D;
+#pragma unroll
+    for (int ks = 0; ks < KS; ++ks) {
+      qfrag[ks] = *reinterpret_cast<const bf16x8_t*>(qp + ks * 32 + kgrp * 8);
+      reg_fence(qfrag[ks]);  // wait the load NOW — a first-use inside the
+      // loop would make hipcc place `s_waitcnt vmcnt(0)` there, draining
+      // the glds pipeline on every iteration (guide §5 trap 4b).
+    }
+  }
+
+  float m_regs[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float l_regs[4] = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t ot[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt) ot[dt] = {0.f, 0.f, 0.f, 0.f};
+
+  const int start = (window > 0 && L > window) ? (L - window) : 0;
+  const int base0 = (start / PD_KT) * PD_KT;
+  const int nsplit = gridDim.z;
+  int range_lo = base0, range_hi = L;
+  if (nsplit > 1) {
+    const int nchunks = (L - base0 + PD_KT - 1) / PD_KT;
+    const int per = (nchunks + nsplit - 1) / nsplit;
+    range_lo = base0 + (int)blockIdx.z * per * PD_KT;
+    range_hi = min(L, range_lo + per * PD_KT);
+  }
+
+  const long kvrow_stride = (long)num_kv_heads * block_size * D;
+
+  // Per-lane source row+offset for K granule g of a chunk (inverse swizzle
+  // on the source: LDS image is lane-linear, content matches swz()).
+  auto k_src = [&](int base, int g) -> const __hip_bfloat16* {
+    const int key = g / CPK;
+    const int r8 = (g % CPK) * 8;
+    const int d = r8 ^ ((key & 7) << 3);
+    const int gkey = base + key;
+    const long blk = bt_l[gkey / block_size];
+    return k_cache + (blk * num_kv_heads + kh) * ((long)block_size * D) +
+           (long)(gkey % block_size) * D + d;
+  };
+  // V: subtile st, in-subtile granule = lane (tr16 image inverse mapping).
+  auto v_src = [&](int base, int st) -> const __hip_bfloat16* {
+    const int ks = st / (D / 16), dtile = st % (D / 16);
+    const int pos = lane * 8;
+    const int bpos = pos / 64, rem = pos % 64;
+    const int qq = ((bpos >> 2) & 1) | ((bpos & 3) << 1);
+    const int key = ks * 32 + qq * 4 + rem / 16;
+    const int dim = dtile * 16 + (rem & 15);
+    const int gkey = base + key;
+    const long blk = bt_l[gkey / block_size];
+    return v_cache + (blk * num_kv_heads + kh) * ((long)block_size * D) +
+           (long)(gkey % block_size) * D + dim;
+  };
+
+  auto issue_k = [&](int base, short* dst) {
+#pragma unroll
+    for (int j = 0; j < NI_K; ++j) {
+      const int inst = wid * NI_K + j;
+      const int off = __builtin_amdgcn_readfirstlane(inst * 512);
+      glds16(k_src(base, inst * 64 + lane), dst + off);
+    }
+  };
+  auto issue_v = [&](int base, short* dst) {
+#pragma unroll
+    for (int j = 0; j < NI_V; ++j) {
+      const int st = wid * NI_V + j;
+      const int off = __builtin_amdgcn_readfirstlane(st * 528);
+      glds16(v_src(base, st), dst + off);
+    }
+  };
+  // Tail chunk (contains L): plain staged writes with zero fill.
+  auto stage_k_plain = [&](int base, short* dst) {
+    for (int c = tid; c < PD_KT * CPK; c += NT) {
+      const int key = c / CPK;
+      const int d8 = (c % CPK) * 8;
+      const int pdst = key * D + swz(key, d8);
+      const int gkey = base + key;
+      if (gkey < L) {
+        const long blk = bt_l[gkey / block_size];
+        const long rowoff =
+            (blk * num_kv_heads + kh) * ((long)block_size * D) +
+            (long)(gkey % block_size) * D + d8;
+        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) =
+            *reinterpret_cast<const bf16x8_t*>(k_cache + rowoff);
+      } else {
+        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+    }
+  };
+  auto stage_v_plain = [&](int base, short* dst) {
+    for (int c = tid; c < PD_KT * CPK; c += NT) {
+      const int key = c / CPK;
+      const int d8 = (c % CPK) * 8;
+      const int dtile = d8 / 16, col0 = d8 & 15;
+      const int qq = (key & 31) >> 2;
+      const int bpos = ((qq & 1) << 2) + (qq >> 1);
+      const int pdst = ((key >> 5) * (D / 16) + dtile) * 528 + bpos * 64 +
+                       (key & 3) * 16 + col0;
+      const int gkey = base + key;
+      if (gkey < L) {
+        const long blk = bt_l[gkey / block_size];
+        const long rowoff =
+            (blk * num_kv_heads + kh) * ((long)block_size * D) +
+            (long)(gkey % block_size) * D + d8;
+        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) =
+            *reinterpret_cast<const bf16x8_t*>(v_cache + rowoff);
+      } else {
+        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+    }
+  };
+
+  auto is_tail = [&](int base) { return base + PD_KT > L; };
+
+  // ---- prologue: block table visible, then K(first) in flight
+  pipe_barrier();  // bt_l ready (plain ds_writes; lgkm drain)
+  const bool have_range = range_lo < range_hi;
+  if (have_range && !is_tail(range_lo)) issue_k(range_lo, kv0);
+
+  int cur = 0;
+  for (int base = range_lo; base < range_hi; base += PD_KT, cur ^= 1) {
+    short* const X = cur ? kv1 : kv0;
+    short* const Y = cur ? kv0 : kv1;
+    const bool tail = is_tail(base);
+    if (tail) {
+      stage_k_plain(base, X);
+      pipe_barrier();
+    } else {
+      pipe_barrier_vm<0>();  // K(base) landed (all waves)
+    }
+
+    // ---- S[16,16] for this wave's slab (waves wid>=SLABS duplicate)
+    f32x4_t s = {0.f, 0.f, 0.f, 0.f};
+    {
+      const int key = slab * 16 + col;
+#pragma unroll
+      for (int ks = 0; ks < KS; ++ks) {
+        const int d8 = ks * 32 + kgrp * 8;
+        bf16x8_t bfrag =
+            *reinterpret_cast<const bf16x8_t*>(&X[key * D + swz(key, d8)]);
+        s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks], bfrag, s, 0, 0, 0);
+      }
+    }
+    float sv[4];
+    {
+      const int key = base + slab * 16 + col;
+      const bool dead = key >= L || key < start;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        float x = s[reg] * scale;
+        if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
+        sv[reg] = dead ? -1e30f : x;
+        float m = sv[reg];
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) m = fmaxf(m, __shfl_xor(m, off, WAVE));
+        if (col == 0 && wid < SLABS) mpart[slab * 16 + kgrp * 4 + reg] = m;
+      }
+    }
+    pipe_barrier();  // S reads of X done everywhere; mpart visible
+
+    // ---- issue V(base)->X then K(next)->Y; both stream under softmax/PV
+    const int next = base + PD_KT;
+    const bool prefetch = next < range_hi && !is_tail(next);
+    if (tail) {
+      stage_v_plain(base, X);
+    } else {
+      issue_v(base, X);
+    }
+    if (prefetch) issue_k(next, Y);
+
+    // ---- combine maxes, build P, update l (redundant on every wave)
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int row = kgrp * 4 + reg;
+      float m_tile = mpart[row];
+#pragma unroll
+      for (int w = 1; w < SLABS; ++w) m_tile = fmaxf(m_tile, mpart[w * 16 + row]);
+      const float m_new = fmaxf(m_regs[reg], m_tile);
+      const float alpha = (m_new > -1e30f) ? __expf(m_regs[reg] - m_new) : 1.f;
+      const float pe = (m_new > -1e30f && sv[reg] > -1e29f)
+                           ? __expf(sv[reg] - m_new) : 0.f;
+      float lsum = pe;
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) lsum += __shfl_xor(lsum, off, WAVE);
+      l_regs[reg] = l_regs[reg] * alpha + lsum;
+      m_regs[reg] = m_new;
+      if (wid == 0 && col == 0) alpha_s[row] = alpha;
+      p_lds2[row * PD_KT + slab * 16 + col] =
+          __bfloat16_as_short(__float2bfloat16(pe));
+    }
+    if (tail) {
+      pipe_barrier();  // V (plain) + P + alpha ready; nothing in flight
+    } else if (prefetch) {
+      pipe_barrier_vm<NI_K>();  // V landed; K(next) stays in flight
+    } else {
+      pipe_barrier_vm<0>();
+    }
+
+    // ---- OT[dims,16q] += V^T P^T over this wave's dim slab
+    const float alpha_q = alpha_s[col];
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
+      ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
+      const int dtile = (wid * D4) / 16 + dt;
+#pragma unroll
+      for (int ks = 0; ks < PD_KT / 32; ++ks) {
+        const int sub = (ks * (D / 16) + dtile) * 528 + lane * 4;
+        bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_bf16x4*)&X[sub]);
+        bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_bf16x4*)&X[sub + 4 * 64]);
+        bf16x8_t a;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          a[j] = __bfloat16_as_short((__hip_bfloat16)lo[j]);
+          a[4 + j] = __bfloat16_as_short((__hip_bfloat16)hi[j]);
+        }
+        bf16x8_t bb = *reinterpret_cast<const bf16x8_t*>(
+            &p_lds2[col * PD_KT + ks * 32 + kgrp * 8]);
+        ot[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bb, ot[dt], 0, 0, 0);
+      }
+    }
+    // No trailing barrier: the next loop-top barrier (vmcnt(0)) both
+    // protects X against the glds of chunk base+2 (issued only after the
+    // NEXT post-S barrier) and confirms K(next) landed in Y.
+  }
+
+  // ---- merge per-slab l, then store (normalised out, or raw partials)
+  if (col == 0 && wid < SLABS) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) l_part[slab * 16 + kgrp * 4 + reg] = l_regs[reg];
+  }
+  pipe_barrier();
+  float l_tot = l_part[col];
+#pragma unroll
+  for (int w = 1; w < SLABS; ++w) l_tot += l_part[w * 16 + col];
+  if (nsplit > 1) {
+    float* slot = scratch +
+        ((((long)b * num_kv_heads + kh) * nsplit + blockIdx.z) * G) * (D + 2);
+    if (col < G) {
+      float* row = slot + (long)col * (D + 2);
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        const int dim0 = wid * D4 + dt * 16 + kgrp * 4;
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) row[dim0 + reg] = ot[dt][reg];
+      }
+    }
+    if (wid == 0 && col == 0) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int r = kgrp * 4 + reg;
+        if (r < G) {
+          float* row = slot + (long)r * (D + 2);
+          row[D] = m_regs[reg];
+          float lt = l_part[r];
+          for (int w = 1; w < SLABS; ++w) lt += l_part[w * 16 + r];
+          row[D + 1] = lt;
+        }
+      }
+    }
+    return;
+  }
+  const float inv = (l_tot > 0.f) ? 1.0f / l_tot : 0.f;
+  if (col < G) {
+    __hip_bfloat16* op = out + (long)b * out_stride + (long)(kh * G + col) * D;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      const int dim0 = wid * D4 + dt * 16 + kgrp * 4;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg)
+        op[dim0 + reg] = __float2bfloat16(ot[dt][reg] * inv);
+    }
+  }
+}
+
 // Combine split-KV partials: out[b, kh*G+g] = Σ_s w_s·acc_s / Σ_s w_s·l_s,
 // w_s = exp(m_s − max_s m). One wave per (b, kh, g).
 template <int HEAD_DIM>

@@ -254,6 +254,42 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
       }
       dim3 sgrid(B, KVH, nsplit);
       const bool fp8_cache = kc.scalar_type() == at::kFloat8_e4m3fn;
+      // Software-pipelined glds kernel (default): LLMQ_DECODE_PIPE selects
+      // the chunk size (64 | 128 keys) or 0 = the plain-staged kernel.
+      // Read per call so microbenches can A/B in-process. Preconditions:
+      // bf16 cache (glds cannot convert fp8 during the LDS write) and the
+      // whole block table staged in LDS (no per-granule LDS-vs-global
+      // select — guide §5 trap 4c).
+      const char* pe = getenv("LLMQ_DECODE_PIPE");
+      const int pipe_kt = pe ? atoi(pe) : 64;
+      if (!fp8_cache && pipe_kt > 0 && max_blocks <= PD_MAX_BT) {
+        auto lp = [&]<int HD, int KT>() {
+          hipLaunchKernelGGL((paged_decode_pipe_kernel<HD, 8, KT>), sgrid,
+                             dim3(8 * 64), 0, stream(),
+                             reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                             reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                             reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()),
+                             reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()),
+                             bt.data_ptr<int>(), cl.data_ptr<int>(), scratch_ptr,
+                             H, KVH, bs, max_blocks, (float)scale,
+                             (float)softcap, (int)window, q.stride(0),
+                             out.stride(0));
+          if (nsplit > 1) {
+            hipLaunchKernelGGL((decode_splitkv_merge_kernel<HD>),
+                               dim3(B, KVH, G), dim3(64), 0, stream(),
+                               reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                               scratch_ptr, KVH, G, nsplit, out.stride(0));
+          }
+        };
+        if (D == 128) {
+          if (pipe_kt >= 128) lp.template operator()<128, 128>();
+          else lp.template operator()<128, 64>();
+        } else {
+          if (pipe_kt >= 128) lp.template operator()<256, 128>();
+          else lp.template operator()<256, 64>();
+        }
+        return;
+      }
       auto lm = [&]<int HD, int NW, typename TC>() {
         hipLaunchKernelGGL((paged_decode_mfma_kernel<HD, NW, TC>), sgrid,
                            dim3(NW * 64), 0, stream(),

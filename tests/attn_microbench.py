@@ -38,10 +38,15 @@ def main() -> None:
     ap.add_argument("--check", action="store_true", help="verify vs torch ref")
     ap.add_argument("--prefill", action="store_true",
                     help="time varlen flash prefill instead of decode")
+    ap.add_argument("--ab", action="store_true",
+                    help="interleaved A/B of decode variants (LLMQ_DECODE_PIPE"
+                    " 0/64/128) with per-variant numerics check")
     args = ap.parse_args()
 
     if args.prefill:
         return prefill_bench(args)
+    if args.ab:
+        return ab_bench(args)
     use_gpu = torch.cuda.is_available()
     if use_gpu:
         assert ops.has_hip_ext()
@@ -98,6 +103,69 @@ def main() -> None:
         err = (out.float() - ref).abs().max().item()
         print(f"max|err| vs f32 ref: {err:.4e}")
         assert err < 0.05
+
+
+def ab_bench(args) -> None:
+    """Within-probe interleaved A/B of the decode kernel variants (guide
+    §5.4 rule 24: perf-Δ claims come from interleaved rounds in ONE
+    process; cross-process noise on these boxes is ~±7%)."""
+    assert torch.cuda.is_available() and ops.has_hip_ext()
+    dev = torch.device("cuda:0")
+    B, L, H, KVH, D, bs = (args.batch, args.ctx, args.heads, args.kv_heads,
+                           args.head_dim, args.block_size)
+    blocks_per_seq = (L + bs - 1) // bs
+    num_blocks = B * blocks_per_seq + 1
+    torch.manual_seed(0)
+    q = torch.randn(B, H, D, device=dev, dtype=torch.bfloat16)
+    k_cache = torch.randn(num_blocks, KVH, bs, D, device=dev, dtype=torch.bfloat16)
+    v_cache = torch.randn(num_blocks, KVH, bs, D, device=dev, dtype=torch.bfloat16)
+    block_tables = torch.arange(
+        1, 1 + B * blocks_per_seq, device=dev, dtype=torch.int32
+    ).reshape(B, blocks_per_seq)
+    context_lens = torch.full((B,), L, device=dev, dtype=torch.int32)
+    scale = D ** -0.5
+    kv_bytes = 2 * B * L * KVH * D * 2
+
+    variants = ["0", "64", "128"]
+
+    def run_once():
+        return ops.paged_decode_attention(
+            q, k_cache, v_cache, block_tables, context_lens,
+            scale, args.softcap, args.window,
+        )
+
+    # numerics first
+    from llmq_amd.ops import torch_ref
+    ref = torch_ref.paged_decode_attention(
+        q.float(), k_cache.float(), v_cache.float(), block_tables,
+        context_lens, scale, args.softcap, args.window,
+    ).float()
+    for v in variants:
+        os.environ["LLMQ_DECODE_PIPE"] = v
+        err = (run_once().float() - ref).abs().max().item()
+        print(f"variant pipe={v}: max|err| vs f32 ref = {err:.4e}")
+        assert err < 0.05, f"variant {v} numerics FAIL"
+
+    rounds = 6
+    times = {v: [] for v in variants}
+    for r in range(rounds):
+        for v in variants:
+            os.environ["LLMQ_DECODE_PIPE"] = v
+            for _ in range(3):
+                run_once()
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(args.iters):
+                run_once()
+            torch.cuda.synchronize()
+            times[v].append((time.perf_counter() - t0) / args.iters)
+    print(f"\nshape B={B} L={L} H={H} KVH={KVH} D={D} bs={bs}  "
+          f"({rounds} interleaved rounds × {args.iters} iters)")
+    for v in variants:
+        ts = sorted(times[v])
+        med, best = ts[len(ts) // 2], ts[0]
+        print(f"  pipe={v:>3}: median {med * 1e3:.3f} ms  best {best * 1e3:.3f} ms"
+              f"   median {kv_bytes / med / 1e12:.2f} TB/s  best {kv_bytes / best / 1e12:.2f} TB/s")
 
 
 def prefill_bench(args) -> None:

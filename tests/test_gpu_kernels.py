@@ -166,6 +166,33 @@ class TestDecodeAttention:
         atol, rtol = TOL[dtype]
         assert_close_to_f32_ref(out.cpu(), ref, atol, 5 * rtol)
 
+    @pytest.mark.parametrize("pipe", ["0", "64", "128"])
+    @pytest.mark.parametrize("B,G,KVH,D,softcap,window", [
+        (9, 2, 8, 256, 0.0, 0),    # headline gemma2 shape
+        (9, 4, 8, 128, 0.0, 0),
+        (5, 2, 8, 256, 50.0, 100), # softcap + window (start mid-chunk)
+        (2, 16, 1, 256, 0.0, 0),   # TP-rank shape: split-KV z-grid engaged
+    ])
+    def test_pipe_variants(self, pipe, B, G, KVH, D, softcap, window, monkeypatch):
+        """The software-pipelined glds decode kernel (LLMQ_DECODE_PIPE=64/128)
+        must match the plain-staged kernel's reference across chunk-boundary
+        contexts (tail chunk short/exact/overflowing) and the split-KV path."""
+        monkeypatch.setenv("LLMQ_DECODE_PIPE", pipe)
+        dtype = torch.bfloat16
+        H = G * KVH
+        torch.manual_seed(4)
+        all_ctx = [1, 63, 64, 65, 127, 128, 129, 200, 1025]
+        ctx = torch.tensor(all_ctx[-B:], dtype=torch.int32, device=DEV)
+        kc, vc, bt = _build_cache(B, KVH, D, 16, 1025, dtype)
+        q = torch.randn(B, H, D, device=DEV, dtype=dtype)
+        scale = D ** -0.5
+        out = ops.paged_decode_attention(q, kc, vc, bt, ctx, scale, softcap, window)
+        ref = torch_ref.paged_decode_attention(
+            q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(), ctx.cpu(),
+            scale, softcap, window,
+        )
+        assert_close_to_f32_ref(out.cpu(), ref, 2e-2, 1e-1)
+
     @pytest.mark.parametrize("softcap,window", [(50.0, 0), (0.0, 64), (50.0, 64)])
     def test_softcap_window(self, softcap, window):
         dtype = torch.bfloat16
