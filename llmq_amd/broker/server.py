@@ -338,11 +338,48 @@ class BrokerServer:
     # -- connection handling ---------------------------------------------
 
     async def _handle_conn(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
+        # Dual protocol on one port: an AMQP 0-9-1 client opens with the
+        # 8-byte protocol header; the in-tree protocol's frames are
+        # '{'-prefixed JSON lines. Sniff the first bytes.
+        try:
+            first = await reader.readexactly(1)
+        except (asyncio.IncompleteReadError, ConnectionError):
+            writer.close()
+            return
+        if first == b"A":
+            try:
+                rest = await reader.readexactly(7)
+            except (asyncio.IncompleteReadError, ConnectionError):
+                writer.close()
+                return
+            if rest[:3] == b"MQP":
+                from llmq_amd.broker.amqp_server import handle_amqp_connection
+
+                try:
+                    await handle_amqp_connection(self, reader, writer)
+                finally:
+                    try:
+                        writer.close()
+                    except Exception:
+                        pass
+                return
+            writer.close()
+            return
         conn = Connection(self, writer)
+        pending_first: Optional[bytes] = first
         try:
             while True:
                 try:
-                    frame = await protocol.read_frame(reader)
+                    if pending_first is not None:
+                        line = pending_first + await reader.readline()
+                        pending_first = None
+                        if not line.endswith(b"\n"):
+                            break
+                        import json as _json
+
+                        frame = _json.loads(line)
+                    else:
+                        frame = await protocol.read_frame(reader)
                 except (asyncio.IncompleteReadError, ConnectionError):
                     break
                 except ValueError:

@@ -58,7 +58,28 @@ class Delivery:
             )
 
 
+AMQP_CLIENT_CLS = None  # set by llmq_amd.core.amqp_client on import
+
+
 class BrokerClient:
+    """Factory + default (in-tree JSON protocol) backend.
+
+    ``BrokerClient(config)`` returns the AMQP 0-9-1 backend
+    (core/amqp_client.py) when the broker URL scheme is ``amqp://`` /
+    ``amqps://`` — the reference's RabbitMQ deployment story — and this
+    class for ``llmq://`` (the in-tree broker, which itself also answers
+    AMQP on the same port)."""
+
+    def __new__(cls, config: Optional[Config] = None):
+        if cls is BrokerClient:
+            cfg = config or get_config()
+            if cfg.broker_url.split("://", 1)[0].lower() in ("amqp", "amqps"):
+                global AMQP_CLIENT_CLS
+                if AMQP_CLIENT_CLS is None:
+                    from llmq_amd.core import amqp_client  # noqa: F401 — registers
+                return super().__new__(AMQP_CLIENT_CLS)
+        return super().__new__(cls)
+
     def __init__(self, config: Optional[Config] = None):
         self.config = config or get_config()
         self._reader: Optional[asyncio.StreamReader] = None

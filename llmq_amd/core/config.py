@@ -137,9 +137,10 @@ class Config(BaseModel):
 
 
 def _parse_url(url: str) -> tuple[str, int]:
-    """Parse llmq://host:port (scheme optional)."""
+    """Parse llmq://host:port or amqp://user:pass@host:port/vhost."""
     rest = url.split("://", 1)[-1]
-    rest = rest.rstrip("/")
+    rest = rest.split("/", 1)[0]        # drop vhost/path
+    rest = rest.rsplit("@", 1)[-1]      # drop credentials
     if ":" in rest:
         host, _, port = rest.rpartition(":")
         return host or "127.0.0.1", int(port)
