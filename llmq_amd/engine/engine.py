@@ -33,8 +33,10 @@ _RESERVE_BYTES = 4 << 30
 
 # Mixed steps fuse decode+prefill into one forward (shared GEMMs); below
 # this many prefill tokens the fusion is not worth losing the decode
-# hipGraph, so the two segments run as separate forwards.
-_MIXED_FUSE_MIN_TOKENS = 128
+# hipGraph, so the two segments run as separate forwards. Env-overridable
+# for steady-state A/B (0 = always fuse; huge = always split and keep the
+# decode hipGraph replaying under continuous arrivals).
+_MIXED_FUSE_MIN_TOKENS = int(os.environ.get("LLMQ_MIXED_FUSE_MIN_TOKENS", "128"))
 
 _TUNED_GEMMS_DONE = False
 
@@ -233,10 +235,13 @@ class LLMEngine:
 
     # -- stepping --------------------------------------------------------
 
+    last_step_kind: str = "decode"  # observability: kind of the last step
+
     def step(self) -> List[RequestOutput]:
         batch = self.scheduler.schedule()
         if batch.empty:
             return []
+        self.last_step_kind = batch.kind
         t0 = time.perf_counter()
         if batch.kind == "prefill":
             tokens = self.runner.execute_prefill(batch.seqs, batch.chunks or None)

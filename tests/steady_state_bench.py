@@ -1,0 +1,123 @@
+#!/usr/bin/env python3
+"""Steady-state serving benchmark (GPU box tool, not a pytest).
+
+Round-1 bench/soak runs were mostly burst-then-drain: admissions happened
+up front, then pure-decode steps replayed the hipGraph. Production serving
+is CONTINUOUS arrivals — jobs finish every step and new prompts prefill in
+their place, so most steps are MIXED and (above the fuse threshold) run
+eager, forfeiting the decode graph (VERDICT r1 weakness 5). This tool
+measures that regime: a feeder keeps the waiting queue non-empty, every
+completed request is immediately replaced, and the report splits steps by
+kind.
+
+A/B the two mixed-step strategies in one process:
+  python tests/steady_state_bench.py --seconds 30            # default fuse
+  LLMQ_MIXED_FUSE_MIN_TOKENS=1000000 python ... (always split: graph decode
+  + separate eager prefill)
+  LLMQ_MIXED_FUSE_MIN_TOKENS=0 python ...        (always fused eager)
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+import torch
+
+from llmq_amd.engine.config import EngineConfig
+from llmq_amd.engine.sampling_params import SamplingParams
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="tower-plus-9b")
+    ap.add_argument("--batch", type=int, default=256)
+    ap.add_argument("--prompt-len", type=int, default=1024)
+    ap.add_argument("--output-len", type=int, default=128)
+    ap.add_argument("--seconds", type=float, default=30.0)
+    ap.add_argument("--max-model-len", type=int, default=4096)
+    args = ap.parse_args()
+
+    from llmq_amd.engine.engine import LLMEngine
+
+    use_gpu = torch.cuda.is_available()
+    model = args.model if use_gpu else "tiny-llama"
+    batch = args.batch if use_gpu else 8
+    plen = args.prompt_len if use_gpu else 32
+    olen = args.output_len if use_gpu else 8
+
+    engine = LLMEngine(EngineConfig(
+        model=model, max_num_seqs=batch,
+        max_model_len=args.max_model_len if use_gpu else 256,
+        max_prefill_tokens=8192, load_weights=False, fast_init=True,
+        device="cuda:0" if use_gpu else "cpu", hipgraph_max_batch=batch,
+    ))
+    rng = np.random.default_rng(7)
+    vocab = engine.spec.vocab_size
+    params = SamplingParams(temperature=0.7, max_tokens=olen, ignore_eos=True)
+    nsub = 0
+
+    def feed(n: int) -> None:
+        nonlocal nsub
+        for _ in range(n):
+            ids = rng.integers(0, vocab, size=plen).tolist()
+            engine.add_request(f"ss-{nsub}", prompt_token_ids=ids, params=params)
+            nsub += 1
+
+    # warm to steady state: full residency + hot graphs
+    feed(batch + 32)
+    t_warm = time.perf_counter()
+    while engine.scheduler.num_running < batch and time.perf_counter() - t_warm < 300:
+        engine.step()
+    for _ in range(16):
+        engine.step()
+
+    kinds = {"decode": [0, 0.0], "mixed": [0, 0.0], "prefill": [0, 0.0]}
+    out_tokens = 0
+    completed = 0
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < args.seconds:
+        # keep the admission queue fed (the broker-prefetch regime)
+        deficit = (batch + 32) - (engine.scheduler.num_running
+                                  + engine.scheduler.num_waiting
+                                  + len(engine.scheduler.prefilling))
+        if deficit > 0:
+            feed(deficit)
+        ts = time.perf_counter()
+        outs = engine.step()
+        if use_gpu:
+            torch.cuda.synchronize()
+        dt = time.perf_counter() - ts
+        # classify the step after the fact from what it produced
+        kind = engine.last_step_kind
+        kinds[kind][0] += 1
+        kinds[kind][1] += dt
+        for o in outs:
+            out_tokens += len(o.new_token_ids)
+            if o.finished:
+                completed += 1
+    elapsed = time.perf_counter() - t0
+
+    fuse_env = os.environ.get("LLMQ_MIXED_FUSE_MIN_TOKENS", "128 (default)")
+    report = {
+        "mode": f"LLMQ_MIXED_FUSE_MIN_TOKENS={fuse_env}",
+        "model": model, "batch": batch, "prompt_len": plen, "output_len": olen,
+        "seconds": round(elapsed, 2),
+        "output_tok_per_s": round(out_tokens / elapsed, 1),
+        "jobs_per_s": round(completed / elapsed, 2),
+        "steps": {k: {"n": v[0], "avg_ms": round(v[1] / v[0] * 1e3, 2) if v[0] else 0}
+                  for k, v in kinds.items()},
+    }
+    print(json.dumps(report))
+
+
+if __name__ == "__main__":
+    main()
