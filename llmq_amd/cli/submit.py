@@ -33,6 +33,68 @@ from llmq_amd.utils.template import create_job_from_data
 logger = logging.getLogger(__name__)
 
 
+class SubmitProgress:
+    """Rich progress with live submit/complete rates (reference parity:
+    submit.py:350-364,437-448). Renders on stderr ONLY when it is a
+    terminal, so ``--stream``'s JSONL on stdout and piped/captured stderr
+    stay clean; otherwise falls back to periodic plain lines."""
+
+    def __init__(self, stream: bool, total: Optional[int] = None):
+        from rich.console import Console
+
+        self._console = Console(file=sys.stderr)
+        self.enabled = self._console.is_terminal
+        self._progress = None
+        self._submit_task = None
+        self._recv_task = None
+        self.stream = stream
+        self.total = total
+
+    def __enter__(self) -> "SubmitProgress":
+        if not self.enabled:
+            return self
+        from rich.progress import (
+            BarColumn, MofNCompleteColumn, Progress, ProgressColumn,
+            SpinnerColumn, TextColumn, TimeElapsedColumn,
+        )
+        from rich.text import Text
+
+        class RateColumn(ProgressColumn):
+            def render(self, task):  # noqa: ANN001
+                speed = task.speed
+                return Text(f"{speed:.1f} jobs/s" if speed else "-- jobs/s",
+                            style="progress.data.speed")
+
+        self._progress = Progress(
+            SpinnerColumn(), TextColumn("[progress.description]{task.description}"),
+            BarColumn(), MofNCompleteColumn(), RateColumn(), TimeElapsedColumn(),
+            console=self._console, transient=False,
+            # rich's Live redirects stdout into the progress console by
+            # default — that would swallow --stream's result JSONL. Keep
+            # both streams untouched; the bar owns only its own console.
+            redirect_stdout=False, redirect_stderr=False,
+        )
+        self._progress.__enter__()
+        self._submit_task = self._progress.add_task("submit", total=self.total)
+        if self.stream:
+            self._recv_task = self._progress.add_task("results", total=self.total)
+        return self
+
+    def __exit__(self, *exc) -> None:
+        if self._progress is not None:
+            self._progress.__exit__(*exc)
+
+    def update_submitted(self, n: int) -> None:
+        if self._progress is not None:
+            self._progress.update(self._submit_task, completed=n)
+        else:
+            print(f"  submitted {n} jobs...", file=sys.stderr, end="\r")
+
+    def update_received(self, n: int, total: Optional[int] = None) -> None:
+        if self._progress is not None and self._recv_task is not None:
+            self._progress.update(self._recv_task, completed=n, total=total)
+
+
 def _looks_like_dataset(source: str) -> bool:
     if source == "-":
         return False
@@ -93,6 +155,7 @@ class JobSubmitter:
         self.client = BrokerClient(self.config)
         self.submitted = 0
         self.received = 0
+        self._progress = SubmitProgress(stream=False)  # replaced in run()
         self._interrupted = 0
 
     def _rows(self) -> Iterator[Dict[str, Any]]:
@@ -124,25 +187,30 @@ class JobSubmitter:
         start = time.time()
         chunk: List[Job] = []
         skipped = 0
-        for idx, row in enumerate(self._rows()):
-            if self._interrupted:
-                break
-            if self.limit is not None and self.submitted + len(chunk) >= self.limit:
-                break
-            try:
-                job = self._make_job(idx, row)
-            except (ValueError, KeyError) as exc:
-                skipped += 1
-                logger.warning("skipping row %d: %s", idx, exc)
-                continue
-            chunk.append(job)
-            if self.stream:
-                expected_ids.add(job.id)
-            if len(chunk) >= self.config.chunk_size:
+        self._progress = SubmitProgress(self.stream, total=self.limit)
+        with self._progress:
+            for idx, row in enumerate(self._rows()):
+                if self._interrupted:
+                    break
+                if self.limit is not None and self.submitted + len(chunk) >= self.limit:
+                    break
+                try:
+                    job = self._make_job(idx, row)
+                except (ValueError, KeyError) as exc:
+                    skipped += 1
+                    logger.warning("skipping row %d: %s", idx, exc)
+                    continue
+                chunk.append(job)
+                if self.stream:
+                    expected_ids.add(job.id)
+                if len(chunk) >= self.config.chunk_size:
+                    await self._submit_chunk(chunk)
+                    chunk = []
+            if chunk and not self._interrupted:
                 await self._submit_chunk(chunk)
-                chunk = []
-        if chunk and not self._interrupted:
-            await self._submit_chunk(chunk)
+            if stream_task is not None:
+                self._progress.update_received(self.received, total=self.submitted)
+                await stream_task
         elapsed = max(time.time() - start, 1e-9)
         print(
             f"Submitted {self.submitted} jobs to '{self.queue_name}' "
@@ -151,7 +219,6 @@ class JobSubmitter:
             file=sys.stderr,
         )
         if stream_task is not None:
-            await stream_task
             print(
                 f"Received {self.received}/{self.submitted} results", file=sys.stderr
             )
@@ -167,7 +234,7 @@ class JobSubmitter:
     async def _submit_chunk(self, jobs: List[Job]) -> None:
         await self.client.publish_jobs(self.queue_name, jobs)
         self.submitted += len(jobs)
-        print(f"  submitted {self.submitted} jobs...", file=sys.stderr, end="\r")
+        self._progress.update_submitted(self.submitted)
 
     async def _consume_results(self, expected_ids: set[str], idle_timeout: float = 60.0) -> None:
         done = asyncio.Event()
@@ -185,6 +252,8 @@ class JobSubmitter:
                 expected_ids.discard(result.id)
                 print(delivery.body, flush=True)
                 self.received += 1
+                self._progress.update_received(
+                    self.received, total=self.submitted or None)
                 await delivery.ack()
                 if not expected_ids and self.submitted:
                     done.set()
