@@ -371,6 +371,7 @@ class AMQPSession:
         s = self.server
         q = s._declare(rkey)
         s._publish(q, body, props.get("message_id", ""))
+        s._sync_if_required(q)  # fsync-before-confirm when enabled
         s._kick(q)
         if conn.confirm_mode:
             conn.publish_seq += 1

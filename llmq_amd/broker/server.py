@@ -26,6 +26,7 @@ from __future__ import annotations
 
 import asyncio
 import logging
+import os
 import time
 from collections import deque
 from dataclasses import dataclass, field
@@ -72,6 +73,7 @@ class Queue:
         data_dir: Optional[Path],
         ttl_ms: int = 0,
         max_retries: int = 3,
+        journal_fsync: bool = False,
     ):
         self.name = name
         self.durable = durable
@@ -84,7 +86,7 @@ class Queue:
         self._seq = 0
         self.journal: Optional[Journal] = None
         if durable and data_dir is not None:
-            self.journal = Journal(data_dir, name)
+            self.journal = Journal(data_dir, name, fsync=journal_fsync)
             for seq, msg_id, body, attempts in self.journal.load():
                 self.ready.append(Message(seq, msg_id, body, attempts))
                 self._seq = max(self._seq, seq)
@@ -160,10 +162,21 @@ class BrokerServer:
         data_dir: Optional[str] = None,
         max_retries: int = 3,
         default_ttl_ms: int = 0,
+        journal_fsync: Optional[bool] = None,
     ):
         self.host = host
         self.port = port
         self.data_dir = Path(data_dir) if data_dir else None
+        # Durability window: by default journals are flush()ed (no fsync)
+        # every 50 ms — a crash of the HOST (not just the broker process)
+        # can lose the last <=50 ms of publishes. LLMQ_JOURNAL_FSYNC=1 (or
+        # journal_fsync=True) fsyncs at every publish(-batch) boundary
+        # BEFORE the client's confirm, matching RabbitMQ's
+        # persistent-message guarantee at a throughput cost.
+        if journal_fsync is None:
+            journal_fsync = os.environ.get("LLMQ_JOURNAL_FSYNC", "").lower() in (
+                "1", "true", "yes", "on")
+        self.journal_fsync = journal_fsync
         self.max_retries = max_retries
         self.default_ttl_ms = default_ttl_ms
         self.queues: Dict[str, Queue] = {}
@@ -245,6 +258,7 @@ class BrokerServer:
                 self.data_dir,
                 ttl_ms if ttl_ms is not None else self.default_ttl_ms,
                 self.max_retries,
+                journal_fsync=self.journal_fsync,
             )
             self.queues[name] = q
         elif ttl_ms is not None:
@@ -329,6 +343,12 @@ class BrokerServer:
             )
             self._send_tasks.add(t)
             t.add_done_callback(self._send_tasks.discard)
+
+    def _sync_if_required(self, q: Queue) -> None:
+        """fsync-before-confirm durability (LLMQ_JOURNAL_FSYNC)."""
+        if self.journal_fsync and q.journal:
+            q.journal.flush()
+            self._dirty_journals.discard(q.name)
 
     def _ack_internal(self, q: Queue, msg: Message) -> None:
         if q.journal:
@@ -436,6 +456,7 @@ class BrokerServer:
         if method == "publish":
             q = self._declare(f["queue"])
             self._publish(q, f["body"], f.get("id", ""))
+            self._sync_if_required(q)
             self._kick(q)
             return None
 
@@ -443,6 +464,7 @@ class BrokerServer:
             q = self._declare(f["queue"])
             for item in f["items"]:
                 self._publish(q, item["body"], item.get("id", ""))
+            self._sync_if_required(q)
             self._kick(q)
             return {"count": len(f["items"])}
 
