@@ -32,7 +32,7 @@ def main() -> None:
     parser.add_argument("--steps", type=int, default=64)
     parser.add_argument("--warmup", type=int, default=16)
     parser.add_argument("--model", default="tower-plus-9b")
-    parser.add_argument("--batch", type=int, default=256,
+    parser.add_argument("--batch", type=int, default=512,
                         help="resident sequences per GPU (max_num_seqs)")
     parser.add_argument("--prompt-len", type=int, default=1024)
     parser.add_argument("--max-model-len", type=int, default=4096)

@@ -31,12 +31,16 @@ logger = logging.getLogger(__name__)
 # Memory the engine leaves free on top of weights+KV (workspace, graphs).
 _RESERVE_BYTES = 4 << 30
 
-# Mixed steps fuse decode+prefill into one forward (shared GEMMs); below
-# this many prefill tokens the fusion is not worth losing the decode
-# hipGraph, so the two segments run as separate forwards. Env-overridable
-# for steady-state A/B (0 = always fuse; huge = always split and keep the
-# decode hipGraph replaying under continuous arrivals).
-_MIXED_FUSE_MIN_TOKENS = int(os.environ.get("LLMQ_MIXED_FUSE_MIN_TOKENS", "128"))
+# Mixed-step policy: fusing decode+prefill into one forward (shared GEMMs)
+# measured WORSE than splitting at every admission size on MI355X
+# (profiles/r2_step2: steady-state 3,967 fused vs 4,247 split tok/s) — the
+# decode segment loses its hipGraph and the fused eager forward costs more
+# than the extra weight read. Default: ALWAYS split (replay the decode
+# graph + run the prefill segment separately, overlapped on two streams).
+# Env-overridable for A/B: a small value re-enables fusion above that many
+# prefill tokens.
+_MIXED_FUSE_MIN_TOKENS = int(
+    os.environ.get("LLMQ_MIXED_FUSE_MIN_TOKENS", str(1 << 30)))
 
 _TUNED_GEMMS_DONE = False
 
@@ -140,6 +144,16 @@ class LLMEngine:
         )
         self.runner = ModelRunner(
             self.model, self.kv_cache, config, self.device, self.max_model_len
+        )
+        # Mixed-step overlap: decode graph on a side stream ∥ prefill on the
+        # default stream (disjoint KV slots; separate sampler buffers).
+        self._overlap_mixed = (
+            self.device.type == "cuda"
+            and tp == 1
+            and os.environ.get("LLMQ_OVERLAP_MIXED", "1") not in ("0", "false")
+        )
+        self._decode_stream = (
+            torch.cuda.Stream(device=self.device) if self._overlap_mixed else None
         )
         self.runner.capture_graphs()
         self._seqs: Dict[str, Sequence] = {}
@@ -248,18 +262,34 @@ class LLMEngine:
         elif batch.kind == "mixed":
             ptoks = sum(e - st for st, e in batch.chunks)
             if ptoks < _MIXED_FUSE_MIN_TOKENS and self.runner.use_graphs:
-                # Tiny admission riding a big decode batch: the shared-GEMM
-                # win is small but fusing forces the whole decode EAGER —
-                # replay the decode hipGraph and prefill separately instead.
-                # .clone(): the fused sampler returns a VIEW of a persistent
-                # output buffer that the second _sample call would overwrite
-                d_tokens = self.runner.execute_decode(
-                    batch.seqs[: batch.n_decode]
-                ).clone()
-                p_tokens = self.runner.execute_prefill(
-                    batch.seqs[batch.n_decode :], batch.chunks
-                )
-                tokens = torch.cat([d_tokens, p_tokens])
+                # SPLIT policy (default, measured fastest): replay the
+                # decode hipGraph and run the prefill segment as its own
+                # forward. On one GPU the two are OVERLAPPED on separate
+                # HIP streams — decode attention is HBM-bound while prefill
+                # GEMMs are MFMA-bound, so the hardware runs them
+                # concurrently (profiles/r2_step2). TP replicas stay
+                # sequential: stream interleaving would make rank-local
+                # RNG/collective ordering nondeterministic.
+                dseqs = batch.seqs[: batch.n_decode]
+                pseqs = batch.seqs[batch.n_decode :]
+                if self._overlap_mixed:
+                    ds = self._decode_stream
+                    cur = torch.cuda.current_stream()
+                    ds.wait_stream(cur)
+                    with torch.cuda.stream(ds):
+                        # .clone(): the fused sampler returns a VIEW of a
+                        # persistent output buffer
+                        d_tokens = self.runner.execute_decode(dseqs).clone()
+                    p_tokens = self.runner.execute_prefill(
+                        pseqs, batch.chunks, buf_name="prefill"
+                    )
+                    cur.wait_stream(ds)
+                    d_tokens.record_stream(cur)
+                    tokens = torch.cat([d_tokens, p_tokens])
+                else:
+                    d_tokens = self.runner.execute_decode(dseqs).clone()
+                    p_tokens = self.runner.execute_prefill(pseqs, batch.chunks)
+                    tokens = torch.cat([d_tokens, p_tokens])
             else:
                 tokens = self.runner.execute_mixed(batch)
         else:

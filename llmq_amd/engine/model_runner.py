@@ -60,14 +60,10 @@ class ModelRunner:
         self.sampling_generator = torch.Generator(device=device if device.type == "cuda" else "cpu")
         self.sampling_generator.manual_seed(config.seed)
         self._sample_step = 0
-        self._sample_out: Optional[torch.Tensor] = None
-        self._sample_keys: Optional[torch.Tensor] = None
-        self._sample_temps: Optional[torch.Tensor] = None
-        self._sample_temps_h: Optional[torch.Tensor] = None
-        self._sample_seeds: Optional[torch.Tensor] = None
-        self._sample_seeds_h: Optional[torch.Tensor] = None
-        self._sample_pos: Optional[torch.Tensor] = None
-        self._sample_pos_h: Optional[torch.Tensor] = None
+        # Persistent sampler buffers, keyed by caller ("main" for the
+        # decode path, "prefill" for the prefill segment of an OVERLAPPED
+        # split mixed step — two concurrent streams must not share them).
+        self._sample_bufs: Dict[str, Dict[str, torch.Tensor]] = {}
         if self.use_graphs:
             self._alloc_static_buffers()
 
@@ -199,7 +195,7 @@ class ModelRunner:
         return ids, pos, slot_mapping, meta, rel_last
 
     @torch.no_grad()
-    def execute_prefill(self, seqs: List[Sequence], chunks=None) -> torch.Tensor:
+    def execute_prefill(self, seqs: List[Sequence], chunks=None, buf_name: str = "main") -> torch.Tensor:
         """Run a packed prefill segment; returns sampled next tokens for the
         rows whose context completed this step (all rows when no chunking)."""
         if chunks is None:
@@ -216,7 +212,7 @@ class ModelRunner:
         last_hidden = hidden[torch.tensor(rel_last, dtype=torch.long, device=dev)]
         logits = self.model.compute_logits(last_hidden)
         sample_seqs = [s for s, (st, e) in zip(seqs, chunks) if e == s.num_tokens]
-        return self._sample(logits, sample_seqs)
+        return self._sample(logits, sample_seqs, buf_name)
 
     # -- mixed (decode rows + ride-along prefill rows, eager) ------------
 
@@ -354,7 +350,25 @@ class ModelRunner:
 
     # -- sampling --------------------------------------------------------
 
-    def _sample(self, logits: torch.Tensor, seqs: List[Sequence]) -> torch.Tensor:
+    def _sample_buf(self, name: str, B: int, dev) -> Dict[str, torch.Tensor]:
+        buf = self._sample_bufs.get(name)
+        if buf is None or buf["out"].numel() < B:
+            cap = max(B, self.config.max_num_seqs)
+            buf = {
+                "out": torch.empty(cap, dtype=torch.int64, device=dev),
+                "keys": torch.empty(cap, dtype=torch.int64, device=dev),
+                "temps": torch.empty(cap, dtype=torch.float32, device=dev),
+                "temps_h": torch.empty(cap, dtype=torch.float32, pin_memory=True),
+                "seeds": torch.empty(cap, dtype=torch.int32, device=dev),
+                "seeds_h": torch.empty(cap, dtype=torch.int32, pin_memory=True),
+                "pos": torch.empty(cap, dtype=torch.int32, device=dev),
+                "pos_h": torch.empty(cap, dtype=torch.int32, pin_memory=True),
+            }
+            self._sample_bufs[name] = buf
+        return buf
+
+    def _sample(self, logits: torch.Tensor, seqs: List[Sequence],
+                buf_name: str = "main") -> torch.Tensor:
         dev = logits.device
         self._sample_step += 1
         simple = all(
@@ -363,19 +377,10 @@ class ModelRunner:
         if simple and dev.type == "cuda":
             # Fused one-pass HIP sampler (gumbel-max / greedy argmax).
             B = len(seqs)
-            if self._sample_out is None or self._sample_out.numel() < B:
-                cap = max(B, self.config.max_num_seqs)
-                self._sample_out = torch.empty(cap, dtype=torch.int64, device=dev)
-                self._sample_keys = torch.empty(cap, dtype=torch.int64, device=dev)
-                self._sample_temps = torch.empty(cap, dtype=torch.float32, device=dev)
-                self._sample_temps_h = torch.empty(cap, dtype=torch.float32, pin_memory=True)
-                self._sample_seeds = torch.empty(cap, dtype=torch.int32, device=dev)
-                self._sample_seeds_h = torch.empty(cap, dtype=torch.int32, pin_memory=True)
-                self._sample_pos = torch.empty(cap, dtype=torch.int32, device=dev)
-                self._sample_pos_h = torch.empty(cap, dtype=torch.int32, pin_memory=True)
-            th = self._sample_temps_h
-            sh = self._sample_seeds_h
-            ph = self._sample_pos_h
+            buf = self._sample_buf(buf_name, B, dev)
+            th = buf["temps_h"]
+            sh = buf["seeds_h"]
+            ph = buf["pos_h"]
             for i, s in enumerate(seqs):
                 th[i] = s.params.temperature
                 rs = s.params.seed
@@ -385,15 +390,15 @@ class ModelRunner:
                     v = rs & 0x7FFFFFFF
                     sh[i] = v if v else 0x1E3779B9  # remap literal seed 0
                 ph[i] = s.output_len
-            self._sample_temps[:B].copy_(th[:B], non_blocking=True)
-            self._sample_seeds[:B].copy_(sh[:B], non_blocking=True)
-            self._sample_pos[:B].copy_(ph[:B], non_blocking=True)
+            buf["temps"][:B].copy_(th[:B], non_blocking=True)
+            buf["seeds"][:B].copy_(sh[:B], non_blocking=True)
+            buf["pos"][:B].copy_(ph[:B], non_blocking=True)
             ops.sample_gumbel_argmax(
-                self._sample_out[:B], self._sample_keys[:B], logits.float(),
-                self._sample_temps[:B], self._sample_seeds[:B],
-                self._sample_pos[:B], self.config.seed, self._sample_step,
+                buf["out"][:B], buf["keys"][:B], logits.float(),
+                buf["temps"][:B], buf["seeds"][:B],
+                buf["pos"][:B], self.config.seed, self._sample_step,
             )
-            return self._sample_out[:B]
+            return buf["out"][:B]
         temps = torch.tensor([s.params.temperature for s in seqs], dtype=torch.float32, device=dev)
         tps = torch.tensor([s.params.top_p for s in seqs], dtype=torch.float32, device=dev)
         tks = torch.tensor([s.params.top_k for s in seqs], dtype=torch.int64, device=dev)
