@@ -230,15 +230,32 @@ def sample_gumbel_argmax(
     req_pos: torch.Tensor,    # [B] int32 output position (seeded rows)
     seed: int,
     step: int,
+    bounds: Optional[torch.Tensor] = None,  # [B] fp32 top-k/p keep-bound
 ) -> None:
     """Fused one-pass sampler: per-row Gumbel-max (== softmax sampling) or
-    argmax for greedy rows. Unseeded rows draw from the counter-based
-    (engine seed, step, row) stream; rows with a request seed draw from
-    (request_seed, output_position) — reproducible for a given request
-    regardless of batch placement."""
+    argmax for greedy rows, optionally truncated to {logit >= bounds[b]}
+    (top-k/top-p, from topk_topp_bound). Unseeded rows draw from the
+    counter-based (engine seed, step, row) stream; rows with a request seed
+    draw from (request_seed, output_position) — reproducible for a given
+    request regardless of batch placement."""
     assert _use_hip(logits)
     keys.zero_()
-    _EXT.sample_gumbel_argmax(out, keys, logits, temps, req_seeds, req_pos, seed, step)
+    _EXT.sample_gumbel_argmax(out, keys, logits, temps, req_seeds, req_pos,
+                              seed, step, bounds)
+
+
+def topk_topp_bound(
+    logits: torch.Tensor,   # [B, V] fp32
+    temps: torch.Tensor,    # [B] fp32
+    top_ps: torch.Tensor,   # [B] fp32
+    top_ks: torch.Tensor,   # [B] int64 (0 = off)
+) -> torch.Tensor:
+    """Per-row logit-space keep-bound realising top-p ∧ top-k truncation
+    via histogram select (3 streaming passes; no full-vocab sort)."""
+    assert _use_hip(logits)
+    out = torch.empty(logits.size(0), dtype=torch.float32, device=logits.device)
+    _EXT.topk_topp_bound(out, logits, temps, top_ps, top_ks)
+    return out
 
 
 def sample_tokens(
@@ -270,19 +287,12 @@ def _sample_tokens_gpu(
     need_topk = bool((top_ks > 0).any()) and bool((top_ks < V).any())
     need_topp = bool((top_ps < 1.0).any())
     if need_topk or need_topp:
-        sorted_l, sorted_i = torch.sort(scaled, dim=-1, descending=True)
-        ranks = torch.arange(V, device=logits.device).expand(B, V)
-        keep = torch.ones_like(sorted_l, dtype=torch.bool)
-        if need_topk:
-            ks = torch.where(top_ks > 0, top_ks, torch.full_like(top_ks, V))
-            keep &= ranks < ks.unsqueeze(1)
-        if need_topp:
-            probs = torch.softmax(sorted_l, dim=-1)
-            cum = torch.cumsum(probs, dim=-1)
-            keep &= (cum - probs) < top_ps.unsqueeze(1)
-        keep[:, 0] = True
-        sorted_l = torch.where(keep, sorted_l, torch.full_like(sorted_l, float("-inf")))
-        scaled = torch.full_like(scaled, float("-inf")).scatter(1, sorted_i, sorted_l)
+        # histogram-select keep-bound (3 streaming passes) instead of the
+        # full-vocab sort — at B=256 × V=256k the sort was a multi-ms cliff
+        bound = topk_topp_bound(logits.float(), temperatures.float(),
+                                top_ps.float(), top_ks)
+        scaled = torch.where(logits.float() >= bound.unsqueeze(1), scaled,
+                             torch.full_like(scaled, float("-inf")))
     # Gumbel-max trick: single fused sample, no multinomial sync.
     u = torch.rand(B, V, device=logits.device, generator=generator)
     gumbel = -torch.log(-torch.log(u.clamp_min(1e-20)).clamp_min(1e-20))

@@ -435,9 +435,24 @@ void varlen_prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,
 
 // ------------------------------------------------------------- sampling --
 
+void topk_topp_bound(at::Tensor out_bound, at::Tensor logits, at::Tensor temps,
+                     at::Tensor top_ps, at::Tensor top_ks) {
+  CHECK_GPU(logits);
+  TORCH_CHECK(logits.scalar_type() == at::kFloat && logits.dim() == 2);
+  TORCH_CHECK(out_bound.scalar_type() == at::kFloat);
+  TORCH_CHECK(top_ps.scalar_type() == at::kFloat);
+  TORCH_CHECK(top_ks.scalar_type() == at::kLong);
+  const int B = logits.size(0), V = logits.size(1);
+  hipLaunchKernelGGL(topk_topp_bound_kernel, dim3(B), dim3(256), 0, stream(),
+                     out_bound.data_ptr<float>(), logits.data_ptr<float>(),
+                     temps.data_ptr<float>(), top_ps.data_ptr<float>(),
+                     top_ks.data_ptr<long>(), V, logits.stride(0));
+}
+
 void sample_gumbel_argmax(at::Tensor out, at::Tensor keys, at::Tensor logits,
                           at::Tensor temps, at::Tensor req_seeds,
-                          at::Tensor req_pos, long seed, long step) {
+                          at::Tensor req_pos, long seed, long step,
+                          c10::optional<at::Tensor> bounds) {
   CHECK_GPU(logits);
   TORCH_CHECK(logits.scalar_type() == at::kFloat && logits.dim() == 2);
   TORCH_CHECK(out.scalar_type() == at::kLong);
@@ -445,6 +460,11 @@ void sample_gumbel_argmax(at::Tensor out, at::Tensor keys, at::Tensor logits,
   TORCH_CHECK(req_seeds.scalar_type() == at::kInt);
   TORCH_CHECK(req_pos.scalar_type() == at::kInt);
   const int B = logits.size(0), V = logits.size(1);
+  const float* bounds_ptr = nullptr;
+  if (bounds.has_value() && bounds->defined()) {
+    TORCH_CHECK(bounds->scalar_type() == at::kFloat && bounds->numel() == B);
+    bounds_ptr = bounds->data_ptr<float>();
+  }
   // enough splits to fill the chip at small B
   int nsplit = 1;
   while (B * nsplit < 2048 && nsplit < 64 && (V / nsplit) > 4096) nsplit *= 2;
@@ -454,7 +474,7 @@ void sample_gumbel_argmax(at::Tensor out, at::Tensor keys, at::Tensor logits,
                      logits.data_ptr<float>(), temps.data_ptr<float>(),
                      reinterpret_cast<const unsigned int*>(req_seeds.data_ptr()),
                      reinterpret_cast<const unsigned int*>(req_pos.data_ptr()),
-                     V, (unsigned int)seed, (unsigned int)step);
+                     bounds_ptr, V, (unsigned int)seed, (unsigned int)step);
   hipLaunchKernelGGL(unpack_keys_kernel, dim3((B + 255) / 256), dim3(256), 0,
                      stream(), out.data_ptr<long>(),
                      reinterpret_cast<const unsigned long long*>(keys.data_ptr()),
@@ -472,7 +492,8 @@ TORCH_LIBRARY(llmq_amd, m) {
   m.def("reshape_and_cache(Tensor key, Tensor value, Tensor(a!) k_cache, Tensor(b!) v_cache, Tensor slot_mapping) -> ()");
   m.def("paged_decode_attention(Tensor(a!) out, Tensor q, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor context_lens, float scale, float softcap, int window) -> ()");
   m.def("varlen_prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, Tensor cu_seqlens_k, int max_seqlen, float scale, float softcap, int window) -> ()");
-  m.def("sample_gumbel_argmax(Tensor(a!) out, Tensor(b!) keys, Tensor logits, Tensor temps, Tensor req_seeds, Tensor req_pos, int seed, int step) -> ()");
+  m.def("sample_gumbel_argmax(Tensor(a!) out, Tensor(b!) keys, Tensor logits, Tensor temps, Tensor req_seeds, Tensor req_pos, int seed, int step, Tensor? bounds=None) -> ()");
+  m.def("topk_topp_bound(Tensor(a!) out_bound, Tensor logits, Tensor temps, Tensor top_ps, Tensor top_ks) -> ()");
   m.def("norm_add_norm(Tensor(a!) x, Tensor(b!) residual, Tensor w_post, Tensor w_pre, float eps, float offset) -> ()");
   m.def("rope_and_cache(Tensor(a!) q, Tensor(b!) k, Tensor value, Tensor(c!) k_cache, Tensor(d!) v_cache, Tensor positions, Tensor cos_sin, Tensor slot_mapping) -> ()");
 }
@@ -487,6 +508,7 @@ TORCH_LIBRARY_IMPL(llmq_amd, CUDA, m) {
   m.impl("paged_decode_attention", &paged_decode_attention);
   m.impl("varlen_prefill_attention", &varlen_prefill_attention);
   m.impl("sample_gumbel_argmax", &sample_gumbel_argmax);
+  m.impl("topk_topp_bound", &topk_topp_bound);
   m.impl("norm_add_norm", &norm_add_norm);
   m.impl("rope_and_cache", &rope_and_cache);
 }

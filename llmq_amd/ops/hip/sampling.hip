@@ -53,6 +53,7 @@ __global__ __launch_bounds__(256) void sample_argmax_kernel(
     const float* __restrict__ temps,            // [B]
     const unsigned int* __restrict__ req_seeds, // [B] (0 = unseeded)
     const unsigned int* __restrict__ req_pos,   // [B] output position
+    const float* __restrict__ bounds,           // [B] top-k/p keep-bound or null
     int V, unsigned int seed, unsigned int step) {
   const int b = blockIdx.x;
   const int nsplit = gridDim.y;
@@ -67,10 +68,14 @@ __global__ __launch_bounds__(256) void sample_argmax_kernel(
   const unsigned int kseed = rs ? rs : seed;
   const unsigned int kstep = rs ? req_pos[b] : step;
   const unsigned int kb = rs ? 0u : (unsigned int)b;
+  // top-k/top-p truncation: only logits >= bound are candidates (the
+  // bound kernel guarantees the row max passes; greedy rows get -inf)
+  const float bound = bounds ? bounds[b] : -3e38f;
 
   float best = -3e38f;
   int best_i = v0;
   for (int v = v0 + threadIdx.x; v < v1; v += 256) {
+    if (row[v] < bound) continue;
     float x = row[v] * inv_t;
     if (!greedy) {
       const float u = uniform01(kseed, kstep, kb, v);
@@ -97,6 +102,137 @@ __global__ __launch_bounds__(256) void sample_argmax_kernel(
     }
     atomicMax(out_keys + b, pack_key(best, best_i));
   }
+}
+
+// ------------------------------------------------- top-k / top-p select --
+// Per-row LOGIT-space bound L_min such that keeping {v : logit_v >= L_min}
+// realises nucleus (top-p) ∧ top-k truncation, WITHOUT the full-vocab sort
+// the torch path needs (at B=256 × V=256k that sort is a multi-ms cliff;
+// this is 3 streaming passes ≈ 100 µs). Histogram select: pass 1 = row max;
+// pass 2 = 512-bin histogram of x=(l−m)/T over [-32, 0] with exp-mass and
+// counts; scan from the top bin until Σp ≥ top_p·Z or count ≥ k; pass 3
+// refines 512× within the crossing bin. The crossing sub-bin is kept whole
+// (overshoot ≤ its mass ≈ Z·p/512² resolution — statistically invisible;
+// the reference's vLLM sorts exactly, our CPU torch_ref too, and the
+// distribution tests compare against it with tolerance).
+// Sampling from the truncated set then uses the same one-pass gumbel-max
+// (argmax over {x: logit ≥ L_min} == renormalised truncated softmax).
+
+#define TH_NBINS 512
+#define TH_TPB 256
+#define TH_XMIN -32.0f
+
+__global__ __launch_bounds__(TH_TPB) void topk_topp_bound_kernel(
+    float* __restrict__ out_bound,     // [B] logit-space keep-bound
+    const float* __restrict__ logits,  // [B, V] fp32
+    const float* __restrict__ temps,   // [B]
+    const float* __restrict__ top_ps,  // [B]
+    const long* __restrict__ top_ks,   // [B] (0 = off)
+    int V, long stride) {
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  const float T0 = temps[b];
+  const float p_lim = top_ps[b];
+  const long k_lim = top_ks[b];
+  const bool need_p = p_lim < 1.0f;
+  const bool need_k = k_lim > 0 && k_lim < V;
+  if (T0 <= 0.f || !(need_p || need_k)) {
+    // greedy rows take argmax regardless; unconstrained rows keep all
+    if (tid == 0) out_bound[b] = -3e38f;
+    return;
+  }
+  const float inv_t = 1.0f / T0;
+  const float* row = logits + (long)b * stride;
+
+  __shared__ float h_sum[TH_NBINS];
+  __shared__ int h_cnt[TH_NBINS];
+  __shared__ float red[TH_TPB / WAVE];
+  __shared__ float sh_state[4];  // xlo, xhi, p_above(normed later), Z
+
+  // ---- pass 1: row max
+  float m = -3e38f;
+  for (int v = tid; v < V; v += TH_TPB) m = fmaxf(m, row[v]);
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    m = fmaxf(m, __shfl_xor(m, off, WAVE));
+  if ((tid & (WAVE - 1)) == 0) red[tid / WAVE] = m;
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < TH_TPB / WAVE; ++w) m = fmaxf(m, red[w]);
+    red[0] = m;
+  }
+  __syncthreads();
+  m = red[0];
+
+  float xlo = TH_XMIN, xhi = 0.0f;
+  float p_above = 0.f;   // exp-mass strictly above xhi (already kept)
+  long n_above = 0;
+  float Z = 0.f;         // total exp-mass (fixed after iter 0)
+  bool keep_all = false;
+
+  for (int iter = 0; iter < 2; ++iter) {
+    for (int i = tid; i < TH_NBINS; i += TH_TPB) { h_sum[i] = 0.f; h_cnt[i] = 0; }
+    __syncthreads();
+    const float w = (xhi - xlo) / TH_NBINS;
+    const float inv_w = 1.0f / w;
+    float tail = 0.f;  // mass at or below xlo (iter 0 only; never kept)
+    for (int v = tid; v < V; v += TH_TPB) {
+      const float x = (row[v] - m) * inv_t;
+      if (x > xhi) continue;           // already kept
+      if (x <= xlo) { tail += __expf(fmaxf(x, -80.f)); continue; }
+      int bin = (int)((x - xlo) * inv_w);
+      bin = bin < 0 ? 0 : (bin >= TH_NBINS ? TH_NBINS - 1 : bin);
+      atomicAdd(&h_sum[bin], __expf(x));
+      atomicAdd(&h_cnt[bin], 1);
+    }
+    // reduce tail mass (only meaningful on iter 0 for Z)
+    for (int off = WAVE / 2; off > 0; off >>= 1) tail += __shfl_xor(tail, off, WAVE);
+    if ((tid & (WAVE - 1)) == 0) red[tid / WAVE] = tail;
+    __syncthreads();
+    if (tid == 0) {
+      float tail_tot = 0.f;
+      for (int w2 = 0; w2 < TH_TPB / WAVE; ++w2) tail_tot += red[w2];
+      if (iter == 0) {
+        Z = tail_tot;
+        for (int i = 0; i < TH_NBINS; ++i) Z += h_sum[i];
+        sh_state[3] = Z;
+      } else {
+        Z = sh_state[3];
+      }
+      const float p_target = need_p ? p_lim * Z : 3e38f;
+      const long k_target = need_k ? k_lim : 0x7fffffffffffffffL;
+      float cum = p_above;
+      long cnt = n_above;
+      int cross = -1;
+      for (int i = TH_NBINS - 1; i >= 0; --i) {
+        cum += h_sum[i];
+        cnt += h_cnt[i];
+        if (cum >= p_target || cnt >= k_target) { cross = i; break; }
+      }
+      if (cross < 0) {
+        // never crossed within [xlo, xhi]: every candidate is kept; the
+        // sub-xlo tail is beyond fp32 softmax relevance -> keep all
+        sh_state[0] = -3e38f;
+        sh_state[1] = -3e38f;
+      } else {
+        float pa = p_above;
+        long na = n_above;
+        for (int i = TH_NBINS - 1; i > cross; --i) { pa += h_sum[i]; na += h_cnt[i]; }
+        sh_state[0] = xlo + cross * w;        // new xlo (crossing bin lo)
+        sh_state[1] = xlo + (cross + 1) * w;  // new xhi
+        red[0] = pa;
+        red[1] = (float)na;  // counts <= V = 262k << 2^24: exact in fp32
+      }
+    }
+    __syncthreads();
+    if (sh_state[0] == -3e38f && sh_state[1] == -3e38f) { keep_all = true; break; }
+    xlo = sh_state[0];
+    xhi = sh_state[1];
+    p_above = red[0];
+    n_above = (long)red[1];
+    __syncthreads();
+  }
+  if (tid == 0)
+    out_bound[b] = keep_all ? -3e38f : (m + xlo * T0);
 }
 
 __global__ void unpack_keys_kernel(long* __restrict__ out,

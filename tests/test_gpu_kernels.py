@@ -311,6 +311,93 @@ class TestFusedSampler:
         out = self._run(logits, temps)
         assert torch.equal(out, logits.argmax(dim=-1))
 
+    def test_topk_topp_bound_matches_exact_sort(self):
+        """The histogram-select keep-bound must reproduce the exact
+        (sort-based) nucleus/top-k keep set up to the documented sub-bin
+        overshoot: every exactly-kept id passes the bound, and the bound
+        admits at most a sliver of extra probability mass."""
+        torch.manual_seed(3)
+        B, V = 32, 50000
+        logits = (torch.randn(B, V, device=DEV) * 3.0).float()
+        temps = torch.full((B,), 0.8, device=DEV)
+        tps = torch.tensor([0.3, 0.7, 0.9, 0.95] * (B // 4), device=DEV)
+        tks = torch.tensor([0, 5, 100, 0] * (B // 4), dtype=torch.int64, device=DEV)
+        bound = ops.topk_topp_bound(logits, temps, tps, tks)
+        scaled = logits / temps.unsqueeze(1)
+        probs = torch.softmax(scaled, dim=-1)
+        sorted_p, sorted_i = torch.sort(probs, dim=-1, descending=True)
+        cum = torch.cumsum(sorted_p, dim=-1)
+        for b in range(B):
+            keep_p = (cum[b] - sorted_p[b]) < tps[b]
+            if tks[b] > 0:
+                keep_k = torch.arange(V, device=DEV) < tks[b]
+                keep = keep_p & keep_k
+            else:
+                keep = keep_p
+            keep[0] = True
+            exact_ids = sorted_i[b][keep]
+            passed = logits[b] >= bound[b]
+            # every exactly-kept id passes the bound
+            assert bool(passed[exact_ids].all()), f"row {b} drops exact ids"
+            # overshoot bounded: extra admitted mass < 1% of the target
+            extra = probs[b][passed].sum() - probs[b][exact_ids].sum()
+            assert float(extra) < 0.01, f"row {b} overshoot {float(extra)}"
+
+    def test_topk1_is_argmax_topk5_stays_in_set(self):
+        torch.manual_seed(4)
+        B, V = 8, 8192
+        logits = torch.randn(B, V, device=DEV).float()
+        temps = torch.full((B,), 1.0, device=DEV)
+        out = torch.empty(B, dtype=torch.int64, device=DEV)
+        keys = torch.empty(B, dtype=torch.int64, device=DEV)
+        zs = torch.zeros(B, dtype=torch.int32, device=DEV)
+        # top_k=1 → only the argmax can ever be drawn
+        b1 = ops.topk_topp_bound(logits, temps, torch.ones(B, device=DEV),
+                                 torch.ones(B, dtype=torch.int64, device=DEV))
+        for step in range(20):
+            ops.sample_gumbel_argmax(out, keys, logits, temps, zs, zs, 11, step, b1)
+            assert torch.equal(out, logits.argmax(dim=-1)), step
+        # top_k=5 → draws stay inside the top-5 set; all 5 seen eventually
+        top5 = logits.topk(5, dim=-1).indices
+        b5 = ops.topk_topp_bound(logits, temps, torch.ones(B, device=DEV),
+                                 torch.full((B,), 5, dtype=torch.int64, device=DEV))
+        seen = [set() for _ in range(B)]
+        for step in range(400):
+            ops.sample_gumbel_argmax(out, keys, logits, temps, zs, zs, 11, step, b5)
+            o = out.cpu()
+            for b in range(B):
+                assert int(o[b]) in set(top5[b].cpu().tolist()), (b, int(o[b]))
+                seen[b].add(int(o[b]))
+        assert all(len(s2) >= 3 for s2 in seen)  # not stuck on one token
+
+    def test_topp_distribution_matches_truncated_softmax(self):
+        """Gumbel-max over the bound-truncated set == renormalised nucleus
+        distribution (statistical check on a small crafted vocab)."""
+        V = 1024
+        base = torch.full((V,), -8.0, device=DEV)
+        base[:4] = torch.tensor([3.0, 2.5, 2.0, 1.0], device=DEV)
+        B = 512  # draw many rows at once (rows are independent RNG streams)
+        logits = base.expand(B, V).contiguous().float()
+        temps = torch.ones(B, device=DEV)
+        tps = torch.full((B,), 0.8, device=DEV)
+        tks = torch.zeros(B, dtype=torch.int64, device=DEV)
+        bound = ops.topk_topp_bound(logits, temps, tps, tks)
+        out = torch.empty(B, dtype=torch.int64, device=DEV)
+        keys = torch.empty(B, dtype=torch.int64, device=DEV)
+        zs = torch.zeros(B, dtype=torch.int32, device=DEV)
+        counts = torch.zeros(V)
+        iters = 40
+        for step in range(iters):
+            ops.sample_gumbel_argmax(out, keys, logits, temps, zs, zs, 5, step, bound)
+            counts += torch.bincount(out.cpu(), minlength=V)
+        # exact nucleus at p=0.8 keeps ids {0,1,2} (cum-before .47/.76/.93)
+        probs = torch.softmax(base, dim=-1)
+        keep = [0, 1, 2]
+        assert counts[3:].sum() == 0, counts[:8]
+        trunc = probs[keep] / probs[keep].sum()
+        freq = counts[keep] / counts.sum()
+        assert torch.allclose(freq, trunc.cpu(), atol=0.02), (freq, trunc)
+
     def test_deterministic_per_step(self):
         torch.manual_seed(0)
         logits = torch.randn(8, 4096, device=DEV)
