@@ -536,6 +536,345 @@ __global__ __launch_bounds__(256) void flash_prefill_bf16_kernel(
   }
 }
 
+// 16-B-per-lane global→LDS DMA, HIDDEN from hipcc in inline asm: with the
+// builtin form the compiler conservatively re-inserts `s_waitcnt vmcnt(0)`
+// before the first ds_read that may alias the DMA destination (verified in
+// the .s: it landed ahead of the PV tr16 reads, draining the K prefetch
+// every chunk). Hand-counted vmcnt in pipe_barrier_vm is the only wait
+// discipline for these, per guide §5.7 (its LDS-DMA recipe: M0 carries the
+// wave-uniform LDS byte address, saved/written/restored in ONE statement;
+// the s_nop is the SALU-write→M0-read wait state).
+DEVINL void glds16(const void* src, void* lds_dst_uniform) {
+  unsigned lds_off = (unsigned)(unsigned long)(
+      (__attribute__((address_space(3))) char*)lds_dst_uniform);
+  lds_off = __builtin_amdgcn_readfirstlane(lds_off);
+  unsigned keep;
+  asm volatile(
+      "s_mov_b32 %0, m0\n\t"
+      "s_mov_b32 m0, %2\n\t"
+      "s_nop 0\n\t"
+      "global_load_lds_dwordx4 %1, off\n\t"
+      "s_mov_b32 m0, %0"
+      : "=&s"(keep)
+      : "v"(src), "s"(lds_off)
+      : "memory");
+}
+
+DEVINL void pipe_barrier() {  // lgkm drain + raw barrier (glds may span)
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+}
+
+template <int N>
+DEVINL void pipe_barrier_vm() {  // + counted vmcnt: N glds may stay in flight
+  asm volatile("s_waitcnt vmcnt(%0) lgkmcnt(0)" :: "i"(N) : "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+}
+
+DEVINL void reg_fence(bf16x8_t& v) {  // force materialisation (wait) HERE
+  asm volatile("" : "+v"(v));
+}
+
+
+// ---------------------------------------------- pipelined MFMA flash prefill --
+// flash_prefill_bf16_kernel with the decode kernel's glds software pipeline
+// (see paged_decode_pipe_kernel below for the full rationale): the serial
+// stage→sync→S→softmax→PV structure exposes the K/V staging latency every
+// tile (L2-resident for the reuse across q-heads/q-tiles, but the VGPR
+// round-trip + barrier drain still serialize). Two buffers: V(kt)
+// overwrites K(kt)'s buffer after S, K(kt+1) prefetches into the other;
+// raw barriers with counted vmcnt keep the DMA in flight through PV.
+// The tile containing Lk falls back to plain zero-filled staging.
+
+template <int HEAD_DIM>
+__global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_pipe_kernel(
+    __hip_bfloat16* __restrict__ out,      // [Tq, H, D]
+    const __hip_bfloat16* __restrict__ q,  // [Tq, H, D]
+    const __hip_bfloat16* __restrict__ k,  // [Tk, KVH, D]
+    const __hip_bfloat16* __restrict__ v,
+    const int* __restrict__ cu_seqlens, const int* __restrict__ cu_seqlens_k,
+    int num_heads, int num_kv_heads,
+    float scale, float softcap, int window, long q_stride, long k_stride,
+    long v_stride, long o_stride) {
+  constexpr int D = HEAD_DIM;
+  constexpr int NT = 256;
+  constexpr int CPK = D / 8;                        // 16B granules per key row
+  constexpr int NI_K = PF_KT * CPK / NT;            // K glds per wave per tile
+  constexpr int NSUB = (PF_KT / 32) * (D / 16);     // V tr-subtiles per tile
+  constexpr int NI_V = NSUB / 4;                    // V glds per wave per tile
+  constexpr int KV_ELEMS = (PF_KT * D > NSUB * 528) ? PF_KT * D : NSUB * 528;
+  const int seq = blockIdx.x;
+  const int h = blockIdx.y;
+  const int kvh = h / (num_heads / num_kv_heads);
+  const int s0 = cu_seqlens[seq];
+  const int L = cu_seqlens[seq + 1] - s0;       // query rows
+  const int s0k = cu_seqlens_k[seq];
+  const int Lk = cu_seqlens_k[seq + 1] - s0k;   // key rows (>= L)
+  const int off = Lk - L;                       // abs position of q row 0
+  const int q_base = blockIdx.z * PF_QT;
+  if (q_base >= L) return;
+
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid & (WAVE - 1);
+  const int col = lane & 15;
+  const int kgrp = lane >> 4;
+
+  // ONE shared array (glds pipeline: hipcc drains vmcnt(0) before ds_reads
+  // when a second __shared__ object exists)
+  constexpr int P_OFF = 2 * KV_ELEMS;                    // shorts
+  // float index of the stat region: 16B-align past P (8 shorts per 16B
+  // unit, 4 floats per unit)
+  constexpr int STAT_F = (P_OFF + 4 * 16 * PF_KT + 7) / 8 * 4;
+  constexpr int TOTAL_BYTES = STAT_F * 4 + 4 * 2 * 16 * 4;
+  __shared__ __attribute__((aligned(16))) char smem[TOTAL_BYTES];
+  short* const kv0 = reinterpret_cast<short*>(smem);
+  short* const kv1 = kv0 + KV_ELEMS;
+  short* const p_base = kv0 + P_OFF;            // [4 waves][16][PF_KT]
+  float* const stat = reinterpret_cast<float*>(smem) + STAT_F;  // [4][2][16]
+
+  // ---- Q fragments
+  constexpr int KS = D / 32;
+  bf16x8_t qfrag[KS];
+  const int my_qrow = q_base + wid * 16 + col;
+  {
+    const int r = (my_qrow < L) ? my_qrow : (L - 1);
+    const __hip_bfloat16* qrow_p = q + (long)(s0 + r) * q_stride + (long)h * D;
+#pragma unroll
+    for (int ks = 0; ks < KS; ++ks) {
+      qfrag[ks] = *reinterpret_cast<const bf16x8_t*>(qrow_p + ks * 32 + kgrp * 8);
+      reg_fence(qfrag[ks]);  // wait HERE, not inside the glds pipeline
+    }
+  }
+
+  float m_run[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float l_run[4] = {0.f, 0.f, 0.f, 0.f};
+  constexpr int DT = D / 16;
+  f32x4_t ot[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt) ot[dt] = {0.f, 0.f, 0.f, 0.f};
+
+  const int wave_max_row = off + min(q_base + wid * 16 + 15, L - 1);
+  const int block_max_row = off + min(q_base + PF_QT - 1, L - 1);
+  const int kv_end = block_max_row + 1;
+  int kv_begin = 0;
+  if (window > 0) {
+    const int wave_min_needed = off + q_base + 1 - window;
+    kv_begin = max(0, (wave_min_needed / PF_KT) * PF_KT);
+  }
+
+  auto is_tail = [&](int kt) { return kt + PF_KT > Lk; };
+  auto issue_k = [&](int kt, short* dst) {
+#pragma unroll
+    for (int j = 0; j < NI_K; ++j) {
+      const int g = (wid * NI_K + j) * 64 + lane;  // granule
+      const int key = g / CPK;
+      const int r8 = (g % CPK) * 8;
+      const int d = r8 ^ ((key & 7) << 3);
+      const int o2 = __builtin_amdgcn_readfirstlane((wid * NI_K + j) * 512);
+      glds16(k + (long)(s0k + kt + key) * k_stride + (long)kvh * D + d, dst + o2);
+    }
+  };
+  auto issue_v = [&](int kt, short* dst) {
+#pragma unroll
+    for (int j = 0; j < NI_V; ++j) {
+      const int st = wid * NI_V + j;
+      const int ks = st / (D / 16), dtile = st % (D / 16);
+      const int pos = lane * 8;
+      const int bpos = pos / 64, rem = pos % 64;
+      const int qq = ((bpos >> 2) & 1) | ((bpos & 3) << 1);
+      const int key = ks * 32 + qq * 4 + rem / 16;
+      const int dim = dtile * 16 + (rem & 15);
+      const int o2 = __builtin_amdgcn_readfirstlane(st * 528);
+      glds16(v + (long)(s0k + kt + key) * v_stride + (long)kvh * D + dim, dst + o2);
+    }
+  };
+  auto stage_k_plain = [&](int kt, short* dst) {
+    for (int c = tid; c < PF_KT * CPK; c += NT) {
+      const int key = c / CPK;
+      const int d8 = (c % CPK) * 8;
+      const int pdst = key * D + swz(key, d8);
+      if (kt + key < Lk) {
+        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) =
+            *reinterpret_cast<const bf16x8_t*>(
+                k + (long)(s0k + kt + key) * k_stride + (long)kvh * D + d8);
+      } else {
+        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+    }
+  };
+  auto stage_v_plain = [&](int kt, short* dst) {
+    for (int c = tid; c < PF_KT * CPK; c += NT) {
+      const int key = c / CPK;
+      const int d8 = (c % CPK) * 8;
+      const int dtile = d8 / 16, col0 = d8 & 15;
+      const int qq = (key & 31) >> 2;
+      const int bpos = ((qq & 1) << 2) + (qq >> 1);
+      const int pdst = ((key >> 5) * (D / 16) + dtile) * 528 + bpos * 64 +
+                       (key & 3) * 16 + col0;
+      if (kt + key < Lk) {
+        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) =
+            *reinterpret_cast<const bf16x8_t*>(
+                v + (long)(s0k + kt + key) * v_stride + (long)kvh * D + d8);
+      } else {
+        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+    }
+  };
+
+  // ---- prologue
+  if (!is_tail(kv_begin)) issue_k(kv_begin, kv0);
+  int cur = 0;
+  for (int kt = kv_begin; kt < kv_end; kt += PF_KT, cur ^= 1) {
+    short* const X = cur ? kv1 : kv0;
+    short* const Y = cur ? kv0 : kv1;
+    const bool tail = is_tail(kt);
+    if (tail) {
+      stage_k_plain(kt, X);
+      pipe_barrier();
+    } else {
+      pipe_barrier_vm<0>();
+    }
+
+    const bool active = kt <= wave_max_row;  // wave-uniform
+    float p[4][4];
+    if (active) {
+      // ---- S = Q K^T for 4 column tiles of 16 keys
+      f32x4_t s[4];
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        s[ct] = {0.f, 0.f, 0.f, 0.f};
+        const int key = ct * 16 + col;
+#pragma unroll
+        for (int ks = 0; ks < KS; ++ks) {
+          const int d8 = ks * 32 + kgrp * 8;
+          bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
+              &X[key * D + swz(key, d8)]);
+          s[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks], bfrag, s[ct], 0, 0, 0);
+        }
+      }
+      // scale, softcap, mask, row max
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int qrow = q_base + wid * 16 + kgrp * 4 + reg;
+        const int row = off + qrow;
+        float mx = -1e30f;
+#pragma unroll
+        for (int ct = 0; ct < 4; ++ct) {
+          const int key = kt + ct * 16 + col;
+          float x = s[ct][reg] * scale;
+          if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
+          const bool dead = key > row || key >= Lk || qrow >= L ||
+                            (window > 0 && key <= row - window);
+          x = dead ? -1e30f : x;
+          p[reg][ct] = x;
+          mx = fmaxf(mx, x);
+        }
+#pragma unroll
+        for (int o3 = 1; o3 < 16; o3 <<= 1) mx = fmaxf(mx, __shfl_xor(mx, o3, WAVE));
+        const float m_new = fmaxf(m_run[reg], mx);
+        const float alpha = (m_new > -1e30f) ? __expf(m_run[reg] - m_new) : 1.f;
+        float lsum = 0.f;
+#pragma unroll
+        for (int ct = 0; ct < 4; ++ct) {
+          const float pe = (m_new > -1e30f && p[reg][ct] > -1e29f)
+                               ? __expf(p[reg][ct] - m_new) : 0.f;
+          p[reg][ct] = pe;
+          lsum += pe;
+        }
+#pragma unroll
+        for (int o3 = 1; o3 < 16; o3 <<= 1) lsum += __shfl_xor(lsum, o3, WAVE);
+        l_run[reg] = l_run[reg] * alpha + lsum;
+        m_run[reg] = m_new;
+        if (col == 0) stat[(wid * 2 + 0) * 16 + kgrp * 4 + reg] = alpha;
+      }
+    }
+    pipe_barrier();  // S reads of X done; alpha visible
+
+    // ---- issue V(kt)->X, K(kt+PF_KT)->Y under the P build + PV
+    const int nxt = kt + PF_KT;
+    const bool prefetch = nxt < kv_end && !is_tail(nxt);
+    if (tail) {
+      stage_v_plain(kt, X);
+    } else {
+      issue_v(kt, X);
+    }
+    if (prefetch) issue_k(nxt, Y);
+
+    if (active) {
+      // write P rows (bf16) to this wave's tile
+      short* const p_lds_w = p_base + wid * 16 * PF_KT;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+#pragma unroll
+        for (int ct = 0; ct < 4; ++ct) {
+          p_lds_w[(kgrp * 4 + reg) * PF_KT + ct * 16 + col] =
+              __bfloat16_as_short(__float2bfloat16(p[reg][ct]));
+        }
+      }
+    }
+    if (tail) {
+      pipe_barrier();
+    } else if (prefetch) {
+      pipe_barrier_vm<NI_K>();
+    } else {
+      pipe_barrier_vm<0>();
+    }
+
+    if (active) {
+      // ---- OT += V^T P^T
+      const float alpha_q = stat[(wid * 2 + 0) * 16 + col];
+      short* const p_lds_w = p_base + wid * 16 * PF_KT;
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
+        ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
+#pragma unroll
+        for (int ks = 0; ks < PF_KT / 32; ++ks) {
+          const int sub = (ks * (D / 16) + dt) * 528 + lane * 4;
+          bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (lds_bf16x4*)&X[sub]);
+          bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (lds_bf16x4*)&X[sub + 4 * 64]);
+          bf16x8_t a;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            a[j] = __bfloat16_as_short((__hip_bfloat16)lo[j]);
+            a[4 + j] = __bfloat16_as_short((__hip_bfloat16)hi[j]);
+          }
+          bf16x8_t bb = *reinterpret_cast<const bf16x8_t*>(
+              &p_lds_w[col * PF_KT + ks * 32 + kgrp * 8]);
+          ot[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bb, ot[dt], 0, 0, 0);
+        }
+      }
+    }
+    // next loop-top barrier protects X (overwritten two iterations out)
+  }
+
+  // ---- epilogue: normalise and store
+  if (col == 0) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const float l = l_run[reg];
+      stat[(wid * 2 + 1) * 16 + kgrp * 4 + reg] = (l > 0.f) ? 1.0f / l : 0.f;
+    }
+  }
+  pipe_barrier();
+  const float inv = stat[(wid * 2 + 1) * 16 + col];
+  const int orow = q_base + wid * 16 + col;
+  if (orow < L) {
+    __hip_bfloat16* op = out + (long)(s0 + orow) * o_stride + (long)h * D;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg)
+        op[dt * 16 + kgrp * 4 + reg] = __float2bfloat16(ot[dt][reg] * inv);
+    }
+  }
+}
+
 // ------------------------------------------------------- MFMA paged decode --
 // Single-token GQA decode attention on matrix cores (bf16, D in {128,256},
 // G <= 16). One workgroup per (sequence, kv_head); the GQA query group is
@@ -868,47 +1207,6 @@ __global__ __launch_bounds__(NW * WAVE) void paged_decode_mfma_kernel(
 //     zero-fill; garbage V rows would reach the PV MFMA as NaN×0).
 // Per-workgroup timeline: the only window with no HBM traffic in flight is
 // the S phase (~hundreds of cycles vs ~6 µs of stream per chunk).
-
-// 16-B-per-lane global→LDS DMA, HIDDEN from hipcc in inline asm: with the
-// builtin form the compiler conservatively re-inserts `s_waitcnt vmcnt(0)`
-// before the first ds_read that may alias the DMA destination (verified in
-// the .s: it landed ahead of the PV tr16 reads, draining the K prefetch
-// every chunk). Hand-counted vmcnt in pipe_barrier_vm is the only wait
-// discipline for these, per guide §5.7 (its LDS-DMA recipe: M0 carries the
-// wave-uniform LDS byte address, saved/written/restored in ONE statement;
-// the s_nop is the SALU-write→M0-read wait state).
-DEVINL void glds16(const void* src, void* lds_dst_uniform) {
-  unsigned lds_off = (unsigned)(unsigned long)(
-      (__attribute__((address_space(3))) char*)lds_dst_uniform);
-  lds_off = __builtin_amdgcn_readfirstlane(lds_off);
-  unsigned keep;
-  asm volatile(
-      "s_mov_b32 %0, m0\n\t"
-      "s_mov_b32 m0, %2\n\t"
-      "s_nop 0\n\t"
-      "global_load_lds_dwordx4 %1, off\n\t"
-      "s_mov_b32 m0, %0"
-      : "=&s"(keep)
-      : "v"(src), "s"(lds_off)
-      : "memory");
-}
-
-DEVINL void pipe_barrier() {  // lgkm drain + raw barrier (glds may span)
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  __builtin_amdgcn_s_barrier();
-  asm volatile("" ::: "memory");
-}
-
-template <int N>
-DEVINL void pipe_barrier_vm() {  // + counted vmcnt: N glds may stay in flight
-  asm volatile("s_waitcnt vmcnt(%0) lgkmcnt(0)" :: "i"(N) : "memory");
-  __builtin_amdgcn_s_barrier();
-  asm volatile("" ::: "memory");
-}
-
-DEVINL void reg_fence(bf16x8_t& v) {  // force materialisation (wait) HERE
-  asm volatile("" : "+v"(v));
-}
 
 // min waves/SIMD: the 64-key chunk fits 2 workgroups/CU by LDS (4 waves/
 // SIMD) — force the register allocator to <=128 VGPRs so VGPRs don't cap

@@ -376,10 +376,26 @@ void launch_prefill(at::Tensor& out, const at::Tensor& q, const at::Tensor& k,
   const int H = q.size(1), D = q.size(2);
   const int KVH = k.size(1);
   if constexpr (std::is_same_v<T, __hip_bfloat16>) {
-    // MFMA flash path (bf16): grid (seq, head, q-tile)
+    // MFMA flash path (bf16): grid (seq, head, q-tile). Default: the glds
+    // software-pipelined kernel; LLMQ_PREFILL_PIPE=0 selects the plain-
+    // staged one (A/B). Read per call for in-process microbenches.
     const int qtiles = (int)((max_seqlen + 63) / 64);
     dim3 fgrid(B, H, std::max(qtiles, 1));
+    const char* ppe = getenv("LLMQ_PREFILL_PIPE");
+    const bool use_pipe = !ppe || atoi(ppe) != 0;
     auto lf = [&]<int HD>() {
+      if (use_pipe) {
+        hipLaunchKernelGGL((flash_prefill_pipe_kernel<HD>), fgrid, dim3(256), 0,
+                           stream(),
+                           reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                           reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                           reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
+                           reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+                           cu.data_ptr<int>(), cu_k.data_ptr<int>(), H, KVH,
+                           (float)scale, (float)softcap, (int)window,
+                           q.stride(0), k.stride(0), v.stride(0), out.stride(0));
+        return;
+      }
       hipLaunchKernelGGL((flash_prefill_bf16_kernel<HD>), fgrid, dim3(256), 0,
                          stream(),
                          reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
