@@ -106,9 +106,10 @@ def main() -> None:
                 completed += 1
     elapsed = time.perf_counter() - t0
 
-    fuse_env = os.environ.get("LLMQ_MIXED_FUSE_MIN_TOKENS", "128 (default)")
+    fuse_env = os.environ.get("LLMQ_MIXED_FUSE_MIN_TOKENS", "split (default)")
+    overlap_env = os.environ.get("LLMQ_OVERLAP_MIXED", "1 (default)")
     report = {
-        "mode": f"LLMQ_MIXED_FUSE_MIN_TOKENS={fuse_env}",
+        "mode": f"fuse={fuse_env} overlap={overlap_env}",
         "model": model, "batch": batch, "prompt_len": plen, "output_len": olen,
         "seconds": round(elapsed, 2),
         "output_tok_per_s": round(out_tokens / elapsed, 1),
