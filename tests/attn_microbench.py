@@ -184,26 +184,37 @@ def prefill_bench(args) -> None:
     def run():
         return ops.varlen_prefill_attention(q, k, v, cu, L, scale)
 
-    for _ in range(args.warmup):
-        run()
-    if use_gpu:
-        torch.cuda.synchronize()
-    t0 = time.perf_counter()
-    for _ in range(args.iters):
-        run()
-    if use_gpu:
-        torch.cuda.synchronize()
-    dt = (time.perf_counter() - t0) / args.iters
     flops = 4.0 * B * H * D * (L * (L + 1) / 2)  # causal QK^T + PV
     print(f"prefill B={B} L={L} H={H} KVH={KVH} D={D}")
-    print(f"{dt * 1e3:.3f} ms/iter   {flops / dt / 1e12:.1f} TF/s")
     if args.check:
         from llmq_amd.ops import torch_ref
 
         ref = torch_ref.varlen_prefill_attention(
             q.float().cpu(), k.float().cpu(), v.float().cpu(), cu.cpu(), scale)
-        err = (run().float().cpu() - ref).abs().max().item()
-        print(f"max|err|: {err:.4e}")
+        for variant in ("0", "1"):
+            os.environ["LLMQ_PREFILL_PIPE"] = variant
+            err = (run().float().cpu() - ref).abs().max().item()
+            print(f"pipe={variant}: max|err|: {err:.4e}")
+            assert err < 0.05
+    variants = ["0", "1"] if args.ab else [os.environ.get("LLMQ_PREFILL_PIPE", "1")]
+    rounds = 5 if args.ab else 1
+    times = {vv: [] for vv in variants}
+    for _ in range(rounds):
+        for vv in variants:
+            os.environ["LLMQ_PREFILL_PIPE"] = vv
+            for _ in range(args.warmup):
+                run()
+            if use_gpu:
+                torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(args.iters):
+                run()
+            if use_gpu:
+                torch.cuda.synchronize()
+            times[vv].append((time.perf_counter() - t0) / args.iters)
+    for vv in variants:
+        dt = sorted(times[vv])[len(times[vv]) // 2]
+        print(f"pipe={vv}: {dt * 1e3:.3f} ms/iter   {flops / dt / 1e12:.1f} TF/s")
 
 
 if __name__ == "__main__":
