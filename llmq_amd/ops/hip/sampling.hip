@@ -163,36 +163,41 @@ __global__ __launch_bounds__(TH_TPB) void topk_topp_bound_kernel(
   __syncthreads();
   m = red[0];
 
+  // Histogram window [xlo, xhi) in x-space; elements >= keep_above are
+  // already-kept (counted in p_above), elements < xlo already-dropped.
+  // Boundary discipline matters: an element EXACTLY at the crossing bin's
+  // lower edge belongs to the refinement window (a `<=` exclusion here
+  // silently lost it and the refinement found no crossing).
   float xlo = TH_XMIN, xhi = 0.0f;
-  float p_above = 0.f;   // exp-mass strictly above xhi (already kept)
+  float keep_above = 3e38f;
+  float p_above = 0.f;
   long n_above = 0;
   float Z = 0.f;         // total exp-mass (fixed after iter 0)
-  bool keep_all = false;
+  float bound_x = TH_XMIN;
+  bool done = false;
 
-  for (int iter = 0; iter < 2; ++iter) {
+  for (int iter = 0; iter < 2 && !done; ++iter) {
     for (int i = tid; i < TH_NBINS; i += TH_TPB) { h_sum[i] = 0.f; h_cnt[i] = 0; }
     __syncthreads();
     const float w = (xhi - xlo) / TH_NBINS;
     const float inv_w = 1.0f / w;
-    float tail = 0.f;  // mass at or below xlo (iter 0 only; never kept)
+    float tail = 0.f;  // mass below xlo (iter 0 only; used for Z)
     for (int v = tid; v < V; v += TH_TPB) {
       const float x = (row[v] - m) * inv_t;
-      if (x > xhi) continue;           // already kept
-      if (x <= xlo) { tail += __expf(fmaxf(x, -80.f)); continue; }
+      if (x >= keep_above) continue;   // already kept (counted in p_above)
+      if (x < xlo) { if (iter == 0) tail += __expf(fmaxf(x, -80.f)); continue; }
       int bin = (int)((x - xlo) * inv_w);
       bin = bin < 0 ? 0 : (bin >= TH_NBINS ? TH_NBINS - 1 : bin);
       atomicAdd(&h_sum[bin], __expf(x));
       atomicAdd(&h_cnt[bin], 1);
     }
-    // reduce tail mass (only meaningful on iter 0 for Z)
+    // reduce tail mass (iter 0: completes Z)
     for (int off = WAVE / 2; off > 0; off >>= 1) tail += __shfl_xor(tail, off, WAVE);
     if ((tid & (WAVE - 1)) == 0) red[tid / WAVE] = tail;
     __syncthreads();
     if (tid == 0) {
-      float tail_tot = 0.f;
-      for (int w2 = 0; w2 < TH_TPB / WAVE; ++w2) tail_tot += red[w2];
       if (iter == 0) {
-        Z = tail_tot;
+        Z = red[0] + red[1] + red[2] + red[3];
         for (int i = 0; i < TH_NBINS; ++i) Z += h_sum[i];
         sh_state[3] = Z;
       } else {
@@ -209,30 +214,38 @@ __global__ __launch_bounds__(TH_TPB) void topk_topp_bound_kernel(
         if (cum >= p_target || cnt >= k_target) { cross = i; break; }
       }
       if (cross < 0) {
-        // never crossed within [xlo, xhi]: every candidate is kept; the
-        // sub-xlo tail is beyond fp32 softmax relevance -> keep all
-        sh_state[0] = -3e38f;
-        sh_state[1] = -3e38f;
+        // No crossing inside the window. Iter 0: the target needs mass
+        // below XMIN too -> keep everything. Refinements: cannot happen
+        // with exact arithmetic (iter 0 placed the crossing here); under
+        // fp drift keep the WHOLE window (conservative over-keep).
+        sh_state[0] = (iter == 0) ? -3e38f : xlo;
+        sh_state[1] = 1.0f;  // done marker
       } else {
         float pa = p_above;
         long na = n_above;
         for (int i = TH_NBINS - 1; i > cross; --i) { pa += h_sum[i]; na += h_cnt[i]; }
         sh_state[0] = xlo + cross * w;        // new xlo (crossing bin lo)
-        sh_state[1] = xlo + (cross + 1) * w;  // new xhi
+        sh_state[1] = 0.0f;                   // continue refining
+        sh_state[2] = xlo + (cross + 1) * w;  // new xhi / keep_above
         red[0] = pa;
         red[1] = (float)na;  // counts <= V = 262k << 2^24: exact in fp32
       }
     }
     __syncthreads();
-    if (sh_state[0] == -3e38f && sh_state[1] == -3e38f) { keep_all = true; break; }
-    xlo = sh_state[0];
-    xhi = sh_state[1];
-    p_above = red[0];
-    n_above = (long)red[1];
+    if (sh_state[1] != 0.0f) {
+      bound_x = sh_state[0];
+      done = true;
+    } else {
+      xlo = sh_state[0];
+      keep_above = xhi = sh_state[2];
+      p_above = red[0];
+      n_above = (long)red[1];
+      bound_x = xlo;  // best-so-far: the crossing bin's lower edge
+    }
     __syncthreads();
   }
   if (tid == 0)
-    out_bound[b] = keep_all ? -3e38f : (m + xlo * T0);
+    out_bound[b] = (bound_x <= -3e37f) ? -3e38f : (m + bound_x * T0);
 }
 
 __global__ void unpack_keys_kernel(long* __restrict__ out,
