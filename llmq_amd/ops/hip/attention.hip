@@ -665,7 +665,9 @@ __global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_
     kv_begin = max(0, (wave_min_needed / PF_KT) * PF_KT);
   }
 
-  auto is_tail = [&](int kt) { return kt + PF_KT > Lk; };
+  // Out-of-range keys clamp to row Lk-1 (real finite data): K garbage is
+  // masked in S, dead keys carry P == 0 into PV — every tile stays on the
+  // glds pipeline (see the decode kernel's note).
   auto issue_k = [&](int kt, short* dst) {
 #pragma unroll
     for (int j = 0; j < NI_K; ++j) {
@@ -673,8 +675,9 @@ __global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_
       const int key = g / CPK;
       const int r8 = (g % CPK) * 8;
       const int d = r8 ^ ((key & 7) << 3);
+      const int gk = min(kt + key, Lk - 1);
       const int o2 = __builtin_amdgcn_readfirstlane((wid * NI_K + j) * 512);
-      glds16(k + (long)(s0k + kt + key) * k_stride + (long)kvh * D + d, dst + o2);
+      glds16(k + (long)(s0k + gk) * k_stride + (long)kvh * D + d, dst + o2);
     }
   };
   auto issue_v = [&](int kt, short* dst) {
@@ -687,56 +690,18 @@ __global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_
       const int qq = ((bpos >> 2) & 1) | ((bpos & 3) << 1);
       const int key = ks * 32 + qq * 4 + rem / 16;
       const int dim = dtile * 16 + (rem & 15);
+      const int gk = min(kt + key, Lk - 1);
       const int o2 = __builtin_amdgcn_readfirstlane(st * 528);
-      glds16(v + (long)(s0k + kt + key) * v_stride + (long)kvh * D + dim, dst + o2);
+      glds16(v + (long)(s0k + gk) * v_stride + (long)kvh * D + dim, dst + o2);
     }
   };
-  auto stage_k_plain = [&](int kt, short* dst) {
-    for (int c = tid; c < PF_KT * CPK; c += NT) {
-      const int key = c / CPK;
-      const int d8 = (c % CPK) * 8;
-      const int pdst = key * D + swz(key, d8);
-      if (kt + key < Lk) {
-        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) =
-            *reinterpret_cast<const bf16x8_t*>(
-                k + (long)(s0k + kt + key) * k_stride + (long)kvh * D + d8);
-      } else {
-        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
-      }
-    }
-  };
-  auto stage_v_plain = [&](int kt, short* dst) {
-    for (int c = tid; c < PF_KT * CPK; c += NT) {
-      const int key = c / CPK;
-      const int d8 = (c % CPK) * 8;
-      const int dtile = d8 / 16, col0 = d8 & 15;
-      const int qq = (key & 31) >> 2;
-      const int bpos = ((qq & 1) << 2) + (qq >> 1);
-      const int pdst = ((key >> 5) * (D / 16) + dtile) * 528 + bpos * 64 +
-                       (key & 3) * 16 + col0;
-      if (kt + key < Lk) {
-        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) =
-            *reinterpret_cast<const bf16x8_t*>(
-                v + (long)(s0k + kt + key) * v_stride + (long)kvh * D + d8);
-      } else {
-        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
-      }
-    }
-  };
-
   // ---- prologue
-  if (!is_tail(kv_begin)) issue_k(kv_begin, kv0);
+  if (kv_begin < kv_end) issue_k(kv_begin, kv0);
   int cur = 0;
   for (int kt = kv_begin; kt < kv_end; kt += PF_KT, cur ^= 1) {
     short* const X = cur ? kv1 : kv0;
     short* const Y = cur ? kv0 : kv1;
-    const bool tail = is_tail(kt);
-    if (tail) {
-      stage_k_plain(kt, X);
-      pipe_barrier();
-    } else {
-      pipe_barrier_vm<0>();
-    }
+    pipe_barrier_vm<0>();
 
     const bool active = kt <= wave_max_row;  // wave-uniform
     float p[4][4];
@@ -795,12 +760,8 @@ __global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_
 
     // ---- issue V(kt)->X, K(kt+PF_KT)->Y under the P build + PV
     const int nxt = kt + PF_KT;
-    const bool prefetch = nxt < kv_end && !is_tail(nxt);
-    if (tail) {
-      stage_v_plain(kt, X);
-    } else {
-      issue_v(kt, X);
-    }
+    const bool prefetch = nxt < kv_end;
+    issue_v(kt, X);
     if (prefetch) issue_k(nxt, Y);
 
     if (active) {
@@ -815,9 +776,7 @@ __global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_
         }
       }
     }
-    if (tail) {
-      pipe_barrier();
-    } else if (prefetch) {
+    if (prefetch) {
       pipe_barrier_vm<NI_K>();
     } else {
       pipe_barrier_vm<0>();
@@ -1313,11 +1272,16 @@ __global__ __launch_bounds__(NW * WAVE, (PD_KT <= 64 ? 4 : 2)) void paged_decode
 
   // Per-lane source row+offset for K granule g of a chunk (inverse swizzle
   // on the source: LDS image is lane-linear, content matches swz()).
+  // Out-of-range keys (tail chunk / z-range end) clamp to row L-1: a real
+  // written cache row, so the staged garbage is FINITE; K garbage is
+  // masked in S (-1e30) and dead keys carry P == 0, so finite V garbage
+  // contributes exactly 0 to PV. This keeps EVERY chunk on the glds
+  // pipeline (no zero-filled plain-staged tail).
   auto k_src = [&](int base, int g) -> const __hip_bfloat16* {
     const int key = g / CPK;
     const int r8 = (g % CPK) * 8;
     const int d = r8 ^ ((key & 7) << 3);
-    const int gkey = base + key;
+    const int gkey = min(base + key, L - 1);
     const long blk = bt_l[gkey / block_size];
     return k_cache + (blk * num_kv_heads + kh) * ((long)block_size * D) +
            (long)(gkey % block_size) * D + d;
@@ -1330,7 +1294,7 @@ __global__ __launch_bounds__(NW * WAVE, (PD_KT <= 64 ? 4 : 2)) void paged_decode
     const int qq = ((bpos >> 2) & 1) | ((bpos & 3) << 1);
     const int key = ks * 32 + qq * 4 + rem / 16;
     const int dim = dtile * 16 + (rem & 15);
-    const int gkey = base + key;
+    const int gkey = min(base + key, L - 1);
     const long blk = bt_l[gkey / block_size];
     return v_cache + (blk * num_kv_heads + kh) * ((long)block_size * D) +
            (long)(gkey % block_size) * D + dim;
@@ -1352,66 +1316,15 @@ __global__ __launch_bounds__(NW * WAVE, (PD_KT <= 64 ? 4 : 2)) void paged_decode
       glds16(v_src(base, st), dst + off);
     }
   };
-  // Tail chunk (contains L): plain staged writes with zero fill.
-  auto stage_k_plain = [&](int base, short* dst) {
-    for (int c = tid; c < PD_KT * CPK; c += NT) {
-      const int key = c / CPK;
-      const int d8 = (c % CPK) * 8;
-      const int pdst = key * D + swz(key, d8);
-      const int gkey = base + key;
-      if (gkey < L) {
-        const long blk = bt_l[gkey / block_size];
-        const long rowoff =
-            (blk * num_kv_heads + kh) * ((long)block_size * D) +
-            (long)(gkey % block_size) * D + d8;
-        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) =
-            *reinterpret_cast<const bf16x8_t*>(k_cache + rowoff);
-      } else {
-        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
-      }
-    }
-  };
-  auto stage_v_plain = [&](int base, short* dst) {
-    for (int c = tid; c < PD_KT * CPK; c += NT) {
-      const int key = c / CPK;
-      const int d8 = (c % CPK) * 8;
-      const int dtile = d8 / 16, col0 = d8 & 15;
-      const int qq = (key & 31) >> 2;
-      const int bpos = ((qq & 1) << 2) + (qq >> 1);
-      const int pdst = ((key >> 5) * (D / 16) + dtile) * 528 + bpos * 64 +
-                       (key & 3) * 16 + col0;
-      const int gkey = base + key;
-      if (gkey < L) {
-        const long blk = bt_l[gkey / block_size];
-        const long rowoff =
-            (blk * num_kv_heads + kh) * ((long)block_size * D) +
-            (long)(gkey % block_size) * D + d8;
-        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) =
-            *reinterpret_cast<const bf16x8_t*>(v_cache + rowoff);
-      } else {
-        *reinterpret_cast<bf16x8_t*>(&dst[pdst]) = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
-      }
-    }
-  };
-
-  auto is_tail = [&](int base) { return base + PD_KT > L; };
-
   // ---- prologue: block table visible, then K(first) in flight
   pipe_barrier();  // bt_l ready (plain ds_writes; lgkm drain)
-  const bool have_range = range_lo < range_hi;
-  if (have_range && !is_tail(range_lo)) issue_k(range_lo, kv0);
+  if (range_lo < range_hi) issue_k(range_lo, kv0);
 
   int cur = 0;
   for (int base = range_lo; base < range_hi; base += PD_KT, cur ^= 1) {
     short* const X = cur ? kv1 : kv0;
     short* const Y = cur ? kv0 : kv1;
-    const bool tail = is_tail(base);
-    if (tail) {
-      stage_k_plain(base, X);
-      pipe_barrier();
-    } else {
-      pipe_barrier_vm<0>();  // K(base) landed (all waves)
-    }
+    pipe_barrier_vm<0>();  // K(base) landed (all waves)
 
     // ---- S[16,16] for this wave's slab (waves wid>=SLABS duplicate)
     f32x4_t s = {0.f, 0.f, 0.f, 0.f};
@@ -1444,12 +1357,8 @@ __global__ __launch_bounds__(NW * WAVE, (PD_KT <= 64 ? 4 : 2)) void paged_decode
 
     // ---- issue V(base)->X then K(next)->Y; both stream under softmax/PV
     const int next = base + PD_KT;
-    const bool prefetch = next < range_hi && !is_tail(next);
-    if (tail) {
-      stage_v_plain(base, X);
-    } else {
-      issue_v(base, X);
-    }
+    const bool prefetch = next < range_hi;
+    issue_v(base, X);
     if (prefetch) issue_k(next, Y);
 
     // ---- combine maxes, build P, update l (redundant on every wave)
@@ -1472,9 +1381,7 @@ __global__ __launch_bounds__(NW * WAVE, (PD_KT <= 64 ? 4 : 2)) void paged_decode
       p_lds2[row * PD_KT + slab * 16 + col] =
           __bfloat16_as_short(__float2bfloat16(pe));
     }
-    if (tail) {
-      pipe_barrier();  // V (plain) + P + alpha ready; nothing in flight
-    } else if (prefetch) {
+    if (prefetch) {
       pipe_barrier_vm<NI_K>();  // V landed; K(next) stays in flight
     } else {
       pipe_barrier_vm<0>();

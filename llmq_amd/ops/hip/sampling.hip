@@ -147,7 +147,7 @@ __global__ __launch_bounds__(TH_TPB) void topk_topp_bound_kernel(
   __shared__ float h_sum[TH_NBINS];
   __shared__ int h_cnt[TH_NBINS];
   __shared__ float red[TH_TPB / WAVE];
-  __shared__ float sh_state[4];  // xlo, xhi, p_above(normed later), Z
+  __shared__ float sh_state[6];  // new-xlo, flag, new-xhi, Z, keep_above
 
   // ---- pass 1: row max
   float m = -3e38f;
@@ -226,7 +226,14 @@ __global__ __launch_bounds__(TH_TPB) void topk_topp_bound_kernel(
         for (int i = TH_NBINS - 1; i > cross; --i) { pa += h_sum[i]; na += h_cnt[i]; }
         sh_state[0] = xlo + cross * w;        // new xlo (crossing bin lo)
         sh_state[1] = 0.0f;                   // continue refining
-        sh_state[2] = xlo + (cross + 1) * w;  // new xhi / keep_above
+        sh_state[2] = xlo + (cross + 1) * w;  // new xhi (bin geometry)
+        // keep_above: elements EXACTLY at a bin edge were binned below it
+        // (clamp puts the window top INSIDE the top bin) — so when the
+        // crossing IS the top bin, its upper boundary stays the previous
+        // exclusive bound, or the row max itself would be excluded from
+        // the refinement and the bound would land a whole bin low.
+        sh_state[4] = (cross == TH_NBINS - 1) ? keep_above
+                                              : (xlo + (cross + 1) * w);
         red[0] = pa;
         red[1] = (float)na;  // counts <= V = 262k << 2^24: exact in fp32
       }
@@ -237,7 +244,8 @@ __global__ __launch_bounds__(TH_TPB) void topk_topp_bound_kernel(
       done = true;
     } else {
       xlo = sh_state[0];
-      keep_above = xhi = sh_state[2];
+      xhi = sh_state[2];
+      keep_above = sh_state[4];
       p_above = red[0];
       n_above = (long)red[1];
       bound_x = xlo;  // best-so-far: the crossing bin's lower edge
