@@ -19,12 +19,14 @@ def build_tiny_checkpoint(
     path,
     family: str = "llama",
     vocab_size: int = 512,
-    hidden: int = 128,
+    hidden: int = 512,
     layers: int = 2,
     heads: int = 8,
     kv_heads: int = 4,
     seed: int = 0,
 ) -> str:
+    # head_dim = hidden/heads must be a GPU-kernel-supported shape
+    # (64/128/256) — the default 512/8 gives 64.
     """Write a tiny random-weight checkpoint + real tokenizer; returns path."""
     import torch
     import transformers
