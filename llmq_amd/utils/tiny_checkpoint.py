@@ -25,9 +25,10 @@ def build_tiny_checkpoint(
     kv_heads: int = 4,
     seed: int = 0,
 ) -> str:
-    # head_dim = hidden/heads must be a GPU-kernel-supported shape
-    # (64/128/256) — the default 512/8 gives 64.
-    """Write a tiny random-weight checkpoint + real tokenizer; returns path."""
+    """Write a tiny random-weight checkpoint + real tokenizer; returns path.
+
+    head_dim = hidden/heads must be a GPU-kernel-supported shape
+    (64/128/256) — the default 512/8 gives 64."""
     import torch
     import transformers
     from tokenizers import Tokenizer
