@@ -218,6 +218,34 @@ def varlen_prefill_attention(
     )
 
 
+# ---------------------------------------------------------- skinny GEMM --
+
+
+def skinny_gemm(
+    x: torch.Tensor,          # [M, K] bf16
+    w: torch.Tensor,          # [N, K] bf16 (torch linear weight layout)
+    bias: Optional[torch.Tensor] = None,
+    splitk: int = 0,          # 0 = auto
+    out: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Hand-written skinny-M decode-projection GEMM (out = x @ w.T + bias).
+
+    256x256x64 MFMA tile, glds-staged with source-side st_16x32 swizzle,
+    counted-vmcnt pipeline; split-K slabs + fused reduce for narrow-N
+    shapes. Falls back to F.linear off-GPU / unsupported shapes."""
+    M, K = x.shape
+    N = w.shape[0]
+    if not (x.is_cuda and x.dtype == torch.bfloat16 and K % 64 == 0
+            and N % 4 == 0 and has_hip_ext()):
+        import torch.nn.functional as F
+
+        return F.linear(x, w, bias)
+    if out is None:
+        out = torch.empty(M, N, dtype=torch.bfloat16, device=x.device)
+    _EXT.skinny_gemm(out, x, w, bias, splitk)
+    return out
+
+
 # ------------------------------------------------------------- sampling --
 
 

@@ -449,6 +449,54 @@ void varlen_prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,
   });
 }
 
+// ---------------------------------------------------------- skinny GEMM --
+
+void skinny_gemm(at::Tensor out, at::Tensor x, at::Tensor w,
+                 c10::optional<at::Tensor> bias, long splitk) {
+  CHECK_GPU(x);
+  CHECK_LASTDIM(x);
+  CHECK_LASTDIM(w);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 && w.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(out.scalar_type() == at::kBFloat16 && out.is_contiguous());
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N);
+  TORCH_CHECK(K % SKG_BK == 0, "K must be a multiple of 64");
+  TORCH_CHECK(N % 4 == 0, "N must be a multiple of 4");
+  const __hip_bfloat16* bias_ptr = nullptr;
+  if (bias.has_value() && bias->defined()) {
+    TORCH_CHECK(bias->scalar_type() == at::kBFloat16 && bias->numel() == N);
+    bias_ptr = reinterpret_cast<const __hip_bfloat16*>(bias->data_ptr());
+  }
+  const int mt = (M + SKG_BM - 1) / SKG_BM;
+  const int nt = (N + SKG_BN - 1) / SKG_BN;
+  int z = (int)splitk;
+  if (z <= 0) {  // auto: fill the chip (>=256 blocks), keep >=8 K-steps/slice
+    z = 1;
+    while (mt * nt * z < 256 && (K / SKG_BK) / (z * 2) >= 8 && z < 8) z *= 2;
+  }
+  dim3 grid(mt, nt, z);
+  if (z == 1) {
+    hipLaunchKernelGGL((skinny_gemm_kernel<0>), grid, dim3(SKG_NT), 0, stream(),
+                       out.data_ptr(),
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                       bias_ptr, M, N, K, x.stride(0), w.stride(0));
+    return;
+  }
+  at::Tensor slabs = at::empty({z, (long)M * N}, x.options().dtype(at::kFloat));
+  hipLaunchKernelGGL((skinny_gemm_kernel<1>), grid, dim3(SKG_NT), 0, stream(),
+                     slabs.data_ptr(),
+                     reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                     nullptr, M, N, K, x.stride(0), w.stride(0));
+  const long MN = (long)M * N;
+  const long blocks = (MN / 4 + 255) / 256;
+  hipLaunchKernelGGL(skinny_gemm_reduce_kernel, dim3(blocks), dim3(256), 0,
+                     stream(),
+                     reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                     slabs.data_ptr<float>(), bias_ptr, z, MN, N);
+}
+
 // ------------------------------------------------------------- sampling --
 
 void topk_topp_bound(at::Tensor out_bound, at::Tensor logits, at::Tensor temps,
@@ -510,6 +558,7 @@ TORCH_LIBRARY(llmq_amd, m) {
   m.def("varlen_prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor cu_seqlens, Tensor cu_seqlens_k, int max_seqlen, float scale, float softcap, int window) -> ()");
   m.def("sample_gumbel_argmax(Tensor(a!) out, Tensor(b!) keys, Tensor logits, Tensor temps, Tensor req_seeds, Tensor req_pos, int seed, int step, Tensor? bounds=None) -> ()");
   m.def("topk_topp_bound(Tensor(a!) out_bound, Tensor logits, Tensor temps, Tensor top_ps, Tensor top_ks) -> ()");
+  m.def("skinny_gemm(Tensor(a!) out, Tensor x, Tensor w, Tensor? bias, int splitk) -> ()");
   m.def("norm_add_norm(Tensor(a!) x, Tensor(b!) residual, Tensor w_post, Tensor w_pre, float eps, float offset) -> ()");
   m.def("rope_and_cache(Tensor(a!) q, Tensor(b!) k, Tensor value, Tensor(c!) k_cache, Tensor(d!) v_cache, Tensor positions, Tensor cos_sin, Tensor slot_mapping) -> ()");
 }
@@ -525,6 +574,7 @@ TORCH_LIBRARY_IMPL(llmq_amd, CUDA, m) {
   m.impl("varlen_prefill_attention", &varlen_prefill_attention);
   m.impl("sample_gumbel_argmax", &sample_gumbel_argmax);
   m.impl("topk_topp_bound", &topk_topp_bound);
+  m.impl("skinny_gemm", &skinny_gemm);
   m.impl("norm_add_norm", &norm_add_norm);
   m.impl("rope_and_cache", &rope_and_cache);
 }

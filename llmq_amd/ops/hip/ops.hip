@@ -3,5 +3,6 @@
 #include "elementwise.hip"
 #include "kvcache.hip"
 #include "attention.hip"
+#include "skinny_gemm.hip"
 #include "sampling.hip"
 #include "bindings.cpp"
