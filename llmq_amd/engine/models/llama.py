@@ -19,6 +19,7 @@ the model set the reference serves through vLLM). Design notes, MI355X-first:
 from __future__ import annotations
 
 import math
+import os
 from typing import Dict, List, Optional, Union
 
 import torch
@@ -29,6 +30,20 @@ from llmq_amd.engine.forward_meta import DecodeMeta, MixedMeta, PrefillMeta
 from llmq_amd.engine.kv_cache import KVCache
 from llmq_amd.engine.model_specs import ModelSpec
 from llmq_amd.parallel import get_tp_group
+
+
+# Decode projection GEMMs through the hand-written skinny-M kernel
+# (ops.skinny_gemm) instead of hipBLASLt. Opt-in while A/B-ing
+# (LLMQ_SKINNY_GEMM=1); shapes outside the kernel's envelope fall back.
+_USE_SKINNY_GEMM = os.environ.get("LLMQ_SKINNY_GEMM", "0") not in ("0", "false", "")
+
+
+def _proj(x: torch.Tensor, w: torch.Tensor, bias=None) -> torch.Tensor:
+    if (_USE_SKINNY_GEMM and x.is_cuda and x.dtype == torch.bfloat16
+            and x.shape[0] <= 1024 and w.shape[1] % 64 == 0
+            and w.shape[0] % 4 == 0):
+        return ops.skinny_gemm(x, w, bias)
+    return F.linear(x, w, bias)
 
 
 class LayerWeights:
@@ -232,7 +247,7 @@ class CausalLM:
         for i, lw in enumerate(self.layers):
             # ---- attention block (h = normed input, set by the previous
             # block's fused norm — launch count is the decode-step bound)
-            qkv = F.linear(h, lw.qkv, lw.qkv_bias)
+            qkv = _proj(h, lw.qkv, lw.qkv_bias)
             q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
             T = q.shape[0]
             q = q.view(T, self.heads, s.head_dim)
@@ -262,7 +277,7 @@ class CausalLM:
                     q, kv_cache.k[i], kv_cache.v[i], meta.block_tables,
                     meta.context_lens, s.scale, s.attn_softcap, window,
                 )
-            attn_out = F.linear(attn.reshape(T, self.q_size), lw.o)
+            attn_out = _proj(attn.reshape(T, self.q_size), lw.o)
             if tp is not None:
                 attn_out = tp.all_reduce(attn_out)
             if s.post_norms:
@@ -277,9 +292,9 @@ class CausalLM:
                     attn_out, residual, lw.pre_mlp_norm, s.rms_eps, self.norm_offset
                 )
             # ---- MLP block
-            gate_up = F.linear(h, lw.gate_up)
+            gate_up = _proj(h, lw.gate_up)
             act = ops.gelu_tanh_and_mul(gate_up) if s.gelu else ops.silu_and_mul(gate_up)
-            mlp_out = F.linear(act, lw.down)
+            mlp_out = _proj(act, lw.down)
             if tp is not None:
                 mlp_out = tp.all_reduce(mlp_out)
             # The next block's input norm (or the final norm) fuses with this
@@ -301,7 +316,7 @@ class CausalLM:
     @torch.no_grad()
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
         """hidden [N, hidden] → logits [N, vocab] (fp32)."""
-        logits = F.linear(hidden, self.lm_head).float()
+        logits = _proj(hidden, self.lm_head).float()
         cap = self.spec.final_softcap
         if cap and cap > 0:
             logits = torch.tanh(logits / cap) * cap
