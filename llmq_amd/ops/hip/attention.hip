@@ -1466,6 +1466,288 @@ __global__ __launch_bounds__(NW * WAVE, (PD_KT <= 64 ? 4 : 2)) void paged_decode
   }
 }
 
+// ------------------------------------------- ring-buffered MFMA decode --
+// Variant of paged_decode_pipe_kernel with SEPARATE K and V slot pairs and
+// 32-key chunks: both V(i) and K(i+1) are issued at the LOOP TOP, so every
+// DMA gets a full iteration to stream (the shared-buffer kernel can only
+// issue V after S finishes reading the buffer — V's stream window is just
+// the softmax phase). Costs 3 barriers per 32 keys (vs 3 per 64); the bet
+// is that removing the V stall beats the barrier overhead. A/B via
+// LLMQ_DECODE_PIPE=32.
+template <int HEAD_DIM, int NW>
+__global__ __launch_bounds__(NW * WAVE, 4) void paged_decode_ring_kernel(
+    __hip_bfloat16* __restrict__ out,      // [B, H, D]
+    const __hip_bfloat16* __restrict__ q,  // [B, H, D]
+    const __hip_bfloat16* __restrict__ k_cache,  // [nb, KVH, bs, D]
+    const __hip_bfloat16* __restrict__ v_cache,
+    const int* __restrict__ block_tables,  // [B, max_blocks<=PD_MAX_BT]
+    const int* __restrict__ context_lens,  // [B]
+    float* __restrict__ scratch,           // [B,KVH,NSPLIT,G,D+2] (NSPLIT>1)
+    int num_heads, int num_kv_heads, int block_size, int max_blocks,
+    float scale, float softcap, int window, long q_stride, long out_stride) {
+  constexpr int D = HEAD_DIM;
+  constexpr int KS = D / 32;
+  constexpr int PD_KT = 32;
+  constexpr int NT = NW * WAVE;
+  constexpr int SLABS = PD_KT / 16;   // 2
+  constexpr int D4 = D / NW;
+  constexpr int DT = D4 / 16;
+  constexpr int CPK = D / 8;
+  constexpr int NI_K = PD_KT * CPK / NT;
+  constexpr int NSUB = (PD_KT / 32) * (D / 16);
+  constexpr int NI_V = NSUB / NW;
+  constexpr int K_ELEMS = PD_KT * D;
+  constexpr int V_ELEMS = NSUB * 528;
+  static_assert(NI_K >= 1 && NI_V >= 1 && D % NW == 0 && D4 % 16 == 0, "");
+
+  const int b = blockIdx.x;
+  const int kh = blockIdx.y;
+  const int G = num_heads / num_kv_heads;
+  const int L = context_lens[b];
+  if (L <= 0) return;
+
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid & (WAVE - 1);
+  const int col = lane & 15;
+  const int kgrp = lane >> 4;
+  const int slab = wid % SLABS;
+
+  constexpr int P_OFF = 2 * K_ELEMS + 2 * V_ELEMS;      // shorts
+  constexpr int F_BASE = (P_OFF + 16 * PD_KT + 7) / 8 * 4;  // float idx
+  constexpr int ALPHA_F = F_BASE + SLABS * 16;
+  constexpr int LPART_F = ALPHA_F + 16;
+  constexpr int BT_I = LPART_F + SLABS * 16;
+  constexpr int TOTAL_BYTES = BT_I * 4 + PD_MAX_BT * 4;
+  __shared__ __attribute__((aligned(16))) char smem[TOTAL_BYTES];
+  short* const ks0 = reinterpret_cast<short*>(smem);
+  short* const ks1 = ks0 + K_ELEMS;
+  short* const vs0 = ks1 + K_ELEMS;
+  short* const vs1 = vs0 + V_ELEMS;
+  short* const p_lds2 = ks0 + P_OFF;                    // [16][PD_KT]
+  float* const fbase = reinterpret_cast<float*>(smem);
+  float* const mpart = fbase + F_BASE;
+  float* const alpha_s = fbase + ALPHA_F;
+  float* const l_part = fbase + LPART_F;
+  int* const bt_l = reinterpret_cast<int*>(smem) + BT_I;
+
+  const int* bt_glob = block_tables + (long)b * max_blocks;
+  {
+    const int nb = (L + block_size - 1) / block_size;
+    for (int i = tid; i < nb; i += NT) bt_l[i] = bt_glob[i];
+  }
+
+  bf16x8_t qfrag[KS];
+  {
+    const int qrow = (col < G) ? col : 0;
+    const __hip_bfloat16* qp = q + (long)b * q_stride + (long)(kh * G + qrow) * D;
+#pragma unroll
+    for (int ks = 0; ks < KS; ++ks) {
+      qfrag[ks] = *reinterpret_cast<const bf16x8_t*>(qp + ks * 32 + kgrp * 8);
+      reg_fence(qfrag[ks]);
+    }
+  }
+
+  float m_regs[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float l_regs[4] = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t ot[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt) ot[dt] = {0.f, 0.f, 0.f, 0.f};
+
+  const int start = (window > 0 && L > window) ? (L - window) : 0;
+  const int base0 = (start / PD_KT) * PD_KT;
+  const int nsplit = gridDim.z;
+  int range_lo = base0, range_hi = L;
+  if (nsplit > 1) {
+    const int nchunks = (L - base0 + PD_KT - 1) / PD_KT;
+    const int per = (nchunks + nsplit - 1) / nsplit;
+    range_lo = base0 + (int)blockIdx.z * per * PD_KT;
+    range_hi = min(L, range_lo + per * PD_KT);
+  }
+
+  auto k_src = [&](int base, int g) -> const __hip_bfloat16* {
+    const int key = g / CPK;
+    const int r8 = (g % CPK) * 8;
+    const int d = r8 ^ ((key & 7) << 3);
+    const int gkey = min(base + key, L - 1);
+    const long blk = bt_l[gkey / block_size];
+    return k_cache + (blk * num_kv_heads + kh) * ((long)block_size * D) +
+           (long)(gkey % block_size) * D + d;
+  };
+  auto v_src = [&](int base, int st) -> const __hip_bfloat16* {
+    const int ks = st / (D / 16), dtile = st % (D / 16);
+    const int pos = lane * 8;
+    const int bpos = pos / 64, rem = pos % 64;
+    const int qq = ((bpos >> 2) & 1) | ((bpos & 3) << 1);
+    const int key = ks * 32 + qq * 4 + rem / 16;
+    const int dim = dtile * 16 + (rem & 15);
+    const int gkey = min(base + key, L - 1);
+    const long blk = bt_l[gkey / block_size];
+    return v_cache + (blk * num_kv_heads + kh) * ((long)block_size * D) +
+           (long)(gkey % block_size) * D + dim;
+  };
+  auto issue_k = [&](int base, short* dst) {
+#pragma unroll
+    for (int j = 0; j < NI_K; ++j) {
+      const int inst = wid * NI_K + j;
+      const int off = __builtin_amdgcn_readfirstlane(inst * 512);
+      glds16(k_src(base, inst * 64 + lane), dst + off);
+    }
+  };
+  auto issue_v = [&](int base, short* dst) {
+#pragma unroll
+    for (int j = 0; j < NI_V; ++j) {
+      const int st = wid * NI_V + j;
+      const int off = __builtin_amdgcn_readfirstlane(st * 528);
+      glds16(v_src(base, st), dst + off);
+    }
+  };
+
+  pipe_barrier();  // bt_l ready
+  if (range_lo < range_hi) issue_k(range_lo, ks0);
+
+  int cur = 0;
+  for (int base = range_lo; base < range_hi; base += PD_KT, cur ^= 1) {
+    short* const KX = cur ? ks1 : ks0;
+    short* const KY = cur ? ks0 : ks1;
+    short* const VX = cur ? vs1 : vs0;
+    // K(base) landed; V slot VX held V(base-2): PV(base-2) finished two
+    // barriers ago; KY held K(base-1): S(base-1) done.
+    pipe_barrier_vm<0>();
+    issue_v(base, VX);
+    const int next = base + PD_KT;
+    const bool prefetch = next < range_hi;
+    if (prefetch) issue_k(next, KY);
+
+    // ---- S[16,16] for this wave's slab
+    f32x4_t s = {0.f, 0.f, 0.f, 0.f};
+    {
+      const int key = slab * 16 + col;
+#pragma unroll
+      for (int ks = 0; ks < KS; ++ks) {
+        const int d8 = ks * 32 + kgrp * 8;
+        bf16x8_t bfrag =
+            *reinterpret_cast<const bf16x8_t*>(&KX[key * D + swz(key, d8)]);
+        s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks], bfrag, s, 0, 0, 0);
+      }
+    }
+    float sv[4];
+    {
+      const int key = base + slab * 16 + col;
+      const bool dead = key >= L || key < start;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        float x = s[reg] * scale;
+        if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
+        sv[reg] = dead ? -1e30f : x;
+        float m = sv[reg];
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) m = fmaxf(m, __shfl_xor(m, off, WAVE));
+        if (col == 0 && wid < SLABS) mpart[slab * 16 + kgrp * 4 + reg] = m;
+      }
+    }
+    pipe_barrier();  // mparts visible (K reads done; no buffer handoff)
+
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int row = kgrp * 4 + reg;
+      float m_tile = mpart[row];
+#pragma unroll
+      for (int w = 1; w < SLABS; ++w) m_tile = fmaxf(m_tile, mpart[w * 16 + row]);
+      const float m_new = fmaxf(m_regs[reg], m_tile);
+      const float alpha = (m_new > -1e30f) ? __expf(m_regs[reg] - m_new) : 1.f;
+      const float pe = (m_new > -1e30f && sv[reg] > -1e29f)
+                           ? __expf(sv[reg] - m_new) : 0.f;
+      float lsum = pe;
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) lsum += __shfl_xor(lsum, off, WAVE);
+      l_regs[reg] = l_regs[reg] * alpha + lsum;
+      m_regs[reg] = m_new;
+      if (wid == 0 && col == 0) alpha_s[row] = alpha;
+      p_lds2[row * PD_KT + slab * 16 + col] =
+          __bfloat16_as_short(__float2bfloat16(pe));
+    }
+    if (prefetch) {
+      pipe_barrier_vm<NI_K>();  // V(base) landed; K(next) stays in flight
+    } else {
+      pipe_barrier_vm<0>();
+    }
+
+    const float alpha_q = alpha_s[col];
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
+      ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
+      const int dtile = (wid * D4) / 16 + dt;
+#pragma unroll
+      for (int ks = 0; ks < PD_KT / 32; ++ks) {
+        const int sub = (ks * (D / 16) + dtile) * 528 + lane * 4;
+        bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_bf16x4*)&VX[sub]);
+        bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_bf16x4*)&VX[sub + 4 * 64]);
+        bf16x8_t a;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          a[j] = __bfloat16_as_short((__hip_bfloat16)lo[j]);
+          a[4 + j] = __bfloat16_as_short((__hip_bfloat16)hi[j]);
+        }
+        bf16x8_t bb = *reinterpret_cast<const bf16x8_t*>(
+            &p_lds2[col * PD_KT + ks * 32 + kgrp * 8]);
+        ot[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bb, ot[dt], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- merge per-slab l, then store
+  if (col == 0 && wid < SLABS) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) l_part[slab * 16 + kgrp * 4 + reg] = l_regs[reg];
+  }
+  pipe_barrier();
+  float l_tot = l_part[col];
+#pragma unroll
+  for (int w = 1; w < SLABS; ++w) l_tot += l_part[w * 16 + col];
+  if (nsplit > 1) {
+    float* slot = scratch +
+        ((((long)b * num_kv_heads + kh) * nsplit + blockIdx.z) * G) * (D + 2);
+    if (col < G) {
+      float* row = slot + (long)col * (D + 2);
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        const int dim0 = wid * D4 + dt * 16 + kgrp * 4;
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) row[dim0 + reg] = ot[dt][reg];
+      }
+    }
+    if (wid == 0 && col == 0) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int r = kgrp * 4 + reg;
+        if (r < G) {
+          float* row = slot + (long)r * (D + 2);
+          row[D] = m_regs[reg];
+          float lt = l_part[r];
+          for (int w = 1; w < SLABS; ++w) lt += l_part[w * 16 + r];
+          row[D + 1] = lt;
+        }
+      }
+    }
+    return;
+  }
+  const float inv = (l_tot > 0.f) ? 1.0f / l_tot : 0.f;
+  if (col < G) {
+    __hip_bfloat16* op = out + (long)b * out_stride + (long)(kh * G + col) * D;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      const int dim0 = wid * D4 + dt * 16 + kgrp * 4;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg)
+        op[dim0 + reg] = __float2bfloat16(ot[dt][reg] * inv);
+    }
+  }
+}
+
 // Combine split-KV partials: out[b, kh*G+g] = Σ_s w_s·acc_s / Σ_s w_s·l_s,
 // w_s = exp(m_s − max_s m). One wave per (b, kh, g).
 template <int HEAD_DIM>

@@ -281,6 +281,29 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
                                scratch_ptr, KVH, G, nsplit, out.stride(0));
           }
         };
+        if (pipe_kt == 32) {  // ring variant: separate K/V slots, 32-key
+          auto lr = [&]<int HD>() {
+            hipLaunchKernelGGL((paged_decode_ring_kernel<HD, 8>), sgrid,
+                               dim3(8 * 64), 0, stream(),
+                               reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                               reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                               reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()),
+                               reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()),
+                               bt.data_ptr<int>(), cl.data_ptr<int>(), scratch_ptr,
+                               H, KVH, bs, max_blocks, (float)scale,
+                               (float)softcap, (int)window, q.stride(0),
+                               out.stride(0));
+            if (nsplit > 1) {
+              hipLaunchKernelGGL((decode_splitkv_merge_kernel<HD>),
+                                 dim3(B, KVH, G), dim3(64), 0, stream(),
+                                 reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                                 scratch_ptr, KVH, G, nsplit, out.stride(0));
+            }
+          };
+          if (D == 128) lr.template operator()<128>();
+          else lr.template operator()<256>();
+          return;
+        }
         if (D == 128) {
           if (pipe_kt >= 128) lp.template operator()<128, 128>();
           else lp.template operator()<128, 64>();
@@ -475,7 +498,19 @@ void skinny_gemm(at::Tensor out, at::Tensor x, at::Tensor w,
     while (mt * nt * z < 256 && (K / SKG_BK) / (z * 2) >= 8 && z < 8) z *= 2;
   }
   dim3 grid(mt, nt, z);
+  static const int sync_dbg = [] {
+    const char* e = getenv("LLMQ_SKG_SYNC");
+    return e ? atoi(e) : 0;
+  }();
   if (z == 1) {
+    if (sync_dbg) {
+      hipLaunchKernelGGL((skinny_gemm_kernel<0, 1>), grid, dim3(SKG_NT), 0,
+                         stream(), out.data_ptr(),
+                         reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         bias_ptr, M, N, K, x.stride(0), w.stride(0));
+      return;
+    }
     hipLaunchKernelGGL((skinny_gemm_kernel<0>), grid, dim3(SKG_NT), 0, stream(),
                        out.data_ptr(),
                        reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),

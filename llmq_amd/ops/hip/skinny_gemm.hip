@@ -37,24 +37,28 @@ typedef __attribute__((ext_vector_type(4))) float skg_f32x4;
 
 DEVINL int skg_swz(int row, int col) { return col ^ (((row >> 3) & 1) << 4); }
 
-// Stage one [256][64] bf16 tile (A or B) via glds: 1024 granules of 16B,
-// 2 instructions per wave (8 waves x 64 lanes x 2). Source row clamped.
+// Stage one [256][64] bf16 tile (A or B) via glds: 2048 granules of 16B
+// (32 KB), 4 instructions per wave (8 waves x 64 lanes x 4). Source row
+// clamped to the tensor (finite garbage beyond the tail; outputs there
+// are never stored).
 DEVINL void skg_stage(const __hip_bfloat16* __restrict__ src, long src_stride,
                       int row0, int rows_total, int k0, short* dst,
                       int wid, int lane) {
 #pragma unroll
-  for (int j = 0; j < 2; ++j) {
-    const int g = (wid * 2 + j) * 64 + lane;   // granule 0..1023
+  for (int j = 0; j < 4; ++j) {
+    const int g = (wid * 4 + j) * 64 + lane;   // granule 0..2047
     const int row = g >> 3;                    // 8 granules per 64-elem row
     const int col = (g & 7) * 8;
     const int srow = min(row0 + row, rows_total - 1);
     const int scol = k0 + skg_swz(row, col);
-    const int off = __builtin_amdgcn_readfirstlane((wid * 2 + j) * 512);
+    const int off = __builtin_amdgcn_readfirstlane((wid * 4 + j) * 512);
     glds16(src + (long)srow * src_stride + scol, dst + off);
   }
 }
 
-template <int SPLITK_TAG>  // 0: bf16 direct out; 1: fp32 slab out
+// SYNC_DEBUG: plain-load + ds_write staging with __syncthreads (no glds,
+// no pipeline) — used to isolate staging-pipeline bugs from layout bugs.
+template <int SPLITK_TAG, int SYNC_DEBUG = 0>
 __global__ __launch_bounds__(SKG_NT, 2) void skinny_gemm_kernel(
     void* __restrict__ out_raw,            // bf16 [M,N] or f32 [Z,M,N]
     const __hip_bfloat16* __restrict__ x,  // [M,K]
@@ -96,22 +100,47 @@ __global__ __launch_bounds__(SKG_NT, 2) void skinny_gemm_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  auto stage_sync = [&](const __hip_bfloat16* src, long stride, int row0,
+                        int rows_total, int k0, short* dst) {
+    for (int g = tid * 2; g < 2048; g += SKG_NT * 2) {
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const int gg = g + j;
+        const int row = gg >> 3;
+        const int col = (gg & 7) * 8;
+        const int srow = min(row0 + row, rows_total - 1);
+        const int scol = k0 + skg_swz(row, col);
+        *reinterpret_cast<skg_bf16x8*>(&dst[gg * 8]) =
+            *reinterpret_cast<const skg_bf16x8*>(src + (long)srow * stride + scol);
+      }
+    }
+  };
+
   // prologue: stage K-step ks0 into buffer 0
-  skg_stage(x, x_stride, row_a0, M, ks0 * SKG_BK, a0, wid, lane);
-  skg_stage(w, w_stride, row_b0, N, ks0 * SKG_BK, b0, wid, lane);
+  if constexpr (!SYNC_DEBUG) {
+    skg_stage(x, x_stride, row_a0, M, ks0 * SKG_BK, a0, wid, lane);
+    skg_stage(w, w_stride, row_b0, N, ks0 * SKG_BK, b0, wid, lane);
+  }
 
   for (int ks = ks0; ks < ks1; ++ks) {
     short* const a = (ks - ks0) & 1 ? a1 : a0;
     short* const b = (ks - ks0) & 1 ? b1 : b0;
     short* const an = (ks - ks0) & 1 ? a0 : a1;
     short* const bn = (ks - ks0) & 1 ? b0 : b1;
-    // current tile landed (the 4 staging glds of this buffer)
-    pipe_barrier_vm<0>();
-    // prefetch next K-step into the other buffer (stays in flight through
-    // the MFMA phase; drained by the next iteration's barrier)
-    if (ks + 1 < ks1) {
-      skg_stage(x, x_stride, row_a0, M, (ks + 1) * SKG_BK, an, wid, lane);
-      skg_stage(w, w_stride, row_b0, N, (ks + 1) * SKG_BK, bn, wid, lane);
+    if constexpr (SYNC_DEBUG) {
+      __syncthreads();
+      stage_sync(x, x_stride, row_a0, M, ks * SKG_BK, a);
+      stage_sync(w, w_stride, row_b0, N, ks * SKG_BK, b);
+      __syncthreads();
+    } else {
+      // current tile landed (the 4 staging glds of this buffer)
+      pipe_barrier_vm<0>();
+      // prefetch next K-step into the other buffer (stays in flight
+      // through the MFMA phase; drained by the next iteration's barrier)
+      if (ks + 1 < ks1) {
+        skg_stage(x, x_stride, row_a0, M, (ks + 1) * SKG_BK, an, wid, lane);
+        skg_stage(w, w_stride, row_b0, N, (ks + 1) * SKG_BK, bn, wid, lane);
+      }
     }
     // MFMA over this K-step: per wave 8 m-frags x 4 n-frags x 2 k-substeps
 #pragma unroll

@@ -126,7 +126,7 @@ def ab_bench(args) -> None:
     scale = D ** -0.5
     kv_bytes = 2 * B * L * KVH * D * 2
 
-    variants = ["0", "64", "128"]
+    variants = ["0", "32", "64", "128"]
 
     def run_once():
         return ops.paged_decode_attention(
