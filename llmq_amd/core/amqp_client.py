@@ -72,6 +72,7 @@ class AMQPBrokerClient(BrokerClient):
         self._cid_ctag: Dict[int, str] = {}
         self._next_cid = 0
         self._get_future: Optional[asyncio.Future] = None
+        self._mgmt_ok: Optional[bool] = None  # memoised mgmt-API reachability
 
     # -- connection ------------------------------------------------------
 
@@ -433,7 +434,61 @@ class AMQPBrokerClient(BrokerClient):
 
     # -- stats / admin ---------------------------------------------------
 
+    # RabbitMQ management HTTP API (reference broker.py:244-289): full
+    # stats incl. message bytes + queue listing. Port = AMQP port + 10000
+    # (RabbitMQ's 5672 -> 15672 convention) unless LLMQ_MGMT_URL overrides.
+    def _mgmt_base(self) -> str:
+        import os
+
+        return os.environ.get(
+            "LLMQ_MGMT_URL",
+            f"http://{self.amqp_host}:{self.amqp_port + 10000}",
+        ).rstrip("/")
+
+    async def _mgmt_get(self, path: str):
+        import httpx
+
+        vhost = urllib.parse.quote(self.amqp_vhost, safe="")
+        async with httpx.AsyncClient(timeout=5.0) as client:
+            r = await client.get(
+                f"{self._mgmt_base()}{path.format(vhost=vhost)}",
+                auth=(self.amqp_user, self.amqp_password),
+            )
+            r.raise_for_status()
+            return r.json()
+
+    @staticmethod
+    def _stats_from_mgmt(blob: Dict[str, Any]) -> QueueStats:
+        return QueueStats(
+            queue_name=blob.get("name", ""),
+            message_count=blob.get("messages", 0) or 0,
+            message_count_ready=blob.get("messages_ready", 0) or 0,
+            message_count_unacknowledged=blob.get("messages_unacknowledged", 0) or 0,
+            consumer_count=blob.get("consumers", 0) or 0,
+            message_bytes=blob.get("message_bytes", 0) or 0,
+            message_bytes_ready=blob.get("message_bytes_ready", 0) or 0,
+            message_bytes_unacknowledged=blob.get(
+                "message_bytes_unacknowledged", 0) or 0,
+            stats_source="management_api",
+        )
+
     async def get_queue_stats(self, queue_name: str) -> QueueStats:
+        # 1) management API (full stats); 2) passive declare fallback
+        # (the reference's amqp_fallback, broker.py:235-239)
+        if self._mgmt_ok is not False:
+            try:
+                blob = await self._mgmt_get(
+                    "/api/queues/{vhost}/" + urllib.parse.quote(queue_name, safe=""))
+                self._mgmt_ok = True
+                return self._stats_from_mgmt(blob)
+            except Exception as exc:  # noqa: BLE001 — mgmt API optional
+                import httpx
+
+                if isinstance(exc, httpx.HTTPStatusError) and exc.response.status_code == 404:
+                    self._mgmt_ok = True  # API reachable; queue absent
+                    return QueueStats(queue_name=queue_name,
+                                      stats_source="unavailable")
+                self._mgmt_ok = False
         try:
             await self._ensure_channel()
             reply = await self._rpc(
@@ -446,17 +501,21 @@ class AMQPBrokerClient(BrokerClient):
                 message_count=reply["message_count"],
                 message_count_ready=reply["message_count"],
                 consumer_count=reply["consumer_count"],
-                stats_source="amqp",
+                stats_source="amqp_fallback",
             )
         except (ChannelClosed, ConnectionError, asyncio.TimeoutError):
             return QueueStats(queue_name=queue_name, stats_source="unavailable")
 
     async def list_queues(self) -> List[QueueStats]:
-        raise NotImplementedError(
-            "AMQP has no queue-listing method; use the RabbitMQ management "
-            "API or the in-tree broker (llmq:// URL) for `llmq status` "
-            "without an explicit queue name"
-        )
+        try:
+            blobs = await self._mgmt_get("/api/queues/{vhost}")
+            return [self._stats_from_mgmt(b) for b in blobs]
+        except Exception as exc:  # noqa: BLE001
+            raise RuntimeError(
+                "queue listing over AMQP needs the RabbitMQ management API "
+                f"({self._mgmt_base()}; set LLMQ_MGMT_URL) — or use the "
+                "in-tree broker (llmq:// URL)"
+            ) from exc
 
     async def _basic_get(self, queue: str):
         async with self._rpc_lock:
