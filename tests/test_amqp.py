@@ -260,6 +260,35 @@ def test_dummy_worker_full_stack_over_amqp(tmp_path):
     run_async(main())
 
 
+def test_amqp_large_body_multi_frame():
+    """Bodies larger than frame_max must split into multiple body frames on
+    publish and reassemble on both server and client (spec §2.3.7)."""
+
+    async def main():
+        async with live_broker() as (server, _cfg):
+            client = BrokerClient(_amqp_config(server))
+            await client.connect()
+            await client.setup_queue_infrastructure("big")
+            blob = "x" * 400_000 + "END"  # ~3 body frames at 128 KiB frame_max
+            await client.publish_job("big", Job(id="big1", prompt=blob))
+
+            got = asyncio.Event()
+            bodies = []
+
+            async def cb(d):
+                bodies.append(d.body)
+                await d.ack()
+                got.set()
+
+            await client.consume_jobs("big", cb)
+            await asyncio.wait_for(got.wait(), 15)
+            job = Job.model_validate_json(bodies[0])
+            assert job.prompt == blob
+            await client.disconnect()
+
+    run_async(main())
+
+
 # --------------------------------------------------------- real RabbitMQ --
 
 @pytest.mark.rabbitmq
