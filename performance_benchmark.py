@@ -290,7 +290,7 @@ def main() -> None:
     ap = argparse.ArgumentParser(description=__doc__)
     ap.add_argument("--model", default="tower-plus-2b")
     ap.add_argument("--samples", type=int, default=1000)
-    ap.add_argument("--batch-sizes", default="16,32,64,128,256",
+    ap.add_argument("--batch-sizes", default="128,256,512",
                     help="max_num_seqs sweep (reference default, performance_benchmark.py:645)")
     ap.add_argument("--workers", type=int, default=1)
     ap.add_argument("--gpus", type=int, default=0, help="GPUs to spread workers over (0=CPU)")
