@@ -95,6 +95,15 @@ class AMQPBrokerClient(BrokerClient):
         )
 
     async def _connect_once(self) -> None:
+        # fresh connection = fresh channel: confirm tags restart server-side
+        self._pub_seq = 0
+        self._confirmed = 0
+        for _seq, f in self._confirm_waiters:
+            if not f.done():
+                f.set_exception(ConnectionError("reconnected"))
+        self._confirm_waiters.clear()
+        self._chan_open = False
+        self._asm = None
         self._reader, self._writer = await asyncio.open_connection(
             self.amqp_host, self.amqp_port
         )
