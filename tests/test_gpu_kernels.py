@@ -490,6 +490,46 @@ class TestFusedRopeCache:
         assert torch.equal(vc.cpu().float(), vc_ref)  # pure copy: bitwise
 
 
+class TestMixedOverlapGPU:
+    def test_overlapped_mixed_step_matches_sequential(self, monkeypatch):
+        """The split mixed step runs decode (hipGraph, side stream) and
+        prefill (default stream) CONCURRENTLY; tokens must be identical to
+        the sequential split (same kernels, same counter-based RNG)."""
+        from llmq_amd.engine.config import EngineConfig
+        from llmq_amd.engine.engine import LLMEngine
+        from llmq_amd.engine.sampling_params import SamplingParams
+
+        def run(overlap: str):
+            monkeypatch.setenv("LLMQ_OVERLAP_MIXED", overlap)
+            eng = LLMEngine(EngineConfig(
+                model="llama-3.2-1b", max_num_seqs=8, max_model_len=256,
+                load_weights=False, num_kv_blocks=512, seed=7,
+            ))
+            params = SamplingParams(temperature=0.7, max_tokens=24, ignore_eos=True)
+            tokens = {}
+            for i in range(4):
+                eng.add_request(f"a{i}", prompt=f"first wave {i}", params=params)
+            # a few decode-only steps, then admit more -> MIXED steps
+            for _ in range(6):
+                for out in eng.step():
+                    tokens.setdefault(out.request_id, []).extend(out.new_token_ids)
+            for i in range(4):
+                eng.add_request(f"b{i}", prompt=f"second wave {i} arrives later",
+                                params=params)
+            while eng.has_unfinished():
+                for out in eng.step():
+                    tokens.setdefault(out.request_id, []).extend(out.new_token_ids)
+            del eng
+            torch.cuda.empty_cache()
+            return tokens
+
+        seq = run("0")
+        ovl = run("1")
+        assert seq.keys() == ovl.keys()
+        for k in seq:
+            assert seq[k] == ovl[k], (k, seq[k][:6], ovl[k][:6])
+
+
 class TestEngineFamiliesGPU:
     """GPU vs CPU engine consistency per model family: the same seeded
     random-init model must sample the same greedy tokens through the HIP
