@@ -8,7 +8,9 @@ N data-parallel engine workers, one rank per MI355X over RCCL.
 A "step" is one continuous-batching decode iteration of the full resident
 batch (the engine's serving step: scheduler → hipGraph decode forward →
 sampling → bookkeeping). The engine is pre-filled with --batch sequences of
---prompt-len synthetic tokens; after --warmup untimed steps, EXACTLY
+--prompt-len synthetic tokens (default batch 512 — 288 GB HBM3E holds far
+more resident sequences than the reference's 750 default, and the larger
+M amortises the per-step weight stream); after --warmup untimed steps, EXACTLY
 --steps steps are timed between barrier+synchronize fences; the slowest
 rank's wall time is used. value = N_ranks × batch × steps ÷ max_elapsed.
 
