@@ -352,3 +352,34 @@ def test_bridge_high_concurrency_fake_engine():
         bridge.shutdown()
 
     run_async(main())
+
+
+def test_engine_worker_roundtrip_over_amqp():
+    """The FULL GPU-worker path (engine worker consume → generate →
+    publish_result → ack) over REAL AMQP 0-9-1 framing, i.e. what a
+    RabbitMQ-backed deployment runs (amqp:// URL selects the from-scratch
+    AMQP client backend against the broker's AMQP front-end)."""
+    from llmq_amd.core.config import Config
+
+    async def main():
+        async with live_broker() as (server, _cfg):
+            config = Config(broker_url=f"amqp://guest:guest@127.0.0.1:{server.port}/")
+            client = BrokerClient(config)
+            await client.connect()
+            await client.setup_queue_infrastructure("aeq")
+            jobs = [
+                Job(id=f"aj{i}", prompt="hello {name}", name=f"n{i}",
+                    max_tokens=4, temperature=0.0)
+                for i in range(4)
+            ]
+            await client.publish_jobs("aeq", jobs)
+            worker = _make_worker("aeq", config)
+            task = await _start_worker(worker)
+            results = await _collect_results(client, "aeq", 4, timeout=60.0)
+            assert {r.id for r in results} == {f"aj{i}" for i in range(4)}
+            for r in results:
+                assert r.output_tokens is not None and r.output_tokens >= 1
+            await _stop_worker(worker, task)
+            await client.disconnect()
+
+    run_async(main())
