@@ -861,12 +861,24 @@ DEVINL bf16x8_t kv8_to_bf16(const __hip_bfloat16* p) {
   return *reinterpret_cast<const bf16x8_t*>(p);
 }
 DEVINL bf16x8_t kv8_to_bf16(const __hip_fp8_e4m3* p) {
+  // gfx950 native packed OCP-fp8 converts: v_cvt_pk_f32_fp8 turns two fp8
+  // (low/high half of a 16-bit pair) into two f32 in one VOP. The generic
+  // static_cast<float>(__hip_fp8_e4m3) lowers to a multi-op bit sequence
+  // per ELEMENT and made fp8 staging VALU-bound (fp8 decode measured
+  // SLOWER than bf16 at batch 512 before this).
   const uint2 raw = *reinterpret_cast<const uint2*>(p);
-  const __hip_fp8_e4m3* e = reinterpret_cast<const __hip_fp8_e4m3*>(&raw);
   bf16x8_t out;
 #pragma unroll
-  for (int j = 0; j < 8; ++j)
-    out[j] = __bfloat16_as_short(__float2bfloat16(static_cast<float>(e[j])));
+  for (int h = 0; h < 2; ++h) {
+    const int w = h ? (int)raw.y : (int)raw.x;
+    typedef __attribute__((ext_vector_type(2))) float f32x2_t;
+    const f32x2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+    const f32x2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
+    out[h * 4 + 0] = __bfloat16_as_short(__float2bfloat16(lo[0]));
+    out[h * 4 + 1] = __bfloat16_as_short(__float2bfloat16(lo[1]));
+    out[h * 4 + 2] = __bfloat16_as_short(__float2bfloat16(hi[0]));
+    out[h * 4 + 3] = __bfloat16_as_short(__float2bfloat16(hi[1]));
+  }
   return out;
 }
 

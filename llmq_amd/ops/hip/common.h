@@ -31,7 +31,12 @@ template <> DEVINL __hip_bfloat16 from_f32<__hip_bfloat16>(float x) {
 }
 template <> DEVINL _Float16 from_f32<_Float16>(float x) { return (_Float16)x; }
 template <> DEVINL __hip_fp8_e4m3 from_f32<__hip_fp8_e4m3>(float x) {
-  return __hip_fp8_e4m3(x);
+  // native v_cvt_pk_fp8_f32 (gfx950): one VOP instead of the library's
+  // per-element software sequence (RNE, saturating — matches OCP e4m3fn)
+  const int packed = __builtin_amdgcn_cvt_pk_fp8_f32(x, x, 0, false);
+  __hip_fp8_e4m3 out;
+  out.__x = (unsigned char)(packed & 0xff);
+  return out;
 }
 
 // ---- vectorized 16-byte access (8 bf16 / 8 fp16 / 4 f32) -------------------
