@@ -1478,6 +1478,335 @@ __global__ __launch_bounds__(NW * WAVE, (PD_KT <= 64 ? 4 : 2)) void paged_decode
   }
 }
 
+// --------------------------------------- fp8-KV pipelined MFMA decode --
+// The glds pipeline for fp8 (OCP e4m3) caches: HALF the KV stream of
+// bf16. glds cannot convert during the LDS write, so:
+//   - K stays fp8 IN LDS ([PD_KT][D] bytes, 16B-granule XOR swizzle on
+//     the source); the S phase reads 8-byte fragments and converts with
+//     the native packed cvt VOPs (one v_cvt_pk_f32_fp8 per 4 elems).
+//   - V stages RAW fp8 into the scratch back-half of the OTHER K/V-image
+//     buffer, then a convert pass (during the softmax window) writes the
+//     bf16 tr16 subtile image the PV path expects.
+// Buffer choreography per chunk i (X = img[i&1], Y = img[(i+1)&1]):
+//   top: vm(0) barrier [K(i) in X front]  ->  S(i) from X (fp8+cvt)
+//   post-S barrier -> issue V(i)raw -> Y back;  K(i+1) -> Y front
+//   softmax/P  ->  vm(NI_K) barrier [V raw landed; K(i+1) flying]
+//   convert pass: Y back (raw) -> X image (overwrites K(i): dead)
+//   lgkm barrier -> PV(i) from X image
+// Y-back is free during i (its image is only rebuilt by i+1's convert);
+// Y-front K(i+1) and Y-back V(i)raw are disjoint ranges. LDS stays at
+// 2 × image + p + bt ≈ 74 KB -> 2 workgroups/CU, same as the bf16 pipe.
+
+DEVINL bf16x8_t fp8x8_to_bf16(unsigned lo, unsigned hi) {
+  typedef __attribute__((ext_vector_type(2))) float cvt_f32x2;
+  bf16x8_t out;
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+    const int w = h ? (int)hi : (int)lo;
+    const cvt_f32x2 a = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+    const cvt_f32x2 b = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
+    out[h * 4 + 0] = __bfloat16_as_short(__float2bfloat16(a[0]));
+    out[h * 4 + 1] = __bfloat16_as_short(__float2bfloat16(a[1]));
+    out[h * 4 + 2] = __bfloat16_as_short(__float2bfloat16(b[0]));
+    out[h * 4 + 3] = __bfloat16_as_short(__float2bfloat16(b[1]));
+  }
+  return out;
+}
+
+template <int HEAD_DIM, int NW>
+__global__ __launch_bounds__(NW * WAVE, 4) void paged_decode_pipe_fp8_kernel(
+    __hip_bfloat16* __restrict__ out,      // [B, H, D]
+    const __hip_bfloat16* __restrict__ q,  // [B, H, D]
+    const __hip_fp8_e4m3* __restrict__ k_cache,  // [nb, KVH, bs, D]
+    const __hip_fp8_e4m3* __restrict__ v_cache,
+    const int* __restrict__ block_tables,  // [B, max_blocks<=PD_MAX_BT]
+    const int* __restrict__ context_lens,  // [B]
+    float* __restrict__ scratch,           // [B,KVH,NSPLIT,G,D+2] (NSPLIT>1)
+    int num_heads, int num_kv_heads, int block_size, int max_blocks,
+    float scale, float softcap, int window, long q_stride, long out_stride) {
+  constexpr int D = HEAD_DIM;
+  constexpr int KS = D / 32;
+  constexpr int PD_KT = 64;
+  constexpr int NT = NW * WAVE;
+  constexpr int SLABS = PD_KT / 16;     // 4 (waves duplicate when NW=8)
+  constexpr int D4 = D / NW;
+  constexpr int DT = D4 / 16;
+  constexpr int G16B = D / 16;          // 16B granules per fp8 key row
+  constexpr int NI_K = PD_KT * G16B / NT;   // fp8-K glds per wave
+  constexpr int NI_VR = NI_K;               // raw fp8-V glds per wave
+  constexpr int NSUB = (PD_KT / 32) * (D / 16);
+  constexpr int IMG_ELEMS = NSUB * 528;     // bf16 tr16 image (shorts)
+  constexpr int RAW_BYTES = PD_KT * D;      // fp8 K front == raw V size
+  static_assert(IMG_ELEMS * 2 >= 2 * RAW_BYTES,
+                "image must cover K-front + V-raw back");
+  static_assert(NI_K >= 1 && D % NW == 0 && D4 % 16 == 0, "");
+
+  const int b = blockIdx.x;
+  const int kh = blockIdx.y;
+  const int G = num_heads / num_kv_heads;
+  const int L = context_lens[b];
+  if (L <= 0) return;
+
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid & (WAVE - 1);
+  const int col = lane & 15;
+  const int kgrp = lane >> 4;
+  const int slab = wid % SLABS;
+
+  constexpr int P_OFF = 2 * IMG_ELEMS;                   // shorts
+  constexpr int F_BASE = (P_OFF + 16 * PD_KT + 7) / 8 * 4;
+  constexpr int ALPHA_F = F_BASE + SLABS * 16;
+  constexpr int LPART_F = ALPHA_F + 16;
+  constexpr int BT_I = LPART_F + SLABS * 16;
+  constexpr int TOTAL_BYTES = BT_I * 4 + PD_MAX_BT * 4;
+  __shared__ __attribute__((aligned(16))) char smem[TOTAL_BYTES];
+  short* const img0 = reinterpret_cast<short*>(smem);
+  short* const img1 = img0 + IMG_ELEMS;
+  short* const p_lds2 = img0 + P_OFF;
+  float* const fbase = reinterpret_cast<float*>(smem);
+  float* const mpart = fbase + F_BASE;
+  float* const alpha_s = fbase + ALPHA_F;
+  float* const l_part = fbase + LPART_F;
+  int* const bt_l = reinterpret_cast<int*>(smem) + BT_I;
+
+  const int* bt_glob = block_tables + (long)b * max_blocks;
+  {
+    const int nb = (L + block_size - 1) / block_size;
+    for (int i = tid; i < nb; i += NT) bt_l[i] = bt_glob[i];
+  }
+
+  bf16x8_t qfrag[KS];
+  {
+    const int qrow = (col < G) ? col : 0;
+    const __hip_bfloat16* qp = q + (long)b * q_stride + (long)(kh * G + qrow) * D;
+#pragma unroll
+    for (int ks = 0; ks < KS; ++ks) {
+      qfrag[ks] = *reinterpret_cast<const bf16x8_t*>(qp + ks * 32 + kgrp * 8);
+      reg_fence(qfrag[ks]);
+    }
+  }
+
+  float m_regs[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float l_regs[4] = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t ot[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt) ot[dt] = {0.f, 0.f, 0.f, 0.f};
+
+  const int start = (window > 0 && L > window) ? (L - window) : 0;
+  const int base0 = (start / PD_KT) * PD_KT;
+  const int nsplit = gridDim.z;
+  int range_lo = base0, range_hi = L;
+  if (nsplit > 1) {
+    const int nchunks = (L - base0 + PD_KT - 1) / PD_KT;
+    const int per = (nchunks + nsplit - 1) / nsplit;
+    range_lo = base0 + (int)blockIdx.z * per * PD_KT;
+    range_hi = min(L, range_lo + per * PD_KT);
+  }
+
+  auto kv_row = [&](int gkey, const __hip_fp8_e4m3* cache) -> const __hip_fp8_e4m3* {
+    const int ck = min(gkey, L - 1);  // clamp: finite real data (masked)
+    const long blk = bt_l[ck / block_size];
+    return cache + (blk * num_kv_heads + kh) * ((long)block_size * D) +
+           (long)(ck % block_size) * D;
+  };
+  // K fp8 image: [PD_KT][D] bytes, 16B-granule XOR swizzle (col16 ^ key&7)
+  auto issue_k = [&](int base, short* dst) {
+#pragma unroll
+    for (int j = 0; j < NI_K; ++j) {
+      const int g = (wid * NI_K + j) * 64 + lane;  // 16B granule
+      const int key = g / G16B;
+      const int c16 = g % G16B;
+      const int src_col = ((c16 ^ (key & 7)) & (G16B - 1)) << 4;
+      const int off = __builtin_amdgcn_readfirstlane((wid * NI_K + j) * 512);
+      glds16(kv_row(base + key, k_cache) + src_col,
+             reinterpret_cast<char*>(dst) + off * 2);
+    }
+  };
+  // V raw fp8, linear [PD_KT][D] bytes into the back half of `dst`
+  auto issue_v_raw = [&](int base, short* dst) {
+    char* const back = reinterpret_cast<char*>(dst) + RAW_BYTES;
+#pragma unroll
+    for (int j = 0; j < NI_VR; ++j) {
+      const int g = (wid * NI_VR + j) * 64 + lane;
+      const int key = g / G16B;
+      const int colb = (g % G16B) * 16;
+      const int off = __builtin_amdgcn_readfirstlane((wid * NI_VR + j) * 1024);
+      glds16(kv_row(base + key, v_cache) + colb, back + off);
+    }
+  };
+
+  pipe_barrier();  // bt_l ready
+  if (range_lo < range_hi) issue_k(range_lo, img0);
+
+  int cur = 0;
+  for (int base = range_lo; base < range_hi; base += PD_KT, cur ^= 1) {
+    short* const X = cur ? img1 : img0;
+    short* const Y = cur ? img0 : img1;
+    pipe_barrier_vm<0>();  // K(base) landed in X front
+
+    // ---- S from fp8 K (convert per fragment with packed cvt VOPs)
+    f32x4_t s = {0.f, 0.f, 0.f, 0.f};
+    {
+      const int key = slab * 16 + col;
+      const char* krow = reinterpret_cast<const char*>(X) + key * D;
+#pragma unroll
+      for (int ks = 0; ks < KS; ++ks) {
+        const int d8 = ks * 32 + kgrp * 8;
+        const int c16 = d8 >> 4;
+        const int lds_col = ((c16 ^ (key & 7)) & (G16B - 1)) * 16 + (d8 & 15);
+        const uint2 raw = *reinterpret_cast<const uint2*>(krow + lds_col);
+        const bf16x8_t bfrag = fp8x8_to_bf16(raw.x, raw.y);
+        s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks], bfrag, s, 0, 0, 0);
+      }
+    }
+    float sv[4];
+    {
+      const int key = base + slab * 16 + col;
+      const bool dead = key >= L || key < start;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        float x = s[reg] * scale;
+        if (softcap > 0.f) x = tanhf(x / softcap) * softcap;
+        sv[reg] = dead ? -1e30f : x;
+        float m = sv[reg];
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) m = fmaxf(m, __shfl_xor(m, off, WAVE));
+        if (col == 0 && wid < SLABS) mpart[slab * 16 + kgrp * 4 + reg] = m;
+      }
+    }
+    pipe_barrier();  // S reads of X-front done; mparts visible
+
+    // ---- issue V(base) raw -> Y back, K(next) -> Y front
+    const int next = base + PD_KT;
+    const bool prefetch = next < range_hi;
+    issue_v_raw(base, Y);
+    if (prefetch) issue_k(next, Y);
+
+    // ---- softmax combine + P
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int row = kgrp * 4 + reg;
+      float m_tile = mpart[row];
+#pragma unroll
+      for (int w = 1; w < SLABS; ++w) m_tile = fmaxf(m_tile, mpart[w * 16 + row]);
+      const float m_new = fmaxf(m_regs[reg], m_tile);
+      const float alpha = (m_new > -1e30f) ? __expf(m_regs[reg] - m_new) : 1.f;
+      const float pe = (m_new > -1e30f && sv[reg] > -1e29f)
+                           ? __expf(sv[reg] - m_new) : 0.f;
+      float lsum = pe;
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) lsum += __shfl_xor(lsum, off, WAVE);
+      l_regs[reg] = l_regs[reg] * alpha + lsum;
+      m_regs[reg] = m_new;
+      if (wid == 0 && col == 0) alpha_s[row] = alpha;
+      p_lds2[row * PD_KT + slab * 16 + col] =
+          __bfloat16_as_short(__float2bfloat16(pe));
+    }
+    if (prefetch) {
+      pipe_barrier_vm<NI_K>();  // V raw landed; K(next) stays in flight
+    } else {
+      pipe_barrier_vm<0>();
+    }
+
+    // ---- convert pass: Y back (raw fp8) -> X bf16 tr16 image
+    {
+      const char* const raw = reinterpret_cast<const char*>(Y) + RAW_BYTES;
+      constexpr int CPK = D / 8;          // 8-elem granules per key row
+      constexpr int NG = PD_KT * CPK;     // 512 at D=256
+#pragma unroll
+      for (int j = 0; j < NG / NT; ++j) {
+        const int c = tid + j * NT;
+        const int key = c / CPK;
+        const int d8 = (c % CPK) * 8;
+        const int dtile = d8 / 16, col0 = d8 & 15;
+        const int qq = (key & 31) >> 2;
+        const int bpos = ((qq & 1) << 2) + (qq >> 1);
+        const int pdst = ((key >> 5) * (D / 16) + dtile) * 528 + bpos * 64 +
+                         (key & 3) * 16 + col0;
+        const uint2 rr = *reinterpret_cast<const uint2*>(raw + key * D + d8);
+        *reinterpret_cast<bf16x8_t*>(&X[pdst]) = fp8x8_to_bf16(rr.x, rr.y);
+      }
+    }
+    pipe_barrier();  // image + P + alpha visible
+
+    // ---- PV from the bf16 tr16 image
+    const float alpha_q = alpha_s[col];
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
+      ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
+      const int dtile = (wid * D4) / 16 + dt;
+#pragma unroll
+      for (int ks = 0; ks < PD_KT / 32; ++ks) {
+        const int sub = (ks * (D / 16) + dtile) * 528 + lane * 4;
+        bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_bf16x4*)&X[sub]);
+        bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_bf16x4*)&X[sub + 4 * 64]);
+        bf16x8_t a;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          a[j] = __bfloat16_as_short((__hip_bfloat16)lo[j]);
+          a[4 + j] = __bfloat16_as_short((__hip_bfloat16)hi[j]);
+        }
+        bf16x8_t bb = *reinterpret_cast<const bf16x8_t*>(
+            &p_lds2[col * PD_KT + ks * 32 + kgrp * 8]);
+        ot[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bb, ot[dt], 0, 0, 0);
+      }
+    }
+    // next loop-top barrier (vm0) protects X/Y rotation
+  }
+
+  // ---- merge per-slab l, then store (same as the bf16 pipe kernel)
+  if (col == 0 && wid < SLABS) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) l_part[slab * 16 + kgrp * 4 + reg] = l_regs[reg];
+  }
+  pipe_barrier();
+  float l_tot = l_part[col];
+#pragma unroll
+  for (int w = 1; w < SLABS; ++w) l_tot += l_part[w * 16 + col];
+  if (nsplit > 1) {
+    float* slot = scratch +
+        ((((long)b * num_kv_heads + kh) * nsplit + blockIdx.z) * G) * (D + 2);
+    if (col < G) {
+      float* row = slot + (long)col * (D + 2);
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        const int dim0 = wid * D4 + dt * 16 + kgrp * 4;
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) row[dim0 + reg] = ot[dt][reg];
+      }
+    }
+    if (wid == 0 && col == 0) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int r = kgrp * 4 + reg;
+        if (r < G) {
+          float* row = slot + (long)r * (D + 2);
+          row[D] = m_regs[reg];
+          float lt = l_part[r];
+          for (int w = 1; w < SLABS; ++w) lt += l_part[w * 16 + r];
+          row[D + 1] = lt;
+        }
+      }
+    }
+    return;
+  }
+  const float inv = (l_tot > 0.f) ? 1.0f / l_tot : 0.f;
+  if (col < G) {
+    __hip_bfloat16* op = out + (long)b * out_stride + (long)(kh * G + col) * D;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      const int dim0 = wid * D4 + dt * 16 + kgrp * 4;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg)
+        op[dim0 + reg] = __float2bfloat16(ot[dt][reg] * inv);
+    }
+  }
+}
+
 // ------------------------------------------- ring-buffered MFMA decode --
 // Variant of paged_decode_pipe_kernel with SEPARATE K and V slot pairs and
 // 32-key chunks: both V(i) and K(i+1) are issued at the LOOP TOP, so every

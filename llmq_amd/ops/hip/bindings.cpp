@@ -262,6 +262,30 @@ void launch_decode(at::Tensor& out, const at::Tensor& q, const at::Tensor& kc,
       // select — guide §5 trap 4c).
       const char* pe = getenv("LLMQ_DECODE_PIPE");
       const int pipe_kt = pe ? atoi(pe) : 64;
+      if (fp8_cache && pipe_kt > 0 && max_blocks <= PD_MAX_BT) {
+        // fp8 glds pipeline: raw-fp8 staging + native packed cvt VOPs
+        auto lf8 = [&]<int HD>() {
+          hipLaunchKernelGGL((paged_decode_pipe_fp8_kernel<HD, 8>), sgrid,
+                             dim3(8 * 64), 0, stream(),
+                             reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                             reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                             reinterpret_cast<const __hip_fp8_e4m3*>(kc.data_ptr()),
+                             reinterpret_cast<const __hip_fp8_e4m3*>(vc.data_ptr()),
+                             bt.data_ptr<int>(), cl.data_ptr<int>(), scratch_ptr,
+                             H, KVH, bs, max_blocks, (float)scale,
+                             (float)softcap, (int)window, q.stride(0),
+                             out.stride(0));
+          if (nsplit > 1) {
+            hipLaunchKernelGGL((decode_splitkv_merge_kernel<HD>),
+                               dim3(B, KVH, G), dim3(64), 0, stream(),
+                               reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                               scratch_ptr, KVH, G, nsplit, out.stride(0));
+          }
+        };
+        if (D == 128) lf8.template operator()<128>();
+        else lf8.template operator()<256>();
+        return;
+      }
       if (!fp8_cache && pipe_kt > 0 && max_blocks <= PD_MAX_BT) {
         auto lp = [&]<int HD, int KT>() {
           hipLaunchKernelGGL((paged_decode_pipe_kernel<HD, 8, KT>), sgrid,
