@@ -633,6 +633,36 @@ class TestFp8KVCache:
         assert len(outs) == 2
         assert eng.kv_cache.k[0].dtype == torch.float8_e4m3fn
 
+    def test_fp8_kv_with_chunked_prefill(self):
+        """Long prompt + fp8 cache: the chunk continuation GATHERS past K/V
+        from the quantised cache (dequantised via .to(dtype)). A dtype
+        reinterpret bug there produces NaN logits → degenerate token-0
+        spam; assert the output is sane and matches the bf16-cache engine's
+        shape of behaviour (token ids legal, not constant)."""
+        from llmq_amd.engine.config import EngineConfig
+        from llmq_amd.engine.engine import LLMEngine
+        from llmq_amd.engine.sampling_params import SamplingParams
+
+        eng = LLMEngine(EngineConfig(
+            model="tiny-llama-d128", max_num_seqs=2, max_model_len=1024,
+            load_weights=False, num_kv_blocks=512, kv_cache_dtype="fp8",
+            max_prefill_tokens=256,  # forces a 600-token prompt into chunks
+        ))
+        torch.manual_seed(11)
+        ids = torch.randint(0, eng.spec.vocab_size, (600,)).tolist()
+        eng.add_request("long", prompt_token_ids=ids,
+                        params=SamplingParams(temperature=0.0, max_tokens=12,
+                                              ignore_eos=True))
+        toks = []
+        steps = 0
+        while eng.has_unfinished() and steps < 64:
+            for out in eng.step():
+                toks.extend(out.new_token_ids)
+            steps += 1
+        assert len(toks) == 12
+        assert all(0 <= t < eng.spec.vocab_size for t in toks)
+        assert len(set(toks)) > 1, f"degenerate output {toks}"
+
 
 class TestChunkedPrefillAttention:
     @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
