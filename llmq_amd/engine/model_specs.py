@@ -155,6 +155,18 @@ PRESETS = {
     "qwen2.5-72b": _qwen2("qwen2.5-72b", 152064, 8192, 29568, 80, 64, 8,
                           max_position_embeddings=32768,
                           eos_token_id=151645, bos_token_id=151643),
+    "qwen2.5-32b": _qwen2("qwen2.5-32b", 152064, 5120, 27648, 64, 40, 8,
+                          max_position_embeddings=32768,
+                          eos_token_id=151645, bos_token_id=151643),
+    # Llama-3.1-8B / 70B (same layout; 70B pairs with TP=8 or fits 1 GPU)
+    "llama-3.1-8b": _llama("llama-3.1-8b", 128256, 4096, 14336, 32, 32, 8,
+                           head_dim=128, rope_theta=500000.0,
+                           max_position_embeddings=131072,
+                           eos_token_id=128001, bos_token_id=128000),
+    "llama-3.1-70b": _llama("llama-3.1-70b", 128256, 8192, 28672, 80, 64, 8,
+                            head_dim=128, rope_theta=500000.0,
+                            max_position_embeddings=131072,
+                            eos_token_id=128001, bos_token_id=128000),
     # Mistral-7B (llama layout, 4096-token sliding window on every layer)
     "mistral-7b": _llama("mistral-7b", 32000, 4096, 14336, 32, 32, 8,
                          rope_theta=10000.0, max_position_embeddings=32768,
@@ -162,6 +174,11 @@ PRESETS = {
     # Gemma-2 (Tower-Plus-9B is built on Gemma-2-9B)
     "gemma-2-9b": _gemma2("gemma-2-9b", 256000, 3584, 14336, 42, 16, 8, 256,
                           max_position_embeddings=8192),
+    # 27b: query_pre_attn_scalar = hidden/heads = 144 (NOT head_dim as on
+    # 9b) -> attention scale 1/sqrt(144)
+    "gemma-2-27b": _gemma2("gemma-2-27b", 256000, 4608, 36864, 46, 32, 16, 128,
+                           attn_scale=144.0 ** -0.5,
+                           max_position_embeddings=8192),
 }
 
 # Aliases for the model names the reference's production configs use.
@@ -178,6 +195,12 @@ ALIASES = {
     "meta-llama/llama-3.2-3b-instruct": "llama-3.2-3b",
     "google/gemma-2-9b": "gemma-2-9b",
     "google/gemma-2-9b-it": "gemma-2-9b",
+    "google/gemma-2-27b": "gemma-2-27b",
+    "meta-llama/llama-3.1-8b-instruct": "llama-3.1-8b",
+    "llama-3.1-8b-instruct": "llama-3.1-8b",
+    "meta-llama/llama-3.1-70b-instruct": "llama-3.1-70b",
+    "llama-3.1-70b-instruct": "llama-3.1-70b",
+    "qwen/qwen2.5-32b-instruct": "qwen2.5-32b",
 }
 
 
