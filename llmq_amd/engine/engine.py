@@ -156,6 +156,22 @@ class LLMEngine:
             torch.cuda.Stream(device=self.device) if self._overlap_mixed else None
         )
         self.runner.capture_graphs()
+        if self.device.type == "cuda":
+            free, _total = torch.cuda.mem_get_info(self.device)
+            if free < (12 << 30):
+                # Prefill activations at max_prefill_tokens rows (plus
+                # hipBLASLt workspaces) live OUTSIDE the KV budget; with
+                # single-digit GB of headroom the allocator thrashes into
+                # retry storms (observed with 72B: 144 GB weights + a full
+                # KV pool + 16k-row prefill segments). Lower
+                # gpu_memory_utilization, max_model_len·max_num_seqs (the
+                # KV cap) or max_prefill_tokens.
+                logger.warning(
+                    "only %.1f GB of HBM headroom left after weights+KV+"
+                    "graphs — large prefill segments may thrash the "
+                    "allocator (reduce max_prefill_tokens=%d or the KV cap)",
+                    free / 2**30, config.max_prefill_tokens,
+                )
         self._seqs: Dict[str, Sequence] = {}
         self._prefill_done_at: Dict[str, float] = {}
         self.steps = 0
