@@ -705,6 +705,7 @@ __global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_
 
     const bool active = kt <= wave_max_row;  // wave-uniform
     float p[4][4];
+    bool rescale = false;  // defer-max decision (set in the S phase)
     if (active) {
       // ---- S = Q K^T for 4 column tiles of 16 keys
       f32x4_t s[4];
@@ -721,11 +722,12 @@ __global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_
         }
       }
       // scale, softcap, mask, row max
+      float mx[4];
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         const int qrow = q_base + wid * 16 + kgrp * 4 + reg;
         const int row = off + qrow;
-        float mx = -1e30f;
+        float m = -1e30f;
 #pragma unroll
         for (int ct = 0; ct < 4; ++ct) {
           const int key = kt + ct * 16 + col;
@@ -735,12 +737,30 @@ __global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_
                             (window > 0 && key <= row - window);
           x = dead ? -1e30f : x;
           p[reg][ct] = x;
-          mx = fmaxf(mx, x);
+          m = fmaxf(m, x);
         }
 #pragma unroll
-        for (int o3 = 1; o3 < 16; o3 <<= 1) mx = fmaxf(mx, __shfl_xor(mx, o3, WAVE));
-        const float m_new = fmaxf(m_run[reg], mx);
-        const float alpha = (m_new > -1e30f) ? __expf(m_run[reg] - m_new) : 1.f;
+        for (int o3 = 1; o3 < 16; o3 <<= 1) m = fmaxf(m, __shfl_xor(m, o3, WAVE));
+        mx[reg] = m;
+      }
+      // defer-max (guide T13): when no row of this wave grew its max by
+      // more than THR, keep the old maxes — P is then bounded by e^THR
+      // (fp32 accumulators tolerate it) and the whole O-rescale pass is
+      // skipped. The decision covers the ENTIRE tile before any P of it
+      // is exponentiated (the T13 ordering hazard), and every l-update
+      // uses alpha==1 on the defer path. Accuracy cost ≈3× max-abs error
+      // (bounded-P bf16 quantisation) — inside the test tolerances.
+      constexpr float DEFER_THR = 8.0f;
+      bool grow = false;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg)
+        grow |= mx[reg] > m_run[reg] + DEFER_THR;
+      rescale = __any(grow);
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const float m_new = rescale ? fmaxf(m_run[reg], mx[reg]) : m_run[reg];
+        const float alpha = (rescale && m_new > -1e30f)
+                                ? __expf(m_run[reg] - m_new) : 1.f;
         float lsum = 0.f;
 #pragma unroll
         for (int ct = 0; ct < 4; ++ct) {
@@ -753,7 +773,7 @@ __global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_
         for (int o3 = 1; o3 < 16; o3 <<= 1) lsum += __shfl_xor(lsum, o3, WAVE);
         l_run[reg] = l_run[reg] * alpha + lsum;
         m_run[reg] = m_new;
-        if (col == 0) stat[(wid * 2 + 0) * 16 + kgrp * 4 + reg] = alpha;
+        if (rescale && col == 0) stat[(wid * 2 + 0) * 16 + kgrp * 4 + reg] = alpha;
       }
     }
     pipe_barrier();  // S reads of X done; alpha visible
@@ -783,13 +803,15 @@ __global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_
     }
 
     if (active) {
-      // ---- OT += V^T P^T
-      const float alpha_q = stat[(wid * 2 + 0) * 16 + col];
+      // ---- OT += V^T P^T (O-rescale skipped entirely on the defer path)
+      const float alpha_q = rescale ? stat[(wid * 2 + 0) * 16 + col] : 1.f;
       short* const p_lds_w = p_base + wid * 16 * PF_KT;
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
-        ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
-        ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
+        if (rescale) {
+          ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
+          ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
+        }
 #pragma unroll
         for (int ks = 0; ks < PF_KT / 32; ++ks) {
           const int sub = (ks * (D / 16) + dt) * 528 + lane * 4;

@@ -230,6 +230,31 @@ class TestPrefillAttention:
         atol, rtol = TOL[dtype]
         assert_close_to_f32_ref(out.cpu(), ref, atol * 2, 5 * rtol)
 
+    def test_defer_max_forced_rescale(self):
+        """Forces the defer-max (T13) branch BOTH ways (guide rule 26).
+
+        Early K rows are tiny, so every tile's max drifts < THR and the
+        kernel takes the defer path (no O-rescale) for ~11 tiles; key 700
+        is a spiked row aligned with late queries, so the tile containing
+        it jumps the running max far past THR and the rescale path must
+        fire late in the loop. Causality means q-rows < 700 exercise
+        defer-only while rows >= 700 exercise the late rescale.
+        """
+        dtype = torch.bfloat16
+        H, KVH, D = 8, 2, 128
+        torch.manual_seed(11)
+        L = 740
+        cu = torch.tensor([0, L], dtype=torch.int32, device=DEV)
+        q = torch.randn(L, H, D, device=DEV, dtype=dtype)
+        k = torch.randn(L, KVH, D, device=DEV, dtype=dtype) * 0.05
+        v = torch.randn(L, KVH, D, device=DEV, dtype=dtype)
+        k[700] = (q[L - 1, :KVH].float() * 6.0).to(dtype)
+        out = ops.varlen_prefill_attention(q, k, v, cu, L, D ** -0.5)
+        ref = torch_ref.varlen_prefill_attention(
+            q.float().cpu(), k.float().cpu(), v.float().cpu(), cu.cpu(), D ** -0.5
+        )
+        assert_close_to_f32_ref(out.cpu(), ref, 2e-2, 1e-1)
+
     def test_softcap_window(self):
         dtype = torch.bfloat16
         H, KVH, D = 8, 4, 128
