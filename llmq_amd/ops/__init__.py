@@ -31,7 +31,11 @@ def _try_load_extension() -> None:
         return
     _EXT_TRIED = True
     here = Path(__file__).parent
-    candidates = sorted(here.glob("_hip_ops*.so"))
+    override = os.environ.get("LLMQ_OPS_SO")
+    if override:
+        candidates = [Path(override)]
+    else:
+        candidates = sorted(here.glob("_hip_ops*.so"))
     if not candidates:
         _EXT_ERROR = (
             f"HIP extension not built (no _hip_ops*.so under {here}). "

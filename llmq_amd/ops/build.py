@@ -68,6 +68,38 @@ def build(verbose: bool = False, force: bool = False) -> Path:
     return OUT
 
 
+def build_ab_defer_off(verbose: bool = False) -> Path:
+    """A/B artifact: same kernels with -DLLMQ_DEFER_OFF (pre-T13 softmax
+    rescale every tile). Output name deliberately does NOT match the
+    `_hip_ops*.so` auto-load glob — select it with LLMQ_OPS_SO=<path>."""
+    out = HERE / "_ab_defer_off.so"
+    if out.is_file() and out.stat().st_mtime >= _sources_mtime():
+        return out
+    SRC.touch()
+    os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+    os.environ.setdefault("MAX_JOBS", "8")
+    from torch.utils.cpp_extension import load
+
+    bdir = HERE / "build_ab"
+    shutil.rmtree(bdir, ignore_errors=True)
+    bdir.mkdir(exist_ok=True)
+    load(
+        name="llmq_amd_hip_ops_ab",
+        sources=[str(SRC)],
+        extra_cflags=["-O3", "-std=c++20", "-DLLMQ_DEFER_OFF"],
+        extra_cuda_cflags=["-O3", "-std=c++20", "-DLLMQ_DEFER_OFF"],
+        build_directory=str(bdir),
+        is_python_module=False,
+        verbose=verbose,
+    )
+    built = next(bdir.glob("*.so"))
+    shutil.copy2(built, out)
+    return out
+
+
 if __name__ == "__main__":
-    out = build(verbose="--verbose" in sys.argv, force="--force" in sys.argv)
+    if "--ab-defer-off" in sys.argv:
+        out = build_ab_defer_off(verbose="--verbose" in sys.argv)
+    else:
+        out = build(verbose="--verbose" in sys.argv, force="--force" in sys.argv)
     print(f"built {out}")

@@ -755,7 +755,11 @@ __global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg)
         grow |= mx[reg] > m_run[reg] + DEFER_THR;
+#ifdef LLMQ_DEFER_OFF
+      rescale = true; (void)grow;  // A/B build: pre-T13 behavior
+#else
       rescale = __any(grow);
+#endif
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         const float m_new = rescale ? fmaxf(m_run[reg], mx[reg]) : m_run[reg];
@@ -1396,14 +1400,34 @@ __global__ __launch_bounds__(NW * WAVE, (PD_KT <= 64 ? 4 : 2)) void paged_decode
     if (prefetch) issue_k(next, Y);
 
     // ---- combine maxes, build P, update l (redundant on every wave)
+    float m_tile[4];
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       const int row = kgrp * 4 + reg;
-      float m_tile = mpart[row];
+      float m = mpart[row];
 #pragma unroll
-      for (int w = 1; w < SLABS; ++w) m_tile = fmaxf(m_tile, mpart[w * 16 + row]);
-      const float m_new = fmaxf(m_regs[reg], m_tile);
-      const float alpha = (m_new > -1e30f) ? __expf(m_regs[reg] - m_new) : 1.f;
+      for (int w = 1; w < SLABS; ++w) m = fmaxf(m, mpart[w * 16 + row]);
+      m_tile[reg] = m;
+    }
+    // defer-max (guide T13): every wave computes m_regs/m_tile from the
+    // same mpart values with the same op order, so the ballot below is
+    // workgroup-uniform with no extra LDS flag. On the defer path
+    // alpha==1 exactly: the alpha_s store AND the whole OT-rescale pass
+    // are skipped; P is bounded by e^THR (l/OT accumulate in fp32).
+    bool grow = false;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) grow |= m_tile[reg] > m_regs[reg] + 8.0f;
+#ifdef LLMQ_DEFER_OFF
+    const bool rescale = true; (void)grow;  // A/B build: pre-T13 behavior
+#else
+    const bool rescale = __any(grow);
+#endif
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int row = kgrp * 4 + reg;
+      const float m_new = rescale ? fmaxf(m_regs[reg], m_tile[reg]) : m_regs[reg];
+      const float alpha = (rescale && m_new > -1e30f)
+                              ? __expf(m_regs[reg] - m_new) : 1.f;
       const float pe = (m_new > -1e30f && sv[reg] > -1e29f)
                            ? __expf(sv[reg] - m_new) : 0.f;
       float lsum = pe;
@@ -1411,7 +1435,7 @@ __global__ __launch_bounds__(NW * WAVE, (PD_KT <= 64 ? 4 : 2)) void paged_decode
       for (int off = 1; off < 16; off <<= 1) lsum += __shfl_xor(lsum, off, WAVE);
       l_regs[reg] = l_regs[reg] * alpha + lsum;
       m_regs[reg] = m_new;
-      if (wid == 0 && col == 0) alpha_s[row] = alpha;
+      if (rescale && wid == 0 && col == 0) alpha_s[row] = alpha;
       p_lds2[row * PD_KT + slab * 16 + col] =
           __bfloat16_as_short(__float2bfloat16(pe));
     }
@@ -1422,11 +1446,13 @@ __global__ __launch_bounds__(NW * WAVE, (PD_KT <= 64 ? 4 : 2)) void paged_decode
     }
 
     // ---- OT[dims,16q] += V^T P^T over this wave's dim slab
-    const float alpha_q = alpha_s[col];
+    const float alpha_q = rescale ? alpha_s[col] : 1.f;
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt) {
-      ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
-      ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
+      if (rescale) {
+        ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
+        ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
+      }
       const int dtile = (wid * D4) / 16 + dt;
 #pragma unroll
       for (int ks = 0; ks < PD_KT / 32; ++ks) {
@@ -1708,14 +1734,34 @@ __global__ __launch_bounds__(NW * WAVE, 4) void paged_decode_pipe_fp8_kernel(
     if (prefetch) issue_k(next, Y);
 
     // ---- softmax combine + P
+    float m_tile[4];
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       const int row = kgrp * 4 + reg;
-      float m_tile = mpart[row];
+      float m = mpart[row];
 #pragma unroll
-      for (int w = 1; w < SLABS; ++w) m_tile = fmaxf(m_tile, mpart[w * 16 + row]);
-      const float m_new = fmaxf(m_regs[reg], m_tile);
-      const float alpha = (m_new > -1e30f) ? __expf(m_regs[reg] - m_new) : 1.f;
+      for (int w = 1; w < SLABS; ++w) m = fmaxf(m, mpart[w * 16 + row]);
+      m_tile[reg] = m;
+    }
+    // defer-max (guide T13): every wave computes m_regs/m_tile from the
+    // same mpart values with the same op order, so the ballot below is
+    // workgroup-uniform with no extra LDS flag. On the defer path
+    // alpha==1 exactly: the alpha_s store AND the whole OT-rescale pass
+    // are skipped; P is bounded by e^THR (l/OT accumulate in fp32).
+    bool grow = false;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) grow |= m_tile[reg] > m_regs[reg] + 8.0f;
+#ifdef LLMQ_DEFER_OFF
+    const bool rescale = true; (void)grow;  // A/B build: pre-T13 behavior
+#else
+    const bool rescale = __any(grow);
+#endif
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int row = kgrp * 4 + reg;
+      const float m_new = rescale ? fmaxf(m_regs[reg], m_tile[reg]) : m_regs[reg];
+      const float alpha = (rescale && m_new > -1e30f)
+                              ? __expf(m_regs[reg] - m_new) : 1.f;
       const float pe = (m_new > -1e30f && sv[reg] > -1e29f)
                            ? __expf(sv[reg] - m_new) : 0.f;
       float lsum = pe;
@@ -1723,7 +1769,7 @@ __global__ __launch_bounds__(NW * WAVE, 4) void paged_decode_pipe_fp8_kernel(
       for (int off = 1; off < 16; off <<= 1) lsum += __shfl_xor(lsum, off, WAVE);
       l_regs[reg] = l_regs[reg] * alpha + lsum;
       m_regs[reg] = m_new;
-      if (wid == 0 && col == 0) alpha_s[row] = alpha;
+      if (rescale && wid == 0 && col == 0) alpha_s[row] = alpha;
       p_lds2[row * PD_KT + slab * 16 + col] =
           __bfloat16_as_short(__float2bfloat16(pe));
     }
@@ -1755,11 +1801,13 @@ __global__ __launch_bounds__(NW * WAVE, 4) void paged_decode_pipe_fp8_kernel(
     pipe_barrier();  // image + P + alpha visible
 
     // ---- PV from the bf16 tr16 image
-    const float alpha_q = alpha_s[col];
+    const float alpha_q = rescale ? alpha_s[col] : 1.f;
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt) {
-      ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
-      ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
+      if (rescale) {
+        ot[dt][0] *= alpha_q; ot[dt][1] *= alpha_q;
+        ot[dt][2] *= alpha_q; ot[dt][3] *= alpha_q;
+      }
       const int dtile = (wid * D4) / 16 + dt;
 #pragma unroll
       for (int ks = 0; ks < PD_KT / 32; ++ks) {
