@@ -68,11 +68,12 @@ def build(verbose: bool = False, force: bool = False) -> Path:
     return OUT
 
 
-def build_ab_defer_off(verbose: bool = False) -> Path:
-    """A/B artifact: same kernels with -DLLMQ_DEFER_OFF (pre-T13 softmax
-    rescale every tile). Output name deliberately does NOT match the
-    `_hip_ops*.so` auto-load glob — select it with LLMQ_OPS_SO=<path>."""
-    out = HERE / "_ab_defer_off.so"
+def build_ab_defer_on(verbose: bool = False) -> Path:
+    """A/B artifact: same kernels with -DLLMQ_DEFER_ON (the T13 defer-max
+    experiment — measured a LOSS, see profiles/r2_step5). Output name
+    deliberately does NOT match the `_hip_ops*.so` auto-load glob —
+    select it with LLMQ_OPS_SO=<path>."""
+    out = HERE / "_ab_defer_on.so"
     if out.is_file() and out.stat().st_mtime >= _sources_mtime():
         return out
     SRC.touch()
@@ -86,8 +87,8 @@ def build_ab_defer_off(verbose: bool = False) -> Path:
     load(
         name="llmq_amd_hip_ops_ab",
         sources=[str(SRC)],
-        extra_cflags=["-O3", "-std=c++20", "-DLLMQ_DEFER_OFF"],
-        extra_cuda_cflags=["-O3", "-std=c++20", "-DLLMQ_DEFER_OFF"],
+        extra_cflags=["-O3", "-std=c++20", "-DLLMQ_DEFER_ON"],
+        extra_cuda_cflags=["-O3", "-std=c++20", "-DLLMQ_DEFER_ON"],
         build_directory=str(bdir),
         is_python_module=False,
         verbose=verbose,
@@ -98,8 +99,8 @@ def build_ab_defer_off(verbose: bool = False) -> Path:
 
 
 if __name__ == "__main__":
-    if "--ab-defer-off" in sys.argv:
-        out = build_ab_defer_off(verbose="--verbose" in sys.argv)
+    if "--ab-defer-on" in sys.argv:
+        out = build_ab_defer_on(verbose="--verbose" in sys.argv)
     else:
         out = build(verbose="--verbose" in sys.argv, force="--force" in sys.argv)
     print(f"built {out}")

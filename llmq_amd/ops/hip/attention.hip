@@ -755,10 +755,10 @@ __global__ __launch_bounds__(256, (HEAD_DIM <= 128 ? 4 : 2)) void flash_prefill_
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg)
         grow |= mx[reg] > m_run[reg] + DEFER_THR;
-#ifdef LLMQ_DEFER_OFF
-      rescale = true; (void)grow;  // A/B build: pre-T13 behavior
-#else
+#ifdef LLMQ_DEFER_ON
       rescale = __any(grow);
+#else
+      rescale = true; (void)grow;  // defer-max measured a loss; see above
 #endif
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
@@ -1417,10 +1417,14 @@ __global__ __launch_bounds__(NW * WAVE, (PD_KT <= 64 ? 4 : 2)) void paged_decode
     bool grow = false;
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) grow |= m_tile[reg] > m_regs[reg] + 8.0f;
-#ifdef LLMQ_DEFER_OFF
-    const bool rescale = true; (void)grow;  // A/B build: pre-T13 behavior
-#else
+#ifdef LLMQ_DEFER_ON
     const bool rescale = __any(grow);
+#else
+    // Measured: the T13 defer guard LOSES here (decode 0.458 vs 0.442 ms,
+    // prefill 0.282 vs 0.261 ms same-box A/B) — the branch around the
+    // O-rescale disturbs the pipelined PV scheduling more than the skipped
+    // VALU pays. Kept compilable behind -DLLMQ_DEFER_ON for re-testing.
+    const bool rescale = true; (void)grow;
 #endif
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
@@ -1751,10 +1755,14 @@ __global__ __launch_bounds__(NW * WAVE, 4) void paged_decode_pipe_fp8_kernel(
     bool grow = false;
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) grow |= m_tile[reg] > m_regs[reg] + 8.0f;
-#ifdef LLMQ_DEFER_OFF
-    const bool rescale = true; (void)grow;  // A/B build: pre-T13 behavior
-#else
+#ifdef LLMQ_DEFER_ON
     const bool rescale = __any(grow);
+#else
+    // Measured: the T13 defer guard LOSES here (decode 0.458 vs 0.442 ms,
+    // prefill 0.282 vs 0.261 ms same-box A/B) — the branch around the
+    // O-rescale disturbs the pipelined PV scheduling more than the skipped
+    // VALU pays. Kept compilable behind -DLLMQ_DEFER_ON for re-testing.
+    const bool rescale = true; (void)grow;
 #endif
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
