@@ -127,8 +127,21 @@ def main() -> None:
         engine.add_request(f"bench-{req_tag}-{i}", prompt_token_ids=ids, params=params)
 
     t_prefill = time.perf_counter()
+    stalled = 0
     while engine.scheduler.num_waiting > 0 or engine.scheduler.prefilling:
+        before = sum(sq.num_tokens for sq in engine.scheduler.running)
         engine.step()  # admission steps (prefill + ride-along decode)
+        after = sum(sq.num_tokens for sq in engine.scheduler.running)
+        # Fail loudly instead of spinning silently if admission makes no
+        # progress (e.g. the KV pool can never seat the requested batch
+        # and preemption churns running -> waiting forever).
+        stalled = stalled + 1 if after <= before and engine.scheduler.num_waiting > 0 else 0
+        if stalled >= 50:
+            raise RuntimeError(
+                f"admission stalled: {engine.scheduler.num_waiting} waiting, "
+                f"{len(engine.scheduler.running)} running, {after} resident tokens "
+                f"— KV pool too small for --batch x --max-model-len?"
+            )
     prefill_s = time.perf_counter() - t_prefill
     prefill_tokens = batch * prompt_len
     # mixed steps decode already-admitted seqs while later ones prefill, so
