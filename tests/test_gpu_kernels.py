@@ -668,13 +668,17 @@ class TestFp8KVCache:
         from llmq_amd.engine.engine import LLMEngine
         from llmq_amd.engine.sampling_params import SamplingParams
 
+        # tiny specs cap max_model_len at 512 positions — keep prompt+output
+        # inside that (the original 600-token prompt was silently truncated
+        # to 511 and the request length-finished after ONE token).
         eng = LLMEngine(EngineConfig(
-            model="tiny-llama-d128", max_num_seqs=2, max_model_len=1024,
+            model="tiny-llama-d128", max_num_seqs=2, max_model_len=512,
             load_weights=False, num_kv_blocks=512, kv_cache_dtype="fp8",
-            max_prefill_tokens=256,  # forces a 600-token prompt into chunks
+            max_prefill_tokens=128,  # forces a 400-token prompt into chunks
         ))
+        assert eng.max_model_len == 512
         torch.manual_seed(11)
-        ids = torch.randint(0, eng.spec.vocab_size, (600,)).tolist()
+        ids = torch.randint(0, eng.spec.vocab_size, (400,)).tolist()
         eng.add_request("long", prompt_token_ids=ids,
                         params=SamplingParams(temperature=0.0, max_tokens=12,
                                               ignore_eos=True))
@@ -684,7 +688,7 @@ class TestFp8KVCache:
             for out in eng.step():
                 toks.extend(out.new_token_ids)
             steps += 1
-        assert len(toks) == 12
+        assert len(toks) == 12, f"expected 12 decode tokens, got {toks}"
         assert all(0 <= t < eng.spec.vocab_size for t in toks)
         # The dtype-reinterpret bug this guards against produces all-NaN
         # logits, whose argmax is constant token 0. A random-init greedy
