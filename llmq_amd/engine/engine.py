@@ -155,6 +155,10 @@ class LLMEngine:
         self._decode_stream = (
             torch.cuda.Stream(device=self.device) if self._overlap_mixed else None
         )
+        # diagnostics: run the overlap's decode EAGER (no graph replay), or
+        # serialise the two streams (two-stream plumbing, zero concurrency)
+        self._overlap_eager = os.environ.get("LLMQ_OVERLAP_EAGER") == "1"
+        self._overlap_sync = os.environ.get("LLMQ_OVERLAP_SYNC") == "1"
         self.runner.capture_graphs()
         if self.device.type == "cuda":
             free, _total = torch.cuda.mem_get_info(self.device)
@@ -295,7 +299,10 @@ class LLMEngine:
                     with torch.cuda.stream(ds):
                         # .clone(): the fused sampler returns a VIEW of a
                         # persistent output buffer
-                        d_tokens = self.runner.execute_decode(dseqs).clone()
+                        d_tokens = self.runner.execute_decode(
+                            dseqs, force_eager=self._overlap_eager).clone()
+                    if self._overlap_sync:
+                        ds.synchronize()
                     p_tokens = self.runner.execute_prefill(
                         pseqs, batch.chunks, buf_name="prefill"
                     )

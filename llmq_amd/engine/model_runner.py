@@ -272,7 +272,7 @@ class ModelRunner:
     # -- decode ----------------------------------------------------------
 
     @torch.no_grad()
-    def execute_decode(self, seqs: List[Sequence]) -> torch.Tensor:
+    def execute_decode(self, seqs: List[Sequence], force_eager: bool = False) -> torch.Tensor:
         dev = self.device
         B = len(seqs)
         bs = self.block_size
@@ -281,7 +281,7 @@ class ModelRunner:
         slots = [s.block_table[p // bs] * bs + (p % bs) for s, p in zip(seqs, pos)]
         ctx = [s.num_tokens for s in seqs]
 
-        bucket = self._bucket_for(B) if self.use_graphs else None
+        bucket = self._bucket_for(B) if (self.use_graphs and not force_eager) else None
         if bucket is not None and bucket in self._graphs:
             return self._decode_with_graph(seqs, bucket, ids, pos, slots, ctx)
 
