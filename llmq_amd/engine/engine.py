@@ -147,10 +147,26 @@ class LLMEngine:
         )
         # Mixed-step overlap: decode graph on a side stream ∥ prefill on the
         # default stream (disjoint KV slots; separate sampler buffers).
+        # gemma-2-27b (the only gemma2 with head_dim 128) deadlocks the GPU
+        # when its decode and prefill kernels run concurrently: GPU busy
+        # 100 %, zero memory traffic, mid-admission, batch >= 128. Bisected
+        # on MI355X (profiles/r2_step5): happens with hipGraphs OFF
+        # (LLMQ_OVERLAP_EAGER) and with either pipe kernel swapped for its
+        # non-pipe fallback; never with AMD_SERIALIZE_KERNEL=3 or with the
+        # overlap off. Root cause not yet isolated — default the overlap
+        # off for this config; LLMQ_OVERLAP_MIXED=1 forces it back on.
+        overlap_env = os.environ.get("LLMQ_OVERLAP_MIXED")
+        overlap_default = not (
+            self.spec.family == "gemma2" and self.spec.head_dim == 128
+        )
         self._overlap_mixed = (
             self.device.type == "cuda"
             and tp == 1
-            and os.environ.get("LLMQ_OVERLAP_MIXED", "1") not in ("0", "false")
+            and (
+                overlap_env not in ("0", "false")
+                if overlap_env is not None
+                else overlap_default
+            )
         )
         self._decode_stream = (
             torch.cuda.Stream(device=self.device) if self._overlap_mixed else None
