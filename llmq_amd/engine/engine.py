@@ -147,14 +147,18 @@ class LLMEngine:
         )
         # Mixed-step overlap: decode graph on a side stream ∥ prefill on the
         # default stream (disjoint KV slots; separate sampler buffers).
-        # gemma-2-27b (the only gemma2 with head_dim 128) deadlocks the GPU
-        # when its decode and prefill kernels run concurrently: GPU busy
-        # 100 %, zero memory traffic, mid-admission, batch >= 128. Bisected
-        # on MI355X (profiles/r2_step5): happens with hipGraphs OFF
-        # (LLMQ_OVERLAP_EAGER) and with either pipe kernel swapped for its
-        # non-pipe fallback; never with AMD_SERIALIZE_KERNEL=3 or with the
-        # overlap off. Root cause not yet isolated — default the overlap
-        # off for this config; LLMQ_OVERLAP_MIXED=1 forces it back on.
+        # gemma-2-27b deadlocks the GPU when its decode and prefill work run
+        # concurrently on two streams: GPU busy 100 %, zero memory traffic.
+        # ROOT CAUSE (bisected on MI355X, profiles/r2_step5): its GEMM
+        # shapes select hipBLASLt stream-k kernels (SK3) whose workgroups
+        # spin on global tile counters assuming the whole grid is resident;
+        # with a second stream's kernels holding CUs, part of the stream-k
+        # grid never launches and the resident part spins forever. Proof:
+        # TENSILE_STREAMK_DATA_PARALLEL=1 (no cross-WG sync) unhangs the
+        # overlap — but also slows 27b's decode GEMMs 37 %, so overlap-off
+        # (sequential split steps) is both the SAFE and the FASTEST config
+        # for this model. LLMQ_OVERLAP_MIXED=1 forces the overlap back on
+        # (combine with TENSILE_STREAMK_DATA_PARALLEL=1 to avoid the hang).
         overlap_env = os.environ.get("LLMQ_OVERLAP_MIXED")
         overlap_default = not (
             self.spec.family == "gemma2" and self.spec.head_dim == 128
