@@ -135,6 +135,18 @@ class Scheduler:
             # Trim oversized prompts from the left (keep the recent window).
             seq.token_ids = seq.token_ids[-self.max_model_len + 1 :]
             seq.prompt_len = len(seq.token_ids)
+        # Reject a prompt the pool could NEVER seat even when empty —
+        # otherwise it blocks the head of the waiting queue forever and the
+        # engine livelocks on empty steps (has_work() stays true while no
+        # batch can ever be built).
+        need = blocks_needed(seq.num_tokens, self.block_size)
+        if need > self.allocator.num_blocks:
+            raise ValueError(
+                f"request {seq.request_id}: prompt needs {need} KV blocks but "
+                f"the pool has only {self.allocator.num_blocks} — it can never "
+                "be admitted (raise gpu_memory_utilization/num_kv_blocks or "
+                "lower max_model_len)"
+            )
         self.waiting.append(seq)
 
     def abort(self, request_id: str) -> bool:
