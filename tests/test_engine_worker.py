@@ -383,3 +383,44 @@ def test_engine_worker_roundtrip_over_amqp():
             await client.disconnect()
 
     run_async(main())
+
+
+def test_never_admittable_job_dead_letters():
+    """A prompt the KV pool can never seat must dead-letter (visible in
+    `llmq errors`), not livelock the engine admission loop. The scheduler
+    raises ValueError at add; the bridge propagates it; the base worker
+    nacks without requeue."""
+    async def main():
+        async with live_broker() as (server, config):
+            client = BrokerClient(config)
+            await client.connect()
+            await client.setup_queue_infrastructure("eqx")
+            await client.publish_jobs("eqx", [
+                Job(id="too-big", prompt="x" * 300, max_tokens=4, temperature=0.0),
+                Job(id="fits", prompt="hello", max_tokens=4, temperature=0.0),
+            ])
+            overrides = dict(TINY_OVERRIDES)
+            overrides["num_kv_blocks"] = 3  # 48 KV slots; 300-token prompt can never fit
+            worker = EngineWorker(
+                "eqx", model="tiny-llama", tensor_parallel_size=1,
+                max_num_seqs=4, max_model_len=512, config=config,
+                engine_overrides=overrides,
+            )
+            task = await _start_worker(worker)
+            # the fitting job completes...
+            results = await _collect_results(client, "eqx", 1, timeout=60.0)
+            assert results[0].id == "fits"
+            # ...and the oversized one is in the DLQ with the loud reason
+            deadline = asyncio.get_event_loop().time() + 30.0
+            failed = []
+            while asyncio.get_event_loop().time() < deadline:
+                failed = await client.get_failed_messages("eqx", limit=10)
+                if failed:
+                    break
+                await asyncio.sleep(0.2)
+            assert failed, "oversized job never dead-lettered"
+            assert any("never" in (f.error_message or "") for f in failed), failed
+            await _stop_worker(worker, task)
+            await client.disconnect()
+
+    run_async(main())
