@@ -81,10 +81,17 @@ def broker_serve(host, port, data_dir, max_retries, ephemeral) -> None:
 @click.option("--template", default=None, help="Prompt template with {placeholders}")
 @click.option("--map", "map_args", multiple=True, help="var=column mapping")
 @click.option("--limit", type=int, default=None, help="Max jobs to submit")
+@click.option("--max-samples", type=int, default=None,
+              help="Alias for --limit (reference main.py:40-42)")
+@click.option("--split", default="train", help="Dataset split (default: train)")
+@click.option("--subset", default=None, help="Dataset subset/config name")
+@click.option("--timeout", "stream_timeout", type=float, default=300.0,
+              help="Idle timeout in seconds waiting for results (with --stream)")
 @click.option("--stream", is_flag=True, help="Echo results to stdout as they arrive")
 @click.option("-p", "--pipeline", "pipeline_path", default=None,
               help="Pipeline YAML: submit to its first stage")
-def submit(queue_name, source, template, map_args, limit, stream, pipeline_path) -> None:
+def submit(queue_name, source, template, map_args, limit, max_samples, split,
+           subset, stream_timeout, stream, pipeline_path) -> None:
     """Submit jobs from SOURCE (jsonl path, '-' for stdin, or dataset id).
 
     With -p/--pipeline, QUEUE_NAME is ignored and jobs go to the pipeline's
@@ -93,10 +100,15 @@ def submit(queue_name, source, template, map_args, limit, stream, pipeline_path)
     from llmq_amd.cli.submit import run_pipeline_submit, run_submit
 
     mapping = _parse_map(map_args)
+    if limit is None:
+        limit = max_samples
     if pipeline_path:
-        run_pipeline_submit(pipeline_path, source, mapping, limit, stream)
+        run_pipeline_submit(pipeline_path, source, mapping, limit, stream,
+                            split=split, subset=subset,
+                            stream_timeout=stream_timeout)
     else:
-        run_submit(queue_name, source, template, mapping, limit, stream)
+        run_submit(queue_name, source, template, mapping, limit, stream,
+                   split=split, subset=subset, stream_timeout=stream_timeout)
 
 
 # --------------------------------------------------------------- receive --
@@ -194,16 +206,30 @@ def worker() -> None:
               help="KV storage dtype; fp8 (OCP e4m3) doubles KV capacity")
 @click.option("--engine-overrides", default=None,
               help='JSON dict of EngineConfig overrides, e.g. \'{"enforce_eager": true}\'')
+@click.option("--data-parallel-size", "-dp", type=int, default=None,
+              help="Model replicas to launch, one process per GPU set "
+                   "(reference main.py:433-439 passes this to vLLM; here it "
+                   "spawns N workers competing on the same queue)")
 def worker_run(model, queue_name, tensor_parallel_size, max_num_seqs, max_model_len,
-               prefetch, kv_cache_dtype, engine_overrides):
+               prefetch, kv_cache_dtype, engine_overrides, data_parallel_size):
     """GPU inference worker (in-tree MI355X engine)."""
     import json as _json
 
-    from llmq_amd.cli.worker import run_engine_worker
+    from llmq_amd.cli.worker import run_engine_worker, run_engine_worker_dp
 
     overrides = _json.loads(engine_overrides) if engine_overrides else {}
     if kv_cache_dtype != "auto":
         overrides.setdefault("kv_cache_dtype", kv_cache_dtype)
+    if data_parallel_size and data_parallel_size > 1:
+        run_engine_worker_dp(
+            model, queue_name, data_parallel_size,
+            tensor_parallel_size=tensor_parallel_size,
+            max_num_seqs=max_num_seqs,
+            max_model_len=max_model_len,
+            prefetch=prefetch,
+            engine_overrides=overrides or None,
+        )
+        return
     run_engine_worker(
         model, queue_name,
         tensor_parallel_size=tensor_parallel_size,
@@ -218,28 +244,38 @@ def worker_run(model, queue_name, tensor_parallel_size, max_num_seqs, max_model_
 @click.argument("queue_name")
 @click.option("--delay", type=float, default=1.0, help="Seconds per job")
 @click.option("--prefetch", type=int, default=None)
-def worker_dummy(queue_name, delay, prefetch):
+@click.option("--concurrency", "-c", type=int, default=None,
+              help="In-flight jobs (reference main.py:466-472; maps to prefetch)")
+def worker_dummy(queue_name, delay, prefetch, concurrency):
     """CPU echo worker (tests/plumbing)."""
     from llmq_amd.cli.worker import run_dummy_worker
 
-    run_dummy_worker(queue_name, delay_s=delay, prefetch=prefetch)
+    run_dummy_worker(queue_name, delay_s=delay, prefetch=prefetch or concurrency)
 
 
 @worker.command("semhash")
 @click.argument("queue_name")
-@click.option("--mode", type=click.Choice(["dedup", "outliers", "representatives"]),
+@click.option("--mode", type=click.Choice([
+                  "dedup", "outliers", "representatives",
+                  # reference mode names (main.py:487-492) accepted as aliases
+                  "deduplicate", "filter_outliers", "find_representative"]),
               default="dedup")
 @click.option("--batch-size", type=int, default=1000)
 @click.option("--threshold", type=float, default=0.9)
 @click.option("--text-field", default=None)
 @click.option("--prefetch", type=int, default=None)
-def worker_semhash(queue_name, mode, batch_size, threshold, text_field, prefetch):
+@click.option("--concurrency", "-c", type=int, default=None,
+              help="In-flight jobs (reference alias; maps to prefetch)")
+def worker_semhash(queue_name, mode, batch_size, threshold, text_field, prefetch,
+                   concurrency):
     """Semantic dedup / filter worker."""
     from llmq_amd.cli.worker import run_semhash_worker
 
+    mode = {"deduplicate": "dedup", "filter_outliers": "outliers",
+            "find_representative": "representatives"}.get(mode, mode)
     run_semhash_worker(
         queue_name, mode=mode, batch_size=batch_size, threshold=threshold,
-        text_field=text_field, prefetch=prefetch,
+        text_field=text_field, prefetch=prefetch or concurrency,
     )
 
 

@@ -121,12 +121,14 @@ def _iter_file_rows(source: str) -> Iterator[Dict[str, Any]]:
             fh.close()
 
 
-def _iter_dataset_rows(source: str, split: str = "train") -> Iterator[Dict[str, Any]]:
+def _iter_dataset_rows(
+    source: str, split: str = "train", subset: Optional[str] = None
+) -> Iterator[Dict[str, Any]]:
     try:
         import datasets  # noqa: PLC0415
     except ImportError as exc:  # pragma: no cover
         raise RuntimeError("`datasets` not available for dataset sources") from exc
-    ds = datasets.load_dataset(source, split=split, streaming=True)
+    ds = datasets.load_dataset(source, name=subset, split=split, streaming=True)
     for row in ds:
         yield dict(row)
 
@@ -141,6 +143,8 @@ class JobSubmitter:
         limit: Optional[int] = None,
         stream: bool = False,
         split: str = "train",
+        subset: Optional[str] = None,
+        stream_timeout: float = 300.0,
         id_prefix: Optional[str] = None,
     ):
         self.queue_name = queue_name
@@ -150,6 +154,8 @@ class JobSubmitter:
         self.limit = limit
         self.stream = stream
         self.split = split
+        self.subset = subset
+        self.stream_timeout = stream_timeout
         self.id_prefix = id_prefix or ("dataset" if _looks_like_dataset(source) else "job")
         self.config = get_config()
         self.client = BrokerClient(self.config)
@@ -160,7 +166,7 @@ class JobSubmitter:
 
     def _rows(self) -> Iterator[Dict[str, Any]]:
         if _looks_like_dataset(self.source):
-            return _iter_dataset_rows(self.source, self.split)
+            return _iter_dataset_rows(self.source, self.split, self.subset)
         return _iter_file_rows(self.source)
 
     def _make_job(self, idx: int, row: Dict[str, Any]) -> Job:
@@ -183,7 +189,8 @@ class JobSubmitter:
         stream_task = None
         expected_ids: set[str] = set()
         if self.stream:
-            stream_task = asyncio.create_task(self._consume_results(expected_ids))
+            stream_task = asyncio.create_task(
+                self._consume_results(expected_ids, idle_timeout=self.stream_timeout))
         start = time.time()
         chunk: List[Job] = []
         skipped = 0
@@ -303,10 +310,14 @@ def run_submit(
     column_mapping: Dict[str, str],
     limit: Optional[int],
     stream: bool,
+    split: str = "train",
+    subset: Optional[str] = None,
+    stream_timeout: float = 300.0,
 ) -> None:
     submitter = JobSubmitter(
         queue_name, source, template=template, column_mapping=column_mapping,
-        limit=limit, stream=stream,
+        limit=limit, stream=stream, split=split, subset=subset,
+        stream_timeout=stream_timeout,
     )
     asyncio.run(submitter.run())
 
@@ -317,9 +328,13 @@ def run_pipeline_submit(
     column_mapping: Dict[str, str],
     limit: Optional[int],
     stream: bool,
+    split: str = "train",
+    subset: Optional[str] = None,
+    stream_timeout: float = 300.0,
 ) -> None:
     pipeline = PipelineConfig.from_yaml_file(pipeline_path)
     submitter = PipelineSubmitter(
-        pipeline, source, column_mapping=column_mapping, limit=limit, stream=stream
+        pipeline, source, column_mapping=column_mapping, limit=limit,
+        stream=stream, split=split, subset=subset, stream_timeout=stream_timeout,
     )
     asyncio.run(submitter.run())

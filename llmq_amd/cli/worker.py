@@ -48,6 +48,59 @@ def run_engine_worker(
     asyncio.run(worker.run())
 
 
+def _dp_child(rank: int, gpus_per_replica: int, kwargs: dict) -> None:
+    import os
+
+    first = rank * gpus_per_replica
+    visible = ",".join(str(first + i) for i in range(gpus_per_replica))
+    os.environ["HIP_VISIBLE_DEVICES"] = visible
+    os.environ.setdefault("CUDA_VISIBLE_DEVICES", visible)
+    run_engine_worker(**kwargs)
+
+
+def run_engine_worker_dp(
+    model: str,
+    queue_name: str,
+    data_parallel_size: int,
+    tensor_parallel_size: Optional[int] = None,
+    **kwargs,
+) -> None:
+    """N independent engine replicas on one node, each pinned to its own GPU
+    slice, all competing on QUEUE_NAME (queue-level data parallelism — the
+    reference's --data-parallel-size, main.py:433-439, realised the way its
+    production scripts actually scale: one process per replica,
+    run_dutch_nemotron.slurm:50-74)."""
+    import multiprocessing as mp
+
+    setup_logging()
+    tp = tensor_parallel_size or 1
+    ctx = mp.get_context("spawn")
+    child_kwargs = dict(
+        model=model, queue_name=queue_name,
+        tensor_parallel_size=tensor_parallel_size, **kwargs,
+    )
+    procs = [
+        ctx.Process(target=_dp_child, args=(r, tp, child_kwargs), daemon=False)
+        for r in range(data_parallel_size)
+    ]
+    for proc in procs:
+        proc.start()
+    logger.info("launched %d engine replicas (tp=%d each) on '%s'",
+                data_parallel_size, tp, queue_name)
+    try:
+        for proc in procs:
+            proc.join()
+    except KeyboardInterrupt:
+        for proc in procs:
+            proc.terminate()
+        for proc in procs:
+            proc.join(timeout=10)
+    finally:
+        bad = [proc.exitcode for proc in procs if proc.exitcode not in (0, None)]
+        if bad:
+            sys.exit(bad[0])
+
+
 def run_dummy_worker(
     queue_name: str,
     delay_s: float = 1.0,

@@ -333,3 +333,43 @@ stages:
     assert res.exit_code == 0, res.output
     lines = [json.loads(l) for l in res.output.splitlines() if l.startswith("{")]
     assert lines and lines[0]["id"] == "pr1"
+
+
+def test_reference_flag_aliases(broker_env, tmp_path, monkeypatch):
+    """Reference CLI parity: --max-samples (submit), -c/--concurrency and
+    reference semhash mode names, -dp on worker run (main.py:40-42, 433-439,
+    466-472, 487-492)."""
+    jobs = tmp_path / "jobs.jsonl"
+    jobs.write_text("\n".join(
+        json.dumps({"id": f"a{i}", "prompt": f"p {i}"}) for i in range(9)
+    ))
+    runner = CliRunner()
+    # --max-samples caps like --limit
+    res = runner.invoke(cli, ["submit", "aliasq", str(jobs), "--max-samples", "4"])
+    assert res.exit_code == 0, res.output
+    assert "Submitted 4 jobs" in res.output
+
+    # semhash reference mode names resolve to the in-tree modes
+    seen = {}
+
+    def fake_semhash(queue_name, mode, batch_size, threshold, text_field, prefetch):
+        seen.update(mode=mode, prefetch=prefetch)
+
+    import llmq_amd.cli.worker as worker_mod
+    monkeypatch.setattr(worker_mod, "run_semhash_worker", fake_semhash)
+    res = runner.invoke(cli, ["worker", "semhash", "sq",
+                              "--mode", "filter_outliers", "-c", "7"])
+    assert res.exit_code == 0, res.output
+    assert seen == {"mode": "outliers", "prefetch": 7}
+
+    # -dp routes to the multi-replica launcher with the right count
+    dp_seen = {}
+
+    def fake_dp(model, queue_name, data_parallel_size, **kw):
+        dp_seen.update(model=model, dp=data_parallel_size,
+                       tp=kw.get("tensor_parallel_size"))
+
+    monkeypatch.setattr(worker_mod, "run_engine_worker_dp", fake_dp)
+    res = runner.invoke(cli, ["worker", "run", "tiny-llama", "dq", "-dp", "3"])
+    assert res.exit_code == 0, res.output
+    assert dp_seen == {"model": "tiny-llama", "dp": 3, "tp": None}
