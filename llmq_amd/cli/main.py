@@ -111,6 +111,38 @@ def submit(queue_name, source, template, map_args, limit, max_samples, split,
                    split=split, subset=subset, stream_timeout=stream_timeout)
 
 
+@cli.command("pipeline", deprecated=True)
+@click.argument("pipeline_config_path")
+@click.argument("jobs_source")
+@click.option("--timeout", "stream_timeout", type=float, default=300.0)
+@click.option("--map", "map_args", multiple=True)
+@click.option("--max-samples", type=int, default=None)
+@click.option("--split", default="train")
+@click.option("--subset", default=None)
+@click.option("--stream", is_flag=True)
+def pipeline_submit_deprecated(pipeline_config_path, jobs_source, stream_timeout,
+                               map_args, max_samples, split, subset, stream):
+    """[deprecated] Use `submit -p PIPELINE.yaml` (reference main.py:150-175
+    keeps this alias; kept for command-line compatibility)."""
+    setup_logging()
+    from llmq_amd.cli.submit import run_pipeline_submit
+
+    run_pipeline_submit(pipeline_config_path, jobs_source, _parse_map(map_args),
+                        max_samples, stream, split=split, subset=subset,
+                        stream_timeout=stream_timeout)
+
+
+@cli.command("receive-pipeline", deprecated=True)
+@click.argument("pipeline_config_path")
+@click.option("--timeout", type=float, default=300.0)
+def receive_pipeline_deprecated(pipeline_config_path, timeout):
+    """[deprecated] Use `receive -p PIPELINE.yaml` (reference main.py:376-378)."""
+    setup_logging()
+    from llmq_amd.cli.receive import run_pipeline_receive
+
+    run_pipeline_receive(pipeline_config_path, timeout, None)
+
+
 # --------------------------------------------------------------- receive --
 
 
