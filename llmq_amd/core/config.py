@@ -37,25 +37,47 @@ def load_dotenv(path: str = ".env") -> None:
         pass
 
 
+# A user switching from the reference keeps a working .env: its names
+# (reference config.py:10-44) are honoured as fallbacks for ours.
+_REFERENCE_ALIASES = {
+    "LLMQ_BROKER_URL": "RABBITMQ_URL",
+    "LLMQ_QUEUE_PREFETCH": "VLLM_QUEUE_PREFETCH",
+    "LLMQ_GPU_MEMORY_UTILIZATION": "VLLM_GPU_MEMORY_UTILIZATION",
+    "LLMQ_MAX_NUM_SEQS": "VLLM_MAX_NUM_SEQS",
+    "LLMQ_MAX_MODEL_LEN": "VLLM_MAX_MODEL_LEN",
+    "LLMQ_MAX_TOKENS": "VLLM_MAX_TOKENS",
+}
+
+
+def _getenv(name: str) -> Optional[str]:
+    value = os.getenv(name)
+    if value is None and name in _REFERENCE_ALIASES:
+        value = os.getenv(_REFERENCE_ALIASES[name])
+    return value
+
+
 def _env(name: str, default: str) -> str:
-    return os.getenv(name, default)
+    value = _getenv(name)
+    return default if value is None else value
 
 
 def _env_int(name: str, default: int) -> int:
-    return int(os.getenv(name, str(default)))
+    value = _getenv(name)
+    return default if value is None else int(value)
 
 
 def _env_float(name: str, default: float) -> float:
-    return float(os.getenv(name, str(default)))
+    value = _getenv(name)
+    return default if value is None else float(value)
 
 
 def _env_opt_int(name: str) -> Optional[int]:
-    value = os.getenv(name)
+    value = _getenv(name)
     return int(value) if value else None
 
 
 def _env_bool(name: str, default: bool) -> bool:
-    value = os.getenv(name)
+    value = _getenv(name)
     if value is None:
         return default
     return value.strip().lower() in ("1", "true", "yes", "on")

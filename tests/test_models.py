@@ -99,3 +99,20 @@ class TestStatsModels:
     def test_error_info(self):
         e = ErrorInfo(job_id="a", error_message="boom", timestamp=utcnow())
         assert e.worker_id is None
+
+
+def test_reference_env_names_accepted(monkeypatch):
+    """A .env written for the reference (RABBITMQ_URL / VLLM_*) configures
+    this framework unchanged (reference config.py:10-44); LLMQ_* wins when
+    both are set."""
+    from llmq_amd.core.config import Config
+
+    monkeypatch.setenv("RABBITMQ_URL", "amqp://u:p@h:5672/vh")
+    monkeypatch.setenv("VLLM_QUEUE_PREFETCH", "1250")
+    monkeypatch.setenv("VLLM_MAX_TOKENS", "4096")
+    cfg = Config()
+    assert cfg.broker_url == "amqp://u:p@h:5672/vh"
+    assert cfg.queue_prefetch == 1250
+    assert cfg.max_tokens == 4096
+    monkeypatch.setenv("LLMQ_QUEUE_PREFETCH", "64")
+    assert Config().queue_prefetch == 64
