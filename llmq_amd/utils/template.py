@@ -68,9 +68,26 @@ def create_job_from_data(
     """
     data = dict(row)
     if column_mapping:
-        for template_var, row_column in column_mapping.items():
-            if row_column in row:
-                data[template_var] = row[row_column]
+        # Three --map value forms, reference parity (submit.py:184-236):
+        #   var=[...]        JSON template, recursively interpolated (messages)
+        #   var=... {col} ...  template string interpolated against the row
+        #   var=column       plain column rename
+        import json as _json
+
+        for template_var, mapping_value in column_mapping.items():
+            mv = mapping_value.strip()
+            if mv.startswith("[") and mv.endswith("]"):
+                try:
+                    template_obj = _json.loads(mv)
+                except _json.JSONDecodeError as exc:
+                    raise ValueError(
+                        f"--map {template_var}: invalid JSON template: {exc}"
+                    ) from exc
+                data[template_var] = format_json_template(template_obj, row)
+            elif "{" in mv and "}" in mv:
+                data[template_var] = resolve_template_string(mv, row)
+            elif mapping_value in row:
+                data[template_var] = row[mapping_value]
     if prompt_template is not None:
         missing = validate_required_fields(prompt_template, data)
         if missing:
@@ -79,6 +96,10 @@ def create_job_from_data(
             )
         extra = {k: v for k, v in data.items() if k not in ("id", "prompt", "messages")}
         return Job(id=job_id, prompt=prompt_template, **extra)
+    if "prompt" not in data and "messages" not in data and "text" in data:
+        # reference fallback (submit.py:226-229): a bare `text` column
+        # becomes the prompt (the common HF-dataset shape)
+        data["prompt"] = str(data["text"])
     if "prompt" in data or "messages" in data:
         data.setdefault("id", job_id)
         data["id"] = data.get("id") or job_id

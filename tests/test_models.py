@@ -116,3 +116,27 @@ def test_reference_env_names_accepted(monkeypatch):
     assert cfg.max_tokens == 4096
     monkeypatch.setenv("LLMQ_QUEUE_PREFETCH", "64")
     assert Config().queue_prefetch == 64
+
+
+def test_map_json_and_template_modes():
+    """--map value forms from the reference (submit.py:184-236): JSON
+    templates build chat messages from rows; {var} strings interpolate; a
+    bare `text` column falls back to prompt."""
+    from llmq_amd.utils.template import create_job_from_data
+
+    row = {"text": "dirty <html>", "lang": "nl"}
+    j = create_job_from_data(
+        row, "d1", None,
+        {"messages": '[{"role": "user", "content": "Clean: {text} ({lang})"}]'})
+    assert j.messages == [{"role": "user", "content": "Clean: dirty <html> (nl)"}]
+    assert j.prompt is None
+
+    j2 = create_job_from_data(row, "d2", None, {"prompt": "Translate {text} to {lang}"})
+    assert j2.prompt == "Translate dirty <html> to nl"
+
+    j3 = create_job_from_data({"text": "as-is"}, "d3", None, None)
+    assert j3.prompt == "as-is"
+
+    import pytest as _pytest
+    with _pytest.raises(ValueError, match="invalid JSON"):
+        create_job_from_data(row, "d4", None, {"messages": "[{broken"})
