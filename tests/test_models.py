@@ -139,4 +139,4 @@ def test_map_json_and_template_modes():
 
     import pytest as _pytest
     with _pytest.raises(ValueError, match="invalid JSON"):
-        create_job_from_data(row, "d4", None, {"messages": "[{broken"})
+        create_job_from_data(row, "d4", None, {"messages": '[{"role": broken}]'})
