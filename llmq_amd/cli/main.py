@@ -311,6 +311,19 @@ def worker_semhash(queue_name, mode, batch_size, threshold, text_field, prefetch
     )
 
 
+@worker.command("filter")
+@click.argument("queue_name")
+@click.argument("field")
+@click.argument("value")
+@click.option("--prefetch", type=int, default=None)
+def worker_filter(queue_name, field, value, prefetch):
+    """Pass jobs whose FIELD equals VALUE; mark the rest filtered
+    (reference README.md:250 documents this worker but never implemented it)."""
+    from llmq_amd.cli.worker import run_filter_worker
+
+    run_filter_worker(queue_name, field, value, prefetch=prefetch)
+
+
 @worker.command("pipeline")
 @click.argument("pipeline_path")
 @click.argument("stage_name")

@@ -144,6 +144,24 @@ def run_semhash_worker(
     asyncio.run(worker.run())
 
 
+def run_filter_worker(
+    queue_name: str,
+    field: str,
+    value: str,
+    prefetch: Optional[int] = None,
+    pipeline: Optional[PipelineConfig] = None,
+    stage_name: Optional[str] = None,
+) -> None:
+    setup_logging(json_output=True)
+    from llmq_amd.workers.filter_worker import FieldFilterWorker
+
+    worker = FieldFilterWorker(
+        queue_name, field=field, value=value, prefetch=prefetch,
+        pipeline=pipeline, stage_name=stage_name,
+    )
+    asyncio.run(worker.run())
+
+
 def run_pipeline_worker(
     pipeline_path: str,
     stage_name: str,
@@ -181,6 +199,15 @@ def run_pipeline_worker(
         run_dummy_worker(
             queue_name,
             delay_s=float(cfg.get("delay_s", 1.0)),
+            prefetch=prefetch,
+            pipeline=pipeline,
+            stage_name=stage_name,
+        )
+    elif stage.worker == "filter" and cfg.get("field") is not None:
+        run_filter_worker(
+            queue_name,
+            field=str(cfg["field"]),
+            value=str(cfg.get("value", "")),
             prefetch=prefetch,
             pipeline=pipeline,
             stage_name=stage_name,

@@ -228,3 +228,33 @@ def test_semhash_dedup_filters_duplicates():
             await client.disconnect()
 
     run_async(main())
+
+
+def test_field_filter_worker():
+    """`worker filter q field value`: matching jobs pass through, the rest
+    are marked filtered (reference README.md:250, unimplemented there)."""
+    from llmq_amd.workers.filter_worker import FieldFilterWorker
+
+    async def main():
+        async with live_broker() as (server, config):
+            client = BrokerClient(config)
+            await client.connect()
+            await client.setup_queue_infrastructure("fq")
+            await client.publish_jobs("fq", [
+                Job(id="keep", prompt="hello", lang="nl"),
+                Job(id="drop", prompt="bye", lang="en"),
+                Job(id="nofield", prompt="zz"),
+            ])
+            worker = FieldFilterWorker("fq", field="lang", value="nl", config=config)
+            task = await _start_worker(worker)
+            results = await _collect_results(client, "fq", 3, timeout=30.0)
+            by_id = {r.id: r for r in results}
+            assert by_id["keep"].result == "hello"
+            assert not getattr(by_id["keep"], "filtered", False)
+            for rid in ("drop", "nofield"):
+                assert getattr(by_id[rid], "filtered", False) is True, by_id[rid]
+                assert by_id[rid].result == ""
+            await _stop_worker(worker, task)
+            await client.disconnect()
+
+    run_async(main())
