@@ -148,7 +148,8 @@ def receive_pipeline_deprecated(pipeline_config_path, timeout):
 
 @cli.command()
 @click.argument("queue_name", required=False)
-@click.option("--timeout", type=float, default=30.0, help="Idle timeout seconds")
+@click.option("--timeout", type=float, default=300.0,
+              help="Idle timeout seconds (reference default 300, main.py:334)")
 @click.option("--limit", type=int, default=None)
 @click.option("--skip-filtered", is_flag=True, help="Drop filtered (semhash) results")
 @click.option("-p", "--pipeline", "pipeline_path", default=None,
@@ -197,7 +198,8 @@ def health(queue_name) -> None:
 
 @cli.command()
 @click.argument("queue_name")
-@click.option("--limit", type=int, default=10)
+@click.option("--limit", type=int, default=100,
+              help="Max errors to show (reference default, main.py:318)")
 def errors(queue_name, limit) -> None:
     """Show dead-lettered jobs from <queue>.failed."""
     setup_logging()
