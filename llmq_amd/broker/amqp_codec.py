@@ -391,6 +391,10 @@ def heartbeat_frame() -> bytes:
     return frame(FRAME_HEARTBEAT, 0, b"")
 
 
+MAX_BODY_SIZE = 64 * 1024 * 1024  # total message body cap (matches the
+                                  # JSON protocol's MAX_FRAME)
+
+
 async def read_frame(reader) -> Tuple[int, int, bytes]:
     """Read one frame from an asyncio StreamReader."""
     hdr = await reader.readexactly(7)

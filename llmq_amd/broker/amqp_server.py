@@ -351,6 +351,14 @@ class AMQPSession:
         if conn.pending_pub is None:
             raise c.AMQPError("content header without basic.publish")
         conn.pending_size, conn.pending_props = c.decode_content_header(payload)
+        # Per-frame size is capped in read_frame, but the declared BODY size
+        # spans many frames — without this bound a client announcing a huge
+        # body makes the server buffer it all (memory DoS).
+        if conn.pending_size > c.MAX_BODY_SIZE:
+            raise c.AMQPError(
+                f"message body too large: {conn.pending_size} "
+                f"(max {c.MAX_BODY_SIZE})"
+            )
         conn.pending_body = b""
         if conn.pending_size == 0:
             self._finish_publish(conn)

@@ -129,3 +129,48 @@ def test_garbage_bytes_do_not_kill_broker():
             await client.disconnect()
 
     run_async(main())
+
+
+def test_huge_declared_body_rejected():
+    """A content header announcing a multi-GB body must be refused before
+    the server buffers it (memory-DoS guard), and the broker survives."""
+    async def main():
+        async with live_broker() as (server, config):
+            host, port = config.broker_host, config.broker_port
+            r, w = await asyncio.open_connection(host, port)
+            w.write(b"AMQP\x00\x00\x09\x01")
+            # minimal handshake: start-ok/tune-ok/open, channel.open, then a
+            # publish with an absurd body size
+            ftype, ch, payload = await c.read_frame(r)       # connection.start
+            w.write(c.method_frame(0, "connection.start-ok",
+                                   client_properties={}, mechanism="PLAIN",
+                                   response="\x00guest\x00guest", locale="en_US"))
+            ftype, ch, payload = await c.read_frame(r)       # connection.tune
+            w.write(c.method_frame(0, "connection.tune-ok",
+                                   channel_max=0, frame_max=131072, heartbeat=0))
+            w.write(c.method_frame(0, "connection.open",
+                                   virtual_host="/", reserved1="", reserved2=False))
+            await c.read_frame(r)                            # connection.open-ok
+            w.write(c.method_frame(1, "channel.open", reserved1=""))
+            await c.read_frame(r)                            # channel.open-ok
+            w.write(c.method_frame(1, "basic.publish", reserved1=0, exchange="",
+                                   routing_key="q", mandatory=False,
+                                   immediate=False))
+            w.write(c.frame(c.FRAME_HEADER, 1,
+                            c.encode_content_header(8 * 1024**3, {})))
+            await w.drain()
+            # server must error the connection, not buffer 8 GB
+            try:
+                data = await asyncio.wait_for(r.read(4096), 5)
+            except (ConnectionError, asyncio.TimeoutError):
+                data = b""
+            w.close()
+            # broker still alive for a normal client
+            client = BrokerClient(config)
+            await client.connect()
+            await client.setup_queue_infrastructure("alive2")
+            await client.publish_job("alive2", Job(id="ok", prompt="x"))
+            assert (await client.get_queue_stats("alive2")).message_count == 1
+            await client.disconnect()
+
+    run_async(main())
