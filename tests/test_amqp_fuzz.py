@@ -174,3 +174,29 @@ def test_huge_declared_body_rejected():
             await client.disconnect()
 
     run_async(main())
+
+
+def test_json_protocol_oversized_line_closes_conn_only():
+    """The native length-bounded JSON protocol: a line beyond MAX_FRAME must
+    close that connection (LimitOverrunError path), never the broker."""
+    async def main():
+        async with live_broker() as (server, config):
+            host, port = config.broker_host, config.broker_port
+            r, w = await asyncio.open_connection(host, port)
+            w.write(b"{" + b"x" * (70 * 1024 * 1024 // 16))  # 4+ MiB, no newline
+            # keep pushing until the server gives up or we hit our own bound
+            try:
+                for _ in range(20):
+                    w.write(b"y" * (4 * 1024 * 1024))
+                    await asyncio.wait_for(w.drain(), 5)
+            except (ConnectionError, asyncio.TimeoutError):
+                pass
+            w.close()
+            client = BrokerClient(config)
+            await client.connect()
+            await client.setup_queue_infrastructure("alive3")
+            await client.publish_job("alive3", Job(id="ok", prompt="x"))
+            assert (await client.get_queue_stats("alive3")).message_count == 1
+            await client.disconnect()
+
+    run_async(main())
