@@ -51,6 +51,7 @@ class ModelSpec:
     sliding_window: int = 0  # 0 = none; Gemma-2: applied on even layers
     embedding_scale: bool = False
     attn_scale: Optional[float] = None  # default 1/sqrt(head_dim)
+    rope_scaling: Optional[dict] = None  # HF rope_scaling (llama3/linear)
     eos_token_id: int = 2
     bos_token_id: int = 1
 
@@ -122,6 +123,15 @@ def _gemma2(name, vocab, hidden, inter, layers, heads, kv_heads, head_dim, **kw)
     )
 
 
+# Llama-3.1/3.2 ship rope_scaling in config.json (rope_type "llama3");
+# real checkpoints are wrong without it, so the presets carry it too.
+_LLAMA31_ROPE = {"rope_type": "llama3", "factor": 8.0, "low_freq_factor": 1.0,
+                 "high_freq_factor": 4.0,
+                 "original_max_position_embeddings": 8192}
+_LLAMA32_ROPE = {"rope_type": "llama3", "factor": 32.0, "low_freq_factor": 1.0,
+                 "high_freq_factor": 4.0,
+                 "original_max_position_embeddings": 8192}
+
 PRESETS = {
     # tiny configs for CPU tests
     # head_dim 64 (not hidden/heads = 16): the HIP attention kernels support
@@ -139,10 +149,12 @@ PRESETS = {
     # Llama 3.2 (vocab 128256, rope theta 500k)
     "llama-3.2-1b": _llama("llama-3.2-1b", 128256, 2048, 8192, 16, 32, 8,
                            head_dim=64, rope_theta=500000.0,
+                           rope_scaling=_LLAMA32_ROPE,
                            max_position_embeddings=131072, tied_embeddings=True,
                            eos_token_id=128001, bos_token_id=128000),
     "llama-3.2-3b": _llama("llama-3.2-3b", 128256, 3072, 8192, 28, 24, 8,
                            head_dim=128, rope_theta=500000.0,
+                           rope_scaling=_LLAMA32_ROPE,
                            max_position_embeddings=131072, tied_embeddings=True,
                            eos_token_id=128001, bos_token_id=128000),
     # Qwen2.5 family (Tower-Plus-2B = Qwen2.5-1.5B arch; -72B = Qwen2.5-72B)
@@ -161,10 +173,12 @@ PRESETS = {
     # Llama-3.1-8B / 70B (same layout; 70B pairs with TP=8 or fits 1 GPU)
     "llama-3.1-8b": _llama("llama-3.1-8b", 128256, 4096, 14336, 32, 32, 8,
                            head_dim=128, rope_theta=500000.0,
+                           rope_scaling=_LLAMA31_ROPE,
                            max_position_embeddings=131072,
                            eos_token_id=128001, bos_token_id=128000),
     "llama-3.1-70b": _llama("llama-3.1-70b", 128256, 8192, 28672, 80, 64, 8,
                             head_dim=128, rope_theta=500000.0,
+                            rope_scaling=_LLAMA31_ROPE,
                             max_position_embeddings=131072,
                             eos_token_id=128001, bos_token_id=128000),
     # Mistral-7B (llama layout, 4096-token sliding window on every layer)
@@ -221,6 +235,13 @@ def spec_from_hf_config(path: Path, name: str) -> ModelSpec:
     heads = cfg["num_attention_heads"]
     hidden = cfg["hidden_size"]
     head_dim = cfg.get("head_dim") or hidden // heads
+    # rope config: transformers <=4.x puts rope_theta at top level and the
+    # scaling dict under "rope_scaling"; transformers 5.x nests BOTH under
+    # "rope_parameters". Normalise to (theta, scaling-or-None).
+    rope_params = cfg.get("rope_scaling") or cfg.get("rope_parameters") or {}
+    rope_theta = cfg.get("rope_theta") or rope_params.get("rope_theta") or 10000.0
+    rope_type = rope_params.get("rope_type") or rope_params.get("type")
+    rope_scaling = dict(rope_params) if rope_type not in (None, "default") else None
     spec = ModelSpec(
         name=name,
         family=family,
@@ -232,7 +253,8 @@ def spec_from_hf_config(path: Path, name: str) -> ModelSpec:
         num_kv_heads=cfg.get("num_key_value_heads", heads),
         head_dim=head_dim,
         rms_eps=cfg.get("rms_norm_eps", 1e-6),
-        rope_theta=cfg.get("rope_theta", 10000.0),
+        rope_theta=rope_theta,
+        rope_scaling=rope_scaling,
         max_position_embeddings=cfg.get("max_position_embeddings", 8192),
         qkv_bias=family == "qwen2",
         tied_embeddings=cfg.get("tie_word_embeddings", False),

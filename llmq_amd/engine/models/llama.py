@@ -104,7 +104,8 @@ class CausalLM:
         self.final_norm: Optional[torch.Tensor] = None
         self.lm_head: Optional[torch.Tensor] = None
         self.rope_cache = ops.build_rope_cache(
-            spec.max_position_embeddings, spec.head_dim, spec.rope_theta, device
+            spec.max_position_embeddings, spec.head_dim, spec.rope_theta, device,
+            rope_scaling=spec.rope_scaling,
         )
         self._alloc()
 

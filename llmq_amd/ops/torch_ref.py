@@ -38,12 +38,42 @@ def fused_add_rmsnorm(
 
 
 def build_rope_cache(
-    max_positions: int, rotary_dim: int, theta: float, device, dtype=torch.float32
+    max_positions: int, rotary_dim: int, theta: float, device,
+    dtype=torch.float32, rope_scaling=None,
 ) -> torch.Tensor:
-    """[max_positions, rotary_dim]: first half cos, second half sin."""
+    """[max_positions, rotary_dim]: first half cos, second half sin.
+
+    rope_scaling: HF config.json dict. Supported: "llama3" (Llama-3.1/3.2
+    wavelength remap, transformers modeling_rope_utils
+    _compute_llama3_parameters) and "linear" (position interpolation —
+    dividing inv_freq by factor is identical to dividing positions).
+    Unknown types raise rather than silently serving wrong rotations.
+    """
+    import math as _math
+
     inv_freq = 1.0 / (
         theta ** (torch.arange(0, rotary_dim, 2, device=device, dtype=torch.float32) / rotary_dim)
     )
+    if rope_scaling:
+        rtype = rope_scaling.get("rope_type") or rope_scaling.get("type")
+        factor = float(rope_scaling.get("factor", 1.0))
+        if rtype == "llama3":
+            low = float(rope_scaling["low_freq_factor"])
+            high = float(rope_scaling["high_freq_factor"])
+            orig = float(rope_scaling["original_max_position_embeddings"])
+            wavelen = 2 * _math.pi / inv_freq
+            smooth = (orig / wavelen - low) / (high - low)
+            smoothed = (1 - smooth) * inv_freq / factor + smooth * inv_freq
+            inv_freq = torch.where(
+                wavelen < orig / high, inv_freq,
+                torch.where(wavelen > orig / low, inv_freq / factor, smoothed),
+            )
+        elif rtype == "linear":
+            inv_freq = inv_freq / factor
+        elif rtype in (None, "default"):
+            pass
+        else:
+            raise ValueError(f"unsupported rope_scaling type: {rtype!r}")
     t = torch.arange(max_positions, device=device, dtype=torch.float32)
     freqs = torch.outer(t, inv_freq)  # [P, rotary_dim/2]
     return torch.cat([freqs.cos(), freqs.sin()], dim=-1).to(dtype)
