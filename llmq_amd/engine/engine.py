@@ -4,6 +4,11 @@ One ``step()`` runs one scheduler iteration (a prefill batch or a decode
 batch), samples, advances sequences, and returns per-request progress.
 The async facade (workers/engine_worker.py AsyncEngineBridge) drives this
 loop on a dedicated engine thread.
+
+This is the MI355X-native replacement for the vLLM AsyncLLMEngine the
+reference consumes (vllm_worker.py:105-123 construction, 183-186
+generate stream): continuous batching up to max_num_seqs,
+paged KV sized for 288 GB HBM3E, hipGraph decode steps, TP over RCCL.
 """
 
 from __future__ import annotations
