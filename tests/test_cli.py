@@ -373,3 +373,24 @@ def test_reference_flag_aliases(broker_env, tmp_path, monkeypatch):
     res = runner.invoke(cli, ["worker", "run", "tiny-llama", "dq", "-dp", "3"])
     assert res.exit_code == 0, res.output
     assert dp_seen == {"model": "tiny-llama", "dp": 3, "tp": None}
+
+
+def test_dp_child_pins_gpu_slices(monkeypatch):
+    """-dp replica r with tp GPUs each must see HIP_VISIBLE_DEVICES
+    [r*tp, (r+1)*tp) — the per-replica pinning the reference's slurm DP
+    loops do by hand (run_dutch_nemotron.slurm:50-74)."""
+    import llmq_amd.cli.worker as worker_mod
+
+    seen = {}
+
+    def fake_run(**kw):
+        import os
+        seen["visible"] = os.environ.get("HIP_VISIBLE_DEVICES")
+        seen["kw"] = kw
+
+    monkeypatch.setattr(worker_mod, "run_engine_worker", fake_run)
+    worker_mod._dp_child(2, 2, {"model": "m", "queue_name": "q"})
+    assert seen["visible"] == "4,5"
+    assert seen["kw"]["model"] == "m"
+    worker_mod._dp_child(0, 1, {"model": "m", "queue_name": "q"})
+    assert seen["visible"] == "0"
