@@ -159,3 +159,60 @@ def test_tp2_logits_match_unsharded(tmp_path):
             solo_tokens.extend(out.new_token_ids)
     assert saved["decode_tokens"] == solo_tokens, (
         saved["decode_tokens"], solo_tokens)
+
+
+def _rank_preempt(rank, world, ckpt, out_dir):
+    import torch.distributed as dist
+
+    from llmq_amd.engine.config import EngineConfig
+    from llmq_amd.engine.engine import LLMEngine
+    from llmq_amd.engine.sampling_params import SamplingParams
+    from llmq_amd.parallel import init_tp
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29597"
+    init_tp(world, rank=rank, backend="gloo")
+    engine = LLMEngine(EngineConfig(
+        model=ckpt, device="cpu", enforce_eager=True,
+        max_num_seqs=4, max_model_len=128,
+        num_kv_blocks=13,  # tight: forces evict+recompute churn mid-decode
+        max_prefill_tokens=64,
+    ), tp_rank=rank, tp_size=world)
+    params = SamplingParams(temperature=0.0, max_tokens=20, ignore_eos=True)
+    for i in range(4):
+        engine.add_request(f"r{i}", prompt_token_ids=list(range(2 + 5 * i, 40 + 5 * i)),
+                           params=params)
+    seqs = list(engine._seqs.values())
+    toks = {f"r{i}": [] for i in range(4)}
+    guard = 0
+    while engine.has_unfinished() and guard < 400:
+        for out in engine.step():
+            toks[out.request_id].extend(out.new_token_ids)
+        guard += 1
+    preempts = max(s.num_preemptions for s in seqs)
+    (out_dir / f"preempt_rank{rank}.json").write_text(
+        json.dumps({"toks": toks, "preempts": preempts, "guard": guard}))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp2_preemption_lockstep(tmp_path):
+    """Preemption-by-recompute under a tight KV pool must be IDENTICAL on
+    every TP rank — divergent evictions would desynchronise the lockstep
+    replicas (collective deadlock or silent token divergence on the 8-GPU
+    tier). Both ranks run the same churn; tokens and preemption counts must
+    match exactly, and preemption must actually fire."""
+    _save_checkpoint(tmp_path)
+
+    import torch.multiprocessing as mp
+
+    mp.spawn(_rank_preempt, args=(2, str(tmp_path), tmp_path), nprocs=2, join=True)
+    r0 = json.loads((tmp_path / "preempt_rank0.json").read_text())
+    r1 = json.loads((tmp_path / "preempt_rank1.json").read_text())
+    assert r0["guard"] < 400 and r1["guard"] < 400
+    assert r0["preempts"] > 0, "pool never forced a preemption - vacuous"
+    assert r0["preempts"] == r1["preempts"]
+    assert r0["toks"] == r1["toks"], "TP ranks diverged under preemption churn"
+    for rid, t in r0["toks"].items():
+        assert len(t) == 20, (rid, t)
