@@ -282,6 +282,15 @@ class PipelineSubmitter(JobSubmitter):
         self.pipeline = pipeline
         first_stage = pipeline.stages[0]
         template = kwargs.pop("template", None) or first_stage.template
+        if template is None and first_stage.messages is not None:
+            # chat-template first stage (reference example-pipeline.yaml):
+            # route rows through the JSON --map machinery so {vars}
+            # interpolate into the messages list per row
+            import json as _json
+
+            mapping = dict(kwargs.pop("column_mapping", None) or {})
+            mapping.setdefault("messages", _json.dumps(first_stage.messages))
+            kwargs["column_mapping"] = mapping
         super().__init__(
             pipeline.get_stage_queue_name(first_stage.name),
             source,

@@ -52,6 +52,13 @@ class PipelineStage(BaseModel):
         return self.config.get("template") or self.config.get("prompt")
 
     @property
+    def messages(self):
+        """Chat-template stage config (reference example-pipeline.yaml uses
+        `messages:` lists with {var} placeholders instead of a string
+        template). Returns the raw template list or None."""
+        return self.config.get("messages")
+
+    @property
     def result_field(self) -> str:
         """Extra-field name under which this stage's output is stored for the
         next stage's template (default: <stage>_result)."""

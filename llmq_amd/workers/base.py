@@ -245,7 +245,17 @@ class BaseWorker(ABC):
         data: Dict[str, Any] = dict(job.extra_fields())
         data[stage.result_field] = result.result
         template = next_stage.template
-        if template is not None:
+        if next_stage.messages is not None:
+            # chat-template stage config (reference example-pipeline.yaml):
+            # interpolate {vars} through the messages list
+            from llmq_amd.utils.template import format_json_template
+
+            next_job = Job(
+                id=job.id,
+                messages=format_json_template(next_stage.messages, data),
+                **data,
+            )
+        elif template is not None:
             next_job = Job(id=job.id, prompt=template, **data)
         else:
             # No template on the next stage: forward the raw result text

@@ -258,3 +258,95 @@ def test_field_filter_worker():
             await client.disconnect()
 
     run_async(main())
+
+
+def test_pipeline_messages_stage_templates(tmp_path):
+    """The reference's example-pipeline.yaml uses `messages:` chat templates
+    in stage config (never applied there — SURVEY §2 quirks). Here: the
+    first stage's messages interpolate per row at submit, and stage N>1
+    messages interpolate the previous stage's result."""
+    import json as _json
+
+    yaml_text = """
+name: msgs-pipe
+stages:
+  - name: first
+    worker: dummy
+    config:
+      messages:
+        - role: user
+          content: "Translate {source_lang}: {source_text}"
+  - name: second
+    worker: dummy
+    config:
+      messages:
+        - role: user
+          content: "Clean up: {first_result}"
+"""
+    cfg_path = tmp_path / "p.yaml"
+    cfg_path.write_text(yaml_text)
+    from llmq_amd.core.pipeline import PipelineConfig
+    from llmq_amd.workers.dummy_worker import DummyWorker
+
+    pipeline = PipelineConfig.from_yaml_file(str(cfg_path))
+
+    async def main():
+        async with live_broker() as (server, config):
+            client = BrokerClient(config)
+            await client.connect()
+            await client.setup_pipeline_infrastructure(pipeline)
+
+            # submit through the PipelineSubmitter mapping path
+            from llmq_amd.cli.submit import PipelineSubmitter
+
+            src = tmp_path / "rows.jsonl"
+            src.write_text(_json.dumps(
+                {"id": "m1", "source_lang": "nl", "source_text": "hallo"}) + "\n")
+            import llmq_amd.core.config as cfg_mod
+            old_get = cfg_mod.get_config
+            cfg_mod.get_config = lambda: config
+            try:
+                sub = PipelineSubmitter(pipeline, str(src))
+            finally:
+                cfg_mod.get_config = old_get
+            sub.config = config
+            sub.client = BrokerClient(config)
+            await sub.client.connect()
+            rows = list(sub._rows())
+            job = sub._job_from_row(rows[0], 0) if hasattr(sub, "_job_from_row") else None
+            if job is None:
+                from llmq_amd.utils.template import create_job_from_data
+                job = create_job_from_data(rows[0], "m1", sub.template, sub.column_mapping)
+            assert job.messages == [
+                {"role": "user", "content": "Translate nl: hallo"}]
+            await sub.client.publish_job(
+                pipeline.get_stage_queue_name("first"), job)
+            await sub.client.disconnect()
+
+            w1 = DummyWorker(pipeline.get_stage_queue_name("first"),
+                             delay_s=0, config=config, pipeline=pipeline,
+                             stage_name="first")
+            w2 = DummyWorker(pipeline.get_stage_queue_name("second"),
+                             delay_s=0, config=config, pipeline=pipeline,
+                             stage_name="second")
+            t1 = await _start_worker(w1)
+            t2 = await _start_worker(w2)
+
+            got = asyncio.Event()
+            final = []
+
+            async def cb(delivery):
+                final.append(Result.model_validate_json(delivery.body))
+                await delivery.ack()
+                got.set()
+
+            await client.consume(
+                pipeline.get_pipeline_results_queue_name(), cb, prefetch=5)
+            await asyncio.wait_for(got.wait(), 30)
+            # stage 1 echoed the chat content; stage 2's messages saw it
+            assert final[0].result == "echo Clean up: echo Translate nl: hallo"
+            await _stop_worker(w1, t1)
+            await _stop_worker(w2, t2)
+            await client.disconnect()
+
+    run_async(main())
